@@ -1,0 +1,423 @@
+"""Base K-FAC preconditioner: hooks + the per-iteration step() pipeline.
+
+Parity surface with reference kfac/base_preconditioner.py:29-479 (same
+constructor contract, hook protocol, step() ordering, state_dict format
+{'steps', scalar hyperparams, 'layers': {name: {'A','G'}}}), with MI355X
+redesigns:
+
+- kl-clip grad scaling is computed entirely on-device: per-layer
+  dot-products accumulate into one fp32 scalar tensor and the final
+  min(1, sqrt(clip/|s|)) is a device op. The reference does a
+  ``.sum().item()`` per layer (base_preconditioner.py:411-435) — a host
+  sync per layer per step that serializes the HIP stream.
+- Layers are walked in reverse registration order in step() so the
+  layers that finished backward first consume their in-flight
+  collectives first (same motivation as the reference, step():310-382).
+"""
+
+from __future__ import annotations
+
+import logging
+import math
+import warnings
+from collections import defaultdict
+from typing import Any
+from typing import Callable
+
+import torch
+
+from kfac_amd.assignment import WorkAssignment
+from kfac_amd.distributed import get_rank
+from kfac_amd.distributed import TorchDistributedCommunicator
+from kfac_amd.layers.base import KFACBaseLayer
+
+logger = logging.getLogger(__name__)
+
+
+class BaseKFACPreconditioner:
+    """Drives per-layer K-FAC state machines through each training step."""
+
+    def __init__(
+        self,
+        layers: dict[torch.nn.Module, tuple[str, KFACBaseLayer]],
+        *,
+        assignment: WorkAssignment,
+        tdc: TorchDistributedCommunicator,
+        factor_update_steps: Callable[[int], int] | int = 1,
+        inv_update_steps: Callable[[int], int] | int = 1,
+        damping: Callable[[int], float] | float = 0.001,
+        factor_decay: Callable[[int], float] | float = 0.95,
+        kl_clip: Callable[[int], float] | float = 0.001,
+        lr: Callable[[int], float] | float = 0.1,
+        accumulation_steps: int = 1,
+        update_factors_in_hook: bool = True,
+        defaults: dict[str, Any] | None = None,
+        loglevel: int = logging.DEBUG,
+    ) -> None:
+        """Init BaseKFACPreconditioner (see KFACPreconditioner for docs)."""
+        if not callable(factor_update_steps) and not 0 < factor_update_steps:
+            raise ValueError('factor_update_steps must be > 0')
+        if not callable(inv_update_steps) and not 0 < inv_update_steps:
+            raise ValueError('inv_update_steps must be > 0')
+        if not callable(damping) and not 0.0 < damping:
+            raise ValueError('damping must be > 0')
+        if not callable(factor_decay) and not 0.0 < factor_decay <= 1:
+            raise ValueError('factor_decay must be in (0, 1]')
+        if (
+            kl_clip is not None
+            and not callable(kl_clip)
+            and not 0.0 < kl_clip
+        ):
+            raise ValueError('kl_clip must be > 0')
+        if not callable(lr) and not 0.0 <= lr:
+            raise ValueError('lr must be >= 0')
+        if not 0 < accumulation_steps:
+            raise ValueError('accumulation_steps must be > 0')
+        if (
+            not callable(inv_update_steps)
+            and not callable(factor_update_steps)
+            and not 0 == inv_update_steps % factor_update_steps
+        ):
+            warnings.warn(
+                'It is suggested that inv_update_steps be an integer '
+                'multiple of factor_update_steps',
+                stacklevel=2,
+            )
+
+        self._accumulation_steps = accumulation_steps
+        self._assignment = assignment
+        self._damping = damping
+        self._defaults = defaults
+        self._factor_decay = factor_decay
+        self._factor_update_steps = factor_update_steps
+        self._inv_update_steps = inv_update_steps
+        self._kl_clip = kl_clip
+        self._layers = layers
+        self._loglevel = loglevel
+        self._lr = lr
+        self._tdc = tdc
+        self._update_factors_in_hook = update_factors_in_hook
+
+        self._steps = 0
+        self._mini_steps: dict[str, int] = defaultdict(int)
+
+        for module in self._layers:
+            module.register_forward_pre_hook(self._save_input)
+            module.register_full_backward_hook(self._save_grad_output)
+
+    def __repr__(self) -> str:
+        params = [
+            ('accumulation_steps', self._accumulation_steps),
+            ('assignment', self._assignment.__class__.__name__),
+            ('damping', self._damping),
+            ('factor_decay', self._factor_decay),
+            ('factor_update_steps', self._factor_update_steps),
+            ('inv_update_steps', self._inv_update_steps),
+            ('kl_clip', self._kl_clip),
+            ('layers', len(self._layers)),
+            ('loglevel', self._loglevel),
+            ('lr', self._lr),
+            ('steps', self.steps),
+            ('update_factors_in_hook', self._update_factors_in_hook),
+        ]
+        if self._defaults is not None:
+            params.extend(list(self._defaults.items()))
+        params = sorted(params, key=lambda x: x[0])
+        body = '\n'.join(f'  {name}={value},' for name, value in params)
+        return f'{self.__class__.__name__}(\n{body}\n)'
+
+    # -- lazily-resolved hyperparameters ----------------------------------
+
+    @property
+    def damping(self) -> float:
+        """Damping at the current step."""
+        return self._damping(self.steps) if callable(self._damping) else self._damping
+
+    @property
+    def factor_decay(self) -> float:
+        """Factor EMA coefficient at the current step."""
+        return (
+            self._factor_decay(self.steps)
+            if callable(self._factor_decay)
+            else self._factor_decay
+        )
+
+    @property
+    def kl_clip(self) -> float | None:
+        """kl-clip at the current step."""
+        return self._kl_clip(self.steps) if callable(self._kl_clip) else self._kl_clip
+
+    @property
+    def lr(self) -> float:
+        """Learning rate at the current step."""
+        return self._lr(self.steps) if callable(self._lr) else self._lr
+
+    @property
+    def factor_update_steps(self) -> int:
+        """Steps between factor updates."""
+        return (
+            self._factor_update_steps(self.steps)
+            if callable(self._factor_update_steps)
+            else self._factor_update_steps
+        )
+
+    @property
+    def inv_update_steps(self) -> int:
+        """Steps between second-order updates."""
+        return (
+            self._inv_update_steps(self.steps)
+            if callable(self._inv_update_steps)
+            else self._inv_update_steps
+        )
+
+    @property
+    def steps(self) -> int:
+        """Completed K-FAC steps."""
+        return self._steps
+
+    # -- checkpointing ------------------------------------------------------
+
+    def state_dict(self, include_factors: bool = True) -> dict[str, Any]:
+        """K-FAC state: steps, non-callable hyperparams, per-layer factors.
+
+        Format matches the reference (base_preconditioner.py:215-247) so
+        checkpoints are interchangeable.
+        """
+        state_dict: dict[str, Any] = {'steps': self.steps}
+        for key, value in (
+            ('factor_update_steps', self._factor_update_steps),
+            ('inv_update_steps', self._inv_update_steps),
+            ('damping', self._damping),
+            ('factor_decay', self._factor_decay),
+            ('kl_clip', self._kl_clip),
+            ('lr', self._lr),
+        ):
+            if not callable(value):
+                state_dict[key] = value
+        if include_factors:
+            state_dict['layers'] = {
+                name: layer.state_dict()
+                for name, layer in self._layers.values()
+            }
+        return state_dict
+
+    def load_state_dict(
+        self,
+        state_dict: dict[str, Any],
+        compute_inverses: bool = True,
+    ) -> None:
+        """Load state; optionally recompute + re-broadcast inverses.
+
+        Reference base_preconditioner.py:249-308.
+        """
+        self._steps = state_dict['steps']
+        for key in (
+            'factor_update_steps',
+            'inv_update_steps',
+            'damping',
+            'factor_decay',
+            'kl_clip',
+            'lr',
+        ):
+            if key in state_dict:
+                setattr(self, f'_{key}', state_dict[key])
+        if 'layers' in state_dict:
+            if len(state_dict['layers']) != len(self._layers):
+                raise ValueError(
+                    'loaded state dict contains a different number of layers',
+                )
+            by_name = {name: layer for name, layer in self._layers.values()}
+            for found_name, layer_state in state_dict['layers'].items():
+                if found_name in by_name:
+                    by_name[found_name].load_state_dict(layer_state)
+        elif compute_inverses:
+            warnings.warn(
+                'Layer factors are not included in the state_dict so '
+                'inverses cannot be computed. Skipping inverse computation.',
+                stacklevel=2,
+            )
+            compute_inverses = False
+        if compute_inverses:
+            for name, layer in self._layers.values():
+                layer.compute_a_inv(damping=self.damping)
+                layer.compute_g_inv(damping=self.damping)
+                if self._assignment.broadcast_inverses():
+                    layer.broadcast_a_inv(
+                        src=self._assignment.inv_worker(name, 'A'),
+                        group=self._assignment.grad_worker_group(name),
+                    )
+                    layer.broadcast_g_inv(
+                        src=self._assignment.inv_worker(name, 'G'),
+                        group=self._assignment.grad_worker_group(name),
+                    )
+
+    # -- the per-iteration pipeline -----------------------------------------
+
+    @torch.no_grad()
+    def step(self) -> None:
+        """One K-FAC step: reduce factors, recompute/broadcast second-order
+        state on schedule, precondition and write back gradients.
+
+        Call after loss.backward() (grads already averaged by DDP) and
+        before optimizer.step(). Reference base_preconditioner.py:310-382.
+        """
+        if (
+            not self._update_factors_in_hook
+            and self.steps % self.factor_update_steps == 0
+        ):
+            for name, layer in reversed(list(self._layers.values())):
+                self._mini_steps[name] = 0
+                layer.update_a_factor(alpha=self.factor_decay)
+                layer.reduce_a_factor(self._assignment.factor_group(name, 'A'))
+                layer.update_g_factor(alpha=self.factor_decay)
+                layer.reduce_g_factor(self._assignment.factor_group(name, 'G'))
+
+        # Launch any trailing factor-allreduce bucket.
+        self._tdc.flush_allreduce_buckets()
+
+        if self.steps % self.inv_update_steps == 0:
+            for name, layer in reversed(list(self._layers.values())):
+                if get_rank() == self._assignment.inv_worker(name, 'A'):
+                    layer.compute_a_inv(damping=self.damping)
+                if (
+                    self._assignment.broadcast_inverses()
+                    and self._assignment.is_grad_worker(name)
+                ):
+                    layer.broadcast_a_inv(
+                        src=self._assignment.inv_worker(name, 'A'),
+                        group=self._assignment.grad_worker_group(name),
+                    )
+                if get_rank() == self._assignment.inv_worker(name, 'G'):
+                    layer.compute_g_inv(damping=self.damping)
+                if (
+                    self._assignment.broadcast_inverses()
+                    and self._assignment.is_grad_worker(name)
+                ):
+                    layer.broadcast_g_inv(
+                        src=self._assignment.inv_worker(name, 'G'),
+                        group=self._assignment.grad_worker_group(name),
+                    )
+            self._tdc.flush_allreduce_buckets()
+
+        for name, layer in reversed(list(self._layers.values())):
+            if self._assignment.is_grad_worker(name):
+                layer.preconditioned_grad(damping=self.damping)
+            if self._assignment.broadcast_gradients():
+                layer.broadcast_grad(
+                    src=self._assignment.src_grad_worker(name),
+                    group=self._assignment.grad_receiver_group(name),
+                )
+        self._tdc.flush_allreduce_buckets()
+
+        scale = None if self.kl_clip is None else self._compute_grad_scale()
+
+        for _, layer in reversed(list(self._layers.values())):
+            layer.update_grad(scale=scale)
+
+        self._steps += 1
+        self._mini_steps = defaultdict(int)
+
+    def reset_batch(self) -> None:
+        """Drop accumulated factor contributions from the current batch."""
+        for _, layer in self._layers.values():
+            layer.reset_batch()
+
+    def memory_usage(self) -> dict[str, int]:
+        """Approximate bytes used by K-FAC state on this rank."""
+        sizes: dict[str, int] = defaultdict(int)
+        self._tdc.flush_allreduce_buckets()
+        for _, layer in self._layers.values():
+            for key, size in layer.memory_usage().items():
+                sizes[key] += size
+        sizes['total'] = sum(sizes.values())
+        return sizes
+
+    def _compute_grad_scale(self) -> float | torch.Tensor:
+        """scale = min(1, sqrt(kl_clip / |sum_l <precon_l, grad_l>| lr^2)).
+
+        On GPU this is computed without any host synchronization: the
+        per-layer dot products accumulate into a single device scalar via
+        a fused reduction and the min/sqrt runs on-device; update_grad()
+        consumes the 0-dim tensor directly. On CPU the float path matches
+        the reference numerics (base_preconditioner.py:411-435).
+        """
+        from kfac_amd import ops
+
+        layers = list(self._layers.values())
+        if len(layers) == 0:
+            return 1.0
+        device = layers[0][1].module.device
+        lr = self.lr
+        kl_clip = self.kl_clip
+        assert kl_clip is not None
+        if device.type == 'cuda':
+            accum = torch.zeros((), dtype=torch.float32, device=device)
+            for _, layer in reversed(layers):
+                grad = layer.grad
+                if grad is None:
+                    raise AssertionError(
+                        'layer gradient has not been preconditioned',
+                    )
+                ops.kl_clip_accum(accum, grad, layer.module.get_grad())
+            return ops.grad_scale_from_accum(accum, kl_clip, lr)
+        vg_sum = 0.0
+        for _, layer in reversed(layers):
+            grad = layer.grad
+            if grad is None:
+                raise AssertionError('layer gradient has not been preconditioned')
+            vg_sum += float(
+                (grad.to(torch.float32) * layer.module.get_grad().to(torch.float32)).sum()
+                * lr
+                * lr,
+            )
+        if vg_sum == 0.0:
+            return 1.0
+        return min(1.0, math.sqrt(kl_clip / abs(vg_sum)))
+
+    # -- hooks ---------------------------------------------------------------
+
+    @torch.no_grad()
+    def _save_input(
+        self,
+        module: torch.nn.Module,
+        input_: list[torch.Tensor],
+    ) -> None:
+        """Forward-pre hook: accumulate the A-factor contribution."""
+        if not module.training:
+            return
+        if self.steps % self.factor_update_steps == 0:
+            name, layer = self._layers[module]
+            layer.save_layer_input(input_)
+            self._mini_steps[name] += 1
+            if (
+                self._update_factors_in_hook
+                and self._mini_steps[name] % self._accumulation_steps == 0
+            ):
+                layer.update_a_factor(alpha=self.factor_decay)
+                layer.reduce_a_factor(self._assignment.factor_group(name, 'A'))
+
+    @torch.no_grad()
+    def _save_grad_output(
+        self,
+        module: torch.nn.Module,
+        grad_input: tuple[torch.Tensor, ...] | torch.Tensor,
+        grad_output: tuple[torch.Tensor, ...] | torch.Tensor,
+    ) -> None:
+        """Backward hook: accumulate the G-factor contribution.
+
+        The factor allreduce launched here rides RCCL's side stream and
+        overlaps the rest of backward (and DDP's own gradient allreduce).
+        """
+        if not module.training:
+            return
+        if self.steps % self.factor_update_steps == 0:
+            name, layer = self._layers[module]
+            if isinstance(grad_output, torch.Tensor):
+                grad_output = (grad_output,)
+            layer.save_layer_grad_output(grad_output)
+            if (
+                self._update_factors_in_hook
+                and self._mini_steps[name] % self._accumulation_steps == 0
+            ):
+                layer.update_g_factor(alpha=self.factor_decay)
+                layer.reduce_g_factor(self._assignment.factor_group(name, 'G'))

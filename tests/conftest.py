@@ -1,0 +1,31 @@
+"""Pytest configuration: register the gpu marker; repo-root imports."""
+
+from __future__ import annotations
+
+import os
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config: pytest.Config) -> None:
+    config.addinivalue_line(
+        'markers',
+        'gpu: test requires a ROCm GPU (run on MI355X via gpurun)',
+    )
+
+
+def pytest_collection_modifyitems(
+    config: pytest.Config,
+    items: list[pytest.Item],
+) -> None:
+    import torch
+
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason='no GPU available')
+    for item in items:
+        if 'gpu' in item.keywords:
+            item.add_marker(skip)
