@@ -40,6 +40,9 @@ def _worker(
     os.environ['RANK'] = str(rank)
     os.environ['WORLD_SIZE'] = str(world_size)
     try:
+        import torch
+
+        torch.set_num_threads(1)
         dist.init_process_group('gloo', rank=rank, world_size=world_size)
         func(*args, **kwargs)
         dist.barrier()

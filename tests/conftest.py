@@ -6,8 +6,14 @@ import os
 import sys
 
 import pytest
+import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+# The test harness forks workers (testing/distributed.py). A fork while the
+# parent's OpenMP/intra-op thread pool holds a lock deadlocks the child, so
+# keep the parent single-threaded for the whole session.
+torch.set_num_threads(1)
 
 
 def pytest_configure(config: pytest.Config) -> None:
