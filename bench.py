@@ -1,0 +1,202 @@
+"""Flagship benchmark: ResNet-50 + K-FAC on MI355X (BASELINE.json metric).
+
+Measures whole-job images/sec for synthetic ImageNet-shaped training
+(bs=64 per GPU, bf16 autocast, random-init weights) with the K-FAC
+preconditioner on the reference example's headline schedule
+(factor_update_steps=10, inv_update_steps=100 —
+reference examples/torch_imagenet_resnet.py:158-167 defaults).
+
+Contract (driver):
+  python bench.py --gpus N --steps K --warmup W
+  N>1 is launched via torch.distributed.run with one rank per GPU (RCCL).
+  W untimed warmup steps; EXACTLY K timed steps bracketed by
+  barrier + torch.cuda.synchronize on both sides; MAX elapsed over ranks;
+  rank 0 prints ONE JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def parse_args() -> argparse.Namespace:
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=100)
+    p.add_argument('--warmup', type=int, default=20)
+    p.add_argument('--batch-size', type=int, default=64, help='per-GPU batch')
+    p.add_argument('--model', type=str, default='resnet50')
+    p.add_argument(
+        '--strategy',
+        type=str,
+        default=None,
+        choices=['comm-opt', 'hybrid-opt', 'mem-opt'],
+        help='KAISA strategy (default: comm-opt at world 1, hybrid-opt else)',
+    )
+    p.add_argument('--factor-update-steps', type=int, default=10)
+    p.add_argument('--inv-update-steps', type=int, default=100)
+    p.add_argument('--no-kfac', action='store_true')
+    return p.parse_args()
+
+
+def main() -> None:
+    args = parse_args()
+    rank = int(os.environ.get('RANK', '0'))
+    world = int(os.environ.get('WORLD_SIZE', '1'))
+    local_rank = int(os.environ.get('LOCAL_RANK', '0'))
+
+    assert torch.cuda.is_available(), 'bench.py requires a GPU'
+    torch.cuda.set_device(local_rank)
+    device = torch.device('cuda', local_rank)
+
+    if world > 1:
+        os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        torch.distributed.init_process_group('nccl')
+
+    from kfac_amd import KFACPreconditioner
+    from kfac_amd import ops
+    from kfac_amd.enums import DistributedStrategy
+    from kfac_amd.models import resnet50, resnet101, resnet152
+
+    assert ops.extension_available(), (
+        'HIP extension not built: run __graft_entry__.build() first'
+    )
+
+    torch.manual_seed(1234 + rank)
+    model_fn = {'resnet50': resnet50, 'resnet101': resnet101, 'resnet152': resnet152}[args.model]
+    model = model_fn().to(device)
+    model.train()
+
+    if world > 1:
+        model = torch.nn.parallel.DistributedDataParallel(
+            model, device_ids=[local_rank],
+        )
+
+    if args.strategy is None:
+        strategy = (
+            DistributedStrategy.COMM_OPT
+            if world == 1
+            else DistributedStrategy.HYBRID_OPT
+        )
+        strategy_name = 'comm-opt' if world == 1 else 'hybrid-opt'
+    else:
+        strategy_name = args.strategy
+        strategy = {
+            'comm-opt': DistributedStrategy.COMM_OPT,
+            'hybrid-opt': DistributedStrategy.HYBRID_OPT,
+            'mem-opt': DistributedStrategy.MEM_OPT,
+        }[args.strategy]
+    # HYBRID at world 2 means grad_worker_fraction 0.5 = 1/world -> MEM_OPT
+    # semantics; KAISAAssignment handles every divisible fraction.
+
+    lr = 0.1
+    optimizer = torch.optim.SGD(
+        model.parameters(), lr=lr, momentum=0.9, weight_decay=5e-5,
+    )
+    precon = None
+    if not args.no_kfac:
+        precon = KFACPreconditioner(
+            model,
+            factor_update_steps=args.factor_update_steps,
+            inv_update_steps=args.inv_update_steps,
+            damping=0.001,
+            factor_decay=0.95,
+            kl_clip=0.001,
+            lr=lr,
+            grad_worker_fraction=strategy,
+            accumulation_steps=1,
+            allreduce_bucket_cap_mb=25.0,
+            compute_eigenvalue_outer_product=True,
+            skip_layers=[],
+        )
+
+    bs = args.batch_size
+    x = torch.randn(bs, 3, 224, 224, device=device)
+    y = torch.randint(0, 1000, (bs,), device=device)
+    criterion = torch.nn.CrossEntropyLoss()
+
+    precond_times: list[float] = []
+
+    def one_step(timed: bool) -> None:
+        optimizer.zero_grad(set_to_none=True)
+        with torch.autocast('cuda', dtype=torch.bfloat16):
+            loss = criterion(model(x), y)
+        loss.backward()
+        if precon is not None:
+            t0 = time.perf_counter() if timed else 0.0
+            precon.step()
+            if timed:
+                precond_times.append(time.perf_counter() - t0)
+        optimizer.step()
+
+    for _ in range(args.warmup):
+        one_step(False)
+
+    if world > 1:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    start = time.perf_counter()
+    for _ in range(args.steps):
+        one_step(True)
+    torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - start
+
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    global_batch = bs * world
+    images_per_sec = global_batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            'metric': 'images/sec (whole node) ResNet-50 + K-FAC precond',
+            'value': images_per_sec,
+            'unit': 'images/sec',
+            'n_gpus': world,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': ms_per_step,
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': 'bf16',
+            'data': 'synthetic',
+            'config': {
+                'model': args.model,
+                'global_batch': global_batch,
+                'seq_len': None,
+                'parallelism': f'dp{world}',
+                'image_size': 224,
+                'kfac': not args.no_kfac,
+                'strategy': strategy_name,
+                'factor_update_steps': args.factor_update_steps,
+                'inv_update_steps': args.inv_update_steps,
+                'precond_step_ms_mean': (
+                    sum(precond_times) / len(precond_times) * 1000.0
+                    if precond_times
+                    else None
+                ),
+            },
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

@@ -1,0 +1,380 @@
+// Python binding for the kfac_amd HIP extension (_kfaccore).
+//
+// Host-side glue only: tensor checks, dtype dispatch, temp allocation,
+// launches on the current HIP stream. Device code lives in syrk.hip /
+// gemm.hip (pure HIP, gfx950).
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <hip/hip_runtime.h>
+
+namespace kfac {
+
+enum class Dtype : int { F32 = 0, BF16 = 1, F16 = 2 };
+
+template <typename T>
+hipError_t cov_linear_t(hipStream_t, const T*, long, int, int, bool, float*, float, float);
+template <typename T>
+hipError_t cov_conv_a_t(hipStream_t, const T*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
+template <typename T>
+hipError_t cov_conv_g_t(hipStream_t, const T*, int, int, int, int, float*, float, float);
+hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float);
+template <typename T>
+hipError_t kl_clip_accum_t(hipStream_t, float*, const T*, const T*, long);
+hipError_t triu_pack_f32(hipStream_t, float*, const float*, int);
+hipError_t triu_unpack_f32(hipStream_t, float*, const float*, int);
+
+}  // namespace kfac
+
+namespace {
+
+#define CHECK_OK(expr)                                                       \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));     \
+  } while (0)
+
+hipStream_t current_stream(const torch::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+void check_gpu_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void check_out_factor(const torch::Tensor& out, int n) {
+  check_gpu_contig(out, "out");
+  TORCH_CHECK(out.scalar_type() == torch::kFloat32, "out must be fp32");
+  TORCH_CHECK(
+      out.dim() == 2 && out.size(0) == n && out.size(1) == n,
+      "out must be (",
+      n,
+      ", ",
+      n,
+      "), got ",
+      out.sizes());
+}
+
+template <typename F32Fn, typename Bf16Fn, typename F16Fn>
+void dispatch_dtype(
+    torch::ScalarType st,
+    F32Fn f32fn,
+    Bf16Fn bf16fn,
+    F16Fn f16fn) {
+  switch (st) {
+    case torch::kFloat32:
+      f32fn();
+      break;
+    case torch::kBFloat16:
+      bf16fn();
+      break;
+    case torch::kFloat16:
+      f16fn();
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype for kfac op: ", st);
+  }
+}
+
+void cov_linear(
+    torch::Tensor a,
+    torch::Tensor out,
+    bool bias,
+    double beta,
+    double coeff) {
+  check_gpu_contig(a, "a");
+  TORCH_CHECK(a.dim() == 2, "a must be 2D");
+  int M = (int)a.size(0);
+  int K = (int)a.size(1);
+  int n = K + (bias ? 1 : 0);
+  check_out_factor(out, n);
+  auto stream = current_stream(a);
+  float* outp = out.data_ptr<float>();
+  dispatch_dtype(
+      a.scalar_type(),
+      [&] {
+        CHECK_OK(kfac::cov_linear_t<float>(
+            stream, a.data_ptr<float>(), K, M, K, bias, outp, (float)beta,
+            (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::cov_linear_t<__hip_bfloat16>(
+            stream, (const __hip_bfloat16*)a.data_ptr(), K, M, K, bias, outp,
+            (float)beta, (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::cov_linear_t<__half>(
+            stream, (const __half*)a.data_ptr(), K, M, K, bias, outp,
+            (float)beta, (float)coeff));
+      });
+}
+
+void cov_conv_a(
+    torch::Tensor x,
+    torch::Tensor out,
+    int64_t kh,
+    int64_t kw,
+    int64_t sh,
+    int64_t sw,
+    int64_t ph,
+    int64_t pw,
+    bool bias,
+    double beta,
+    double coeff_scale) {
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  int Nb = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+      W = (int)x.size(3);
+  int OH = (int)((H + 2 * ph - kh) / sh + 1);
+  int OW = (int)((W + 2 * pw - kw) / sw + 1);
+  TORCH_CHECK(OH > 0 && OW > 0, "empty conv output");
+  int K = (int)(C * kh * kw);
+  int n = K + (bias ? 1 : 0);
+  check_out_factor(out, n);
+  long m = (long)Nb * OH * OW;
+  long s = (long)OH * OW;
+  // patches/spatial (ones included) then cov scale 1/m:
+  // coeff = coeff_scale / (m * s^2)   (see ops/reference.py cov_conv_a)
+  double coeff = coeff_scale / ((double)m * (double)s * (double)s);
+  auto stream = current_stream(x);
+  float* outp = out.data_ptr<float>();
+  dispatch_dtype(
+      x.scalar_type(),
+      [&] {
+        CHECK_OK(kfac::cov_conv_a_t<float>(
+            stream, x.data_ptr<float>(), Nb, C, H, W, (int)kh, (int)kw,
+            (int)sh, (int)sw, (int)ph, (int)pw, bias, outp, (float)beta,
+            (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::cov_conv_a_t<__hip_bfloat16>(
+            stream, (const __hip_bfloat16*)x.data_ptr(), Nb, C, H, W, (int)kh,
+            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, bias, outp,
+            (float)beta, (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::cov_conv_a_t<__half>(
+            stream, (const __half*)x.data_ptr(), Nb, C, H, W, (int)kh,
+            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, bias, outp,
+            (float)beta, (float)coeff));
+      });
+}
+
+void cov_conv_g(
+    torch::Tensor g,
+    torch::Tensor out,
+    double beta,
+    double coeff_scale) {
+  check_gpu_contig(g, "g");
+  TORCH_CHECK(g.dim() == 4, "g must be NCHW");
+  int Nb = (int)g.size(0), C = (int)g.size(1), OH = (int)g.size(2),
+      OW = (int)g.size(3);
+  check_out_factor(out, C);
+  long m = (long)Nb * OH * OW;
+  long s = (long)OH * OW;
+  double coeff = coeff_scale / ((double)m * (double)s * (double)s);
+  auto stream = current_stream(g);
+  float* outp = out.data_ptr<float>();
+  dispatch_dtype(
+      g.scalar_type(),
+      [&] {
+        CHECK_OK(kfac::cov_conv_g_t<float>(
+            stream, g.data_ptr<float>(), Nb, C, OH, OW, outp, (float)beta,
+            (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::cov_conv_g_t<__hip_bfloat16>(
+            stream, (const __hip_bfloat16*)g.data_ptr(), Nb, C, OH, OW, outp,
+            (float)beta, (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::cov_conv_g_t<__half>(
+            stream, (const __half*)g.data_ptr(), Nb, C, OH, OW, outp,
+            (float)beta, (float)coeff));
+      });
+}
+
+// Shared tail of the eigen precondition chain:
+//   given v2 (already divided/multiplied), out = QG @ (v2 @ QA^T).
+torch::Tensor eigen_tail(
+    const torch::Tensor& v2,
+    const torch::Tensor& qa,
+    const torch::Tensor& qg,
+    hipStream_t stream) {
+  int m = (int)v2.size(0);
+  int n = (int)v2.size(1);
+  auto t2 = torch::empty_like(v2);
+  // t2 = v2 @ QA^T : [m,n] x [n,n]^T
+  CHECK_OK(kfac::gemm_f32(
+      stream, t2.data_ptr<float>(), v2.data_ptr<float>(),
+      qa.data_ptr<float>(), m, n, n, false, true, 0, nullptr, nullptr, 0.f));
+  auto out = torch::empty_like(v2);
+  // out = QG @ t2 : [m,m] x [m,n]
+  CHECK_OK(kfac::gemm_f32(
+      stream, out.data_ptr<float>(), qg.data_ptr<float>(),
+      t2.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr, 0.f));
+  return out;
+}
+
+torch::Tensor precond_eigen_fused(
+    torch::Tensor grad,
+    torch::Tensor qa,
+    torch::Tensor qg,
+    torch::Tensor dgda) {
+  check_gpu_contig(grad, "grad");
+  check_gpu_contig(qa, "qa");
+  check_gpu_contig(qg, "qg");
+  check_gpu_contig(dgda, "dgda");
+  auto dtype = grad.scalar_type();
+  auto g32 = grad.to(torch::kFloat32);
+  int m = (int)g32.size(0);
+  int n = (int)g32.size(1);
+  TORCH_CHECK(qa.size(0) == n && qg.size(0) == m, "shape mismatch");
+  auto stream = current_stream(grad);
+  auto t1 = torch::empty_like(g32);
+  // t1 = grad @ QA
+  CHECK_OK(kfac::gemm_f32(
+      stream, t1.data_ptr<float>(), g32.data_ptr<float>(),
+      qa.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f));
+  auto v2 = torch::empty_like(g32);
+  // v2 = (QG^T @ t1) * dgda   (epilogue-fused elementwise)
+  CHECK_OK(kfac::gemm_f32(
+      stream, v2.data_ptr<float>(), qg.data_ptr<float>(),
+      t1.data_ptr<float>(), m, n, m, true, false, 1, dgda.data_ptr<float>(),
+      nullptr, 0.f));
+  return eigen_tail(v2, qa, qg, stream).to(dtype);
+}
+
+torch::Tensor precond_eigen(
+    torch::Tensor grad,
+    torch::Tensor qa,
+    torch::Tensor qg,
+    torch::Tensor dg,
+    torch::Tensor da,
+    double damping) {
+  check_gpu_contig(grad, "grad");
+  check_gpu_contig(qa, "qa");
+  check_gpu_contig(qg, "qg");
+  check_gpu_contig(dg, "dg");
+  check_gpu_contig(da, "da");
+  auto dtype = grad.scalar_type();
+  auto g32 = grad.to(torch::kFloat32);
+  int m = (int)g32.size(0);
+  int n = (int)g32.size(1);
+  auto stream = current_stream(grad);
+  auto t1 = torch::empty_like(g32);
+  CHECK_OK(kfac::gemm_f32(
+      stream, t1.data_ptr<float>(), g32.data_ptr<float>(),
+      qa.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f));
+  auto v2 = torch::empty_like(g32);
+  // v2 = (QG^T @ t1) / (outer(dg, da) + damping)
+  CHECK_OK(kfac::gemm_f32(
+      stream, v2.data_ptr<float>(), qg.data_ptr<float>(),
+      t1.data_ptr<float>(), m, n, m, true, false, 2, dg.data_ptr<float>(),
+      da.data_ptr<float>(), (float)damping));
+  return eigen_tail(v2, qa, qg, stream).to(dtype);
+}
+
+torch::Tensor precond_inverse(
+    torch::Tensor grad,
+    torch::Tensor a_inv,
+    torch::Tensor g_inv) {
+  check_gpu_contig(grad, "grad");
+  check_gpu_contig(a_inv, "a_inv");
+  check_gpu_contig(g_inv, "g_inv");
+  auto dtype = grad.scalar_type();
+  auto g32 = grad.to(torch::kFloat32);
+  auto ai = a_inv.scalar_type() == torch::kFloat32 ? a_inv
+                                                   : a_inv.to(torch::kFloat32);
+  auto gi = g_inv.scalar_type() == torch::kFloat32 ? g_inv
+                                                   : g_inv.to(torch::kFloat32);
+  int m = (int)g32.size(0);
+  int n = (int)g32.size(1);
+  auto stream = current_stream(grad);
+  auto t1 = torch::empty_like(g32);
+  // t1 = G^-1 @ grad
+  CHECK_OK(kfac::gemm_f32(
+      stream, t1.data_ptr<float>(), gi.data_ptr<float>(),
+      g32.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr,
+      0.f));
+  auto out = torch::empty_like(g32);
+  // out = t1 @ A^-1
+  CHECK_OK(kfac::gemm_f32(
+      stream, out.data_ptr<float>(), t1.data_ptr<float>(),
+      ai.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f));
+  return out.to(dtype);
+}
+
+void kl_clip_accum(
+    torch::Tensor accum,
+    torch::Tensor precon,
+    torch::Tensor grad) {
+  check_gpu_contig(accum, "accum");
+  check_gpu_contig(precon, "precon");
+  check_gpu_contig(grad, "grad");
+  TORCH_CHECK(accum.scalar_type() == torch::kFloat32, "accum must be fp32");
+  TORCH_CHECK(precon.scalar_type() == grad.scalar_type(), "dtype mismatch");
+  TORCH_CHECK(precon.numel() == grad.numel(), "numel mismatch");
+  auto stream = current_stream(accum);
+  long n = precon.numel();
+  dispatch_dtype(
+      precon.scalar_type(),
+      [&] {
+        CHECK_OK(kfac::kl_clip_accum_t<float>(
+            stream, accum.data_ptr<float>(), precon.data_ptr<float>(),
+            grad.data_ptr<float>(), n));
+      },
+      [&] {
+        CHECK_OK(kfac::kl_clip_accum_t<__hip_bfloat16>(
+            stream, accum.data_ptr<float>(),
+            (const __hip_bfloat16*)precon.data_ptr(),
+            (const __hip_bfloat16*)grad.data_ptr(), n));
+      },
+      [&] {
+        CHECK_OK(kfac::kl_clip_accum_t<__half>(
+            stream, accum.data_ptr<float>(), (const __half*)precon.data_ptr(),
+            (const __half*)grad.data_ptr(), n));
+      });
+}
+
+torch::Tensor triu_pack(torch::Tensor x) {
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(x.dim() == 2 && x.size(0) == x.size(1), "x must be square");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "triu_pack: fp32 only");
+  int n = (int)x.size(0);
+  auto out = torch::empty(
+      {(long)n * (n + 1) / 2},
+      x.options());
+  CHECK_OK(kfac::triu_pack_f32(
+      current_stream(x), out.data_ptr<float>(), x.data_ptr<float>(), n));
+  return out;
+}
+
+torch::Tensor triu_unpack(torch::Tensor v, int64_t n) {
+  check_gpu_contig(v, "v");
+  TORCH_CHECK(v.scalar_type() == torch::kFloat32, "triu_unpack: fp32 only");
+  TORCH_CHECK(v.numel() == n * (n + 1) / 2, "packed size mismatch");
+  auto out = torch::empty({n, n}, v.options());
+  CHECK_OK(kfac::triu_unpack_f32(
+      current_stream(v), out.data_ptr<float>(), v.data_ptr<float>(), (int)n));
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "kfac_amd CDNA4 (gfx950) kernels";
+  m.def("cov_linear", &cov_linear, "fused [a,1]^T[a,1] SYRK accumulation");
+  m.def("cov_conv_a", &cov_conv_a, "fused im2col+SYRK A-factor accumulation");
+  m.def("cov_conv_g", &cov_conv_g, "fused NCHW-transpose SYRK G-factor");
+  m.def("precond_eigen_fused", &precond_eigen_fused, "Kronecker precondition (prediv)");
+  m.def("precond_eigen", &precond_eigen, "Kronecker precondition (dg/da)");
+  m.def("precond_inverse", &precond_inverse, "G^-1 grad A^-1");
+  m.def("kl_clip_accum", &kl_clip_accum, "device-side kl-clip accumulation");
+  m.def("triu_pack", &triu_pack, "pack upper triangle");
+  m.def("triu_unpack", &triu_unpack, "unpack upper triangle");
+}

@@ -1,0 +1,255 @@
+// Small-matrix MFMA GEMM with fused epilogues for the K-FAC
+// precondition chain (K8/K9 of SURVEY §2.4):
+//
+//   eigen:   v1 = QG^T @ grad @ QA ; v2 = v1 * dGdA (or / (dg x da + damping))
+//            out = QG @ v2 @ QA^T
+//   inverse: out = G^-1 @ grad @ A^-1
+//
+// Factor edges are 64..~4k, so a 64x64-tile mfma_f32_16x16x4_f32 kernel
+// with LDS staging covers the whole size range; the elementwise
+// eigenvalue denominator is fused into the second GEMM's epilogue so the
+// chain is exactly 4 kernel launches per layer with no intermediate
+// elementwise pass over HBM.
+
+#include "common.h"
+
+namespace kfac {
+
+constexpr int GBT = 64;
+constexpr int GBK = 32;
+constexpr int GLDS = GBT + 1;
+
+enum class Epilogue : int { NONE = 0, MUL = 1, DIV_OUTER = 2 };
+
+// Stage op(A)'s [k, rows i0..i0+63] slab into lds[k][i].
+// src is f32; trans selects A[k][i] vs A[i][k].
+__device__ __forceinline__ void stage_gemm(
+    const float* __restrict__ src,
+    int rows,      // logical rows of op(src) slab dimension (i extent)
+    int ks,        // logical k extent
+    long ld,       // leading dim of the PHYSICAL matrix
+    bool trans,    // false: physical[k][i] = src[k*ld+i] feeds lds[k][i]
+                   // true:  physical[i][k] = src[i*ld+k] feeds lds[k][i]
+    int k0,
+    int i0,
+    float (*lds)[GLDS],
+    int tid) {
+#pragma unroll
+  for (int e = 0; e < (GBK * GBT) / 256; ++e) {
+    int idx = tid + e * 256;
+    int k = idx / GBT;
+    int i = idx % GBT;
+    int gk = k0 + k;
+    int gi = i0 + i;
+    float v = 0.0f;
+    if (gk < ks && gi < rows) {
+      v = trans ? src[(long)gi * ld + gk] : src[(long)gk * ld + gi];
+    }
+    lds[k][i] = v;
+  }
+}
+
+template <Epilogue EPI>
+__global__ __launch_bounds__(256) void gemm_kernel(
+    float* __restrict__ c,
+    const float* __restrict__ a,
+    const float* __restrict__ b,
+    int M,
+    int N,
+    int K,
+    bool ta,  // use A^T (A physical is [K_phys x M_phys] interpreted so that
+              // op(A)[m][k]; ta=false: A[m][k] (ld=K); ta=true: A[k][m] (ld=M)
+    bool tb,  // op(B)[k][n]; tb=false: B[k][n] (ld=N); tb=true: B[n][k] (ld=K)
+    const float* __restrict__ e1,  // MUL: dgda [M x N]; DIV_OUTER: dg [M]
+    const float* __restrict__ e2,  // DIV_OUTER: da [N]
+    float damping) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+  const int i0 = blockIdx.x * GBT;  // row block of C
+  const int j0 = blockIdx.y * GBT;  // col block of C
+
+  __shared__ float lds_a[GBK][GLDS];
+  __shared__ float lds_b[GBK][GLDS];
+
+  f32x4 acc[2][2] = {};
+
+  for (int k0 = 0; k0 < K; k0 += GBK) {
+    // op(A) slab: lds_a[k][i] = op(A)[i0+i][k0+k]
+    //   ta=false: A[m][k], physical ld = K -> element A[(i0+i)*K + (k0+k)]
+    //   ta=true : A[k][m], physical ld = M -> element A[(k0+k)*M + (i0+i)]
+    stage_gemm(a, M, K, ta ? (long)M : (long)K, !ta, k0, i0, lds_a, tid);
+    // op(B) slab: lds_b[k][j] = op(B)[k0+k][j0+j]
+    //   tb=false: B[k][n], ld = N -> B[(k0+k)*N + (j0+j)]
+    //   tb=true : B[n][k], ld = K -> B[(j0+j)*K + (k0+k)]
+    stage_gemm(b, N, K, tb ? (long)K : (long)N, tb, k0, j0, lds_b, tid);
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < GBK; kk += 4) {
+      const int krow = kk + (lane >> 4);
+      float a0 = lds_a[krow][wr * 32 + (lane & 15)];
+      float a1 = lds_a[krow][wr * 32 + 16 + (lane & 15)];
+      float b0 = lds_b[krow][wc * 32 + (lane & 15)];
+      float b1 = lds_b[krow][wc * 32 + 16 + (lane & 15)];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i0 + wr * 32 + fi * 16 + (lane >> 4) * 4 + r;
+        int col = j0 + wc * 32 + fj * 16 + (lane & 15);
+        if (row < M && col < N) {
+          float v = acc[fi][fj][r];
+          if constexpr (EPI == Epilogue::MUL) {
+            v *= e1[(long)row * N + col];
+          } else if constexpr (EPI == Epilogue::DIV_OUTER) {
+            v /= (e1[row] * e2[col] + damping);
+          }
+          c[(long)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
+hipError_t gemm_f32(
+    hipStream_t stream,
+    float* c,
+    const float* a,
+    const float* b,
+    int M,
+    int N,
+    int K,
+    bool ta,
+    bool tb,
+    int epilogue,
+    const float* e1,
+    const float* e2,
+    float damping) {
+  dim3 grid(ceil_div(M, GBT), ceil_div(N, GBT));
+  switch (static_cast<Epilogue>(epilogue)) {
+    case Epilogue::NONE:
+      gemm_kernel<Epilogue::NONE><<<grid, 256, 0, stream>>>(
+          c, a, b, M, N, K, ta, tb, nullptr, nullptr, 0.0f);
+      break;
+    case Epilogue::MUL:
+      gemm_kernel<Epilogue::MUL><<<grid, 256, 0, stream>>>(
+          c, a, b, M, N, K, ta, tb, e1, nullptr, 0.0f);
+      break;
+    case Epilogue::DIV_OUTER:
+      gemm_kernel<Epilogue::DIV_OUTER><<<grid, 256, 0, stream>>>(
+          c, a, b, M, N, K, ta, tb, e1, e2, damping);
+      break;
+  }
+  return hipGetLastError();
+}
+
+// ------------------------------------------------------- kl-clip reduction
+
+// accum += sum(precon * grad) without any host round trip (K10).
+template <typename T>
+__global__ void kl_clip_kernel(
+    float* __restrict__ accum,
+    const T* __restrict__ precon,
+    const T* __restrict__ grad,
+    long n) {
+  __shared__ float warp_sums[4];
+  float s = 0.0f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    s += to_f32(precon[i]) * to_f32(grad[i]);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+  }
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  if (lane == 0) warp_sums[wave] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.0f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += warp_sums[w];
+    atomicAdd(accum, t);
+  }
+}
+
+template <typename T>
+hipError_t kl_clip_accum_t(
+    hipStream_t stream,
+    float* accum,
+    const T* precon,
+    const T* grad,
+    long n) {
+  int threads = 256;
+  int blocks = (int)min((n + threads - 1) / threads, (long)1024);
+  blocks = max(blocks, 1);
+  kl_clip_kernel<T><<<blocks, threads, 0, stream>>>(accum, precon, grad, n);
+  return hipGetLastError();
+}
+
+template hipError_t kl_clip_accum_t<float>(hipStream_t, float*, const float*, const float*, long);
+template hipError_t kl_clip_accum_t<__hip_bfloat16>(hipStream_t, float*, const __hip_bfloat16*, const __hip_bfloat16*, long);
+template hipError_t kl_clip_accum_t<__half>(hipStream_t, float*, const __half*, const __half*, long);
+
+// ------------------------------------------------------- triu pack/unpack
+
+// Symmetric wire format (K12): packed index of (i,j), j>=i is
+// i*n - i*(i-1)/2 + (j-i).
+__global__ void triu_pack_kernel(
+    float* __restrict__ packed,
+    const float* __restrict__ x,
+    int n) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long total = (long)n * n;
+  if (idx >= total) return;
+  int i = (int)(idx / n);
+  int j = (int)(idx % n);
+  if (j < i) return;
+  long p = (long)i * n - (long)i * (i - 1) / 2 + (j - i);
+  packed[p] = x[idx];
+}
+
+__global__ void triu_unpack_kernel(
+    float* __restrict__ x,
+    const float* __restrict__ packed,
+    int n) {
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long total = (long)n * n;
+  if (idx >= total) return;
+  int i = (int)(idx / n);
+  int j = (int)(idx % n);
+  int r = min(i, j);
+  int c = max(i, j);
+  long p = (long)r * n - (long)r * (r - 1) / 2 + (c - r);
+  x[idx] = packed[p];
+}
+
+hipError_t triu_pack_f32(hipStream_t stream, float* packed, const float* x, int n) {
+  long total = (long)n * n;
+  int threads = 256;
+  triu_pack_kernel<<<(total + threads - 1) / threads, threads, 0, stream>>>(
+      packed, x, n);
+  return hipGetLastError();
+}
+
+hipError_t triu_unpack_f32(hipStream_t stream, float* x, const float* packed, int n) {
+  long total = (long)n * n;
+  int threads = 256;
+  triu_unpack_kernel<<<(total + threads - 1) / threads, threads, 0, stream>>>(
+      x, packed, n);
+  return hipGetLastError();
+}
+
+}  // namespace kfac

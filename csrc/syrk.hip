@@ -1,0 +1,302 @@
+// Fused covariance (SYRK) kernels for K-FAC factor accumulation.
+//
+// Computes out = beta*out + coeff * (A^T A) for three virtual matrices A
+// without ever materializing them in HBM:
+//   - LinearAcc:    A = [activations, ones] (bias column fused)
+//   - ConvPatchAcc: A = im2col(x) patches [+ ones]  (K2+K3 of SURVEY §2.4)
+//   - ConvGradAcc:  A = NCHW -> (N*OH*OW, C) transpose view
+//
+// MI355X design notes:
+//   * mfma_f32_16x16x4_f32: exact f32 numerics (matches the fp32 torch
+//     reference bit-for-bit in summation class), 157 TF chip peak.
+//   * C = A^T A is symmetric: only tiles ti<=tj are computed; off-diagonal
+//     tiles are written twice (mirrored), making the output exactly
+//     symmetric by construction (the reference symmetrizes explicitly,
+//     kfac/layers/utils.py:55-57).
+//   * The reduction (batch) dimension M is split across blockIdx.z so even
+//     a single-tile factor (64x64) launches enough workgroups to fill 256
+//     CUs; partial tiles accumulate with fp32 atomics (bytes are trivial
+//     against the MFMA work).
+//   * bf16/f16 inputs are converted in the LDS staging pass and accumulated
+//     in fp32 (better numerics than the reference's bf16 matmul).
+
+#include "common.h"
+
+namespace kfac {
+
+constexpr int BT = 64;   // C tile edge
+constexpr int BK = 32;   // m-slice staged per iteration
+constexpr int LDS_STRIDE = BT + 1;  // break bank alignment for b32 reads
+
+// ---------------------------------------------------------------- accessors
+
+template <typename T>
+struct LinearAcc {
+  const T* a;
+  long lda;
+  int M;       // rows
+  int K;       // real columns (bias column at index K if Ncols == K+1)
+  int Ncols;   // K + (bias ? 1 : 0)
+  static constexpr bool kLaneAlongCols = true;
+
+  __device__ __forceinline__ float load(int m, int i) const {
+    if (m >= M || i >= Ncols) return 0.0f;
+    if (i < K) return to_f32(a[(long)m * lda + i]);
+    return 1.0f;  // fused bias-ones column
+  }
+};
+
+template <typename T>
+struct ConvPatchAcc {
+  const T* x;
+  int C, H, W, OH, OW;
+  int kh, kw, sh, sw, ph, pw;
+  int M;     // Nb*OH*OW
+  int K;     // C*kh*kw
+  int Ncols;
+  static constexpr bool kLaneAlongCols = false;  // lanes along m (w-contig)
+
+  __device__ __forceinline__ float load(int m, int i) const {
+    if (m >= M || i >= Ncols) return 0.0f;
+    if (i >= K) return 1.0f;
+    int ow = m % OW;
+    int t = m / OW;
+    int oh = t % OH;
+    int n = t / OH;
+    int s = i % kw;
+    int t2 = i / kw;
+    int r = t2 % kh;
+    int c = t2 / kh;
+    int h = oh * sh - ph + r;
+    int w = ow * sw - pw + s;
+    if (h < 0 || h >= H || w < 0 || w >= W) return 0.0f;
+    return to_f32(x[((long)(n * C + c) * H + h) * W + w]);
+  }
+};
+
+template <typename T>
+struct ConvGradAcc {
+  const T* g;
+  int C, OH, OW;
+  int M;  // Nb*OH*OW
+  int Ncols;
+  static constexpr bool kLaneAlongCols = false;
+
+  __device__ __forceinline__ float load(int m, int j) const {
+    if (m >= M || j >= Ncols) return 0.0f;
+    int ow = m % OW;
+    int t = m / OW;
+    int oh = t % OH;
+    int n = t / OH;
+    return to_f32(g[((long)(n * C + j) * OH + oh) * OW + ow]);
+  }
+};
+
+// ---------------------------------------------------------------- kernel
+
+template <typename Acc>
+__device__ __forceinline__ void stage_tile(
+    const Acc& acc,
+    float (*lds)[LDS_STRIDE],
+    int m0,
+    int col0,
+    int tid) {
+  // 32 x 64 elements, 256 threads -> 8 each.
+#pragma unroll
+  for (int e = 0; e < (BK * BT) / 256; ++e) {
+    int idx = tid + e * 256;
+    int k, i;
+    if (Acc::kLaneAlongCols) {
+      k = idx / BT;
+      i = idx % BT;
+    } else {
+      k = idx % BK;
+      i = idx / BK;
+    }
+    lds[k][i] = acc.load(m0 + k, col0 + i);
+  }
+}
+
+template <typename AccL, typename AccR>
+__global__ __launch_bounds__(256) void syrk_kernel(
+    float* __restrict__ out,
+    int N,            // factor edge (= Ncols)
+    AccL accl,
+    AccR accr,
+    int m_per_split,
+    float coeff,
+    bool same_tile_ok  // accl and accr view the same matrix
+) {
+  const int ti = blockIdx.x;
+  const int tj = blockIdx.y;
+  if (tj < ti) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;  // wave row (0..1) -> 32-row quadrant
+  const int wc = wave & 1;   // wave col
+
+  const int i0 = ti * BT;
+  const int j0 = tj * BT;
+  const int m_begin = blockIdx.z * m_per_split;
+  const int m_end = min(accl.M, m_begin + m_per_split);
+  if (m_begin >= m_end) return;
+
+  __shared__ float lds_l[BK][LDS_STRIDE];
+  __shared__ float lds_r[BK][LDS_STRIDE];
+  const bool diag = same_tile_ok && (ti == tj);
+
+  f32x4 acc[2][2] = {};
+
+  for (int m0 = m_begin; m0 < m_end; m0 += BK) {
+    stage_tile(accl, lds_l, m0, i0, tid);
+    if (!diag) {
+      stage_tile(accr, lds_r, m0, j0, tid);
+    }
+    __syncthreads();
+    auto rbuf = diag ? lds_l : lds_r;
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int krow = kk + (lane >> 4);
+      float a0 = lds_l[krow][wr * 32 + (lane & 15)];
+      float a1 = lds_l[krow][wr * 32 + 16 + (lane & 15)];
+      float b0 = rbuf[krow][wc * 32 + (lane & 15)];
+      float b1 = rbuf[krow][wc * 32 + 16 + (lane & 15)];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // C/D fragment layout (16x16): row = (lane>>4)*4 + r, col = lane&15.
+  const bool mirror = (ti != tj);
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i0 + wr * 32 + fi * 16 + (lane >> 4) * 4 + r;
+        int col = j0 + wc * 32 + fj * 16 + (lane & 15);
+        if (row < N && col < N) {
+          float v = coeff * acc[fi][fj][r];
+          atomicAdd(&out[(long)row * N + col], v);
+          if (mirror) {
+            atomicAdd(&out[(long)col * N + row], v);
+          }
+        }
+      }
+    }
+  }
+}
+
+__global__ void scale_kernel(float* out, long n, float beta) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    out[i] = (beta == 0.0f) ? 0.0f : out[i] * beta;
+  }
+}
+
+// ---------------------------------------------------------------- launchers
+
+static int pick_splits(int M, int n_tiles) {
+  // Fill the chip: want >= ~1024 workgroups; each split handles >= 1 BK.
+  int max_splits = max(1, M / (8 * BK));
+  int want = max(1, 1024 / max(1, n_tiles));
+  return min(max_splits, want);
+}
+
+template <typename AccL>
+static hipError_t launch_syrk(
+    hipStream_t stream,
+    float* out,
+    int N,
+    const AccL& acc,
+    float beta,
+    float coeff) {
+  long n2 = (long)N * N;
+  int threads = 256;
+  scale_kernel<<<(n2 + threads - 1) / threads, threads, 0, stream>>>(
+      out, n2, beta);
+  int nt = ceil_div(N, BT);
+  int n_tiles = nt * (nt + 1) / 2;
+  int splits = pick_splits(acc.M, n_tiles);
+  int m_per_split = ceil_div(ceil_div(acc.M, splits), BK) * BK;
+  splits = ceil_div(acc.M, m_per_split);
+  dim3 grid(nt, nt, splits);
+  syrk_kernel<AccL, AccL><<<grid, 256, 0, stream>>>(
+      out, N, acc, acc, m_per_split, coeff, true);
+  return hipGetLastError();
+}
+
+template <typename T>
+hipError_t cov_linear_t(
+    hipStream_t stream,
+    const T* a,
+    long lda,
+    int M,
+    int K,
+    bool bias,
+    float* out,
+    float beta,
+    float coeff) {
+  LinearAcc<T> acc{a, lda, M, K, K + (bias ? 1 : 0)};
+  return launch_syrk(stream, out, acc.Ncols, acc, beta, coeff);
+}
+
+template <typename T>
+hipError_t cov_conv_a_t(
+    hipStream_t stream,
+    const T* x,
+    int Nb,
+    int C,
+    int H,
+    int W,
+    int kh,
+    int kw,
+    int sh,
+    int sw,
+    int ph,
+    int pw,
+    bool bias,
+    float* out,
+    float beta,
+    float coeff) {
+  int OH = (H + 2 * ph - kh) / sh + 1;
+  int OW = (W + 2 * pw - kw) / sw + 1;
+  int K = C * kh * kw;
+  ConvPatchAcc<T> acc{x,  C,  H,  W,  OH, OW, kh,
+                      kw, sh, sw, ph, pw, Nb * OH * OW,
+                      K,  K + (bias ? 1 : 0)};
+  return launch_syrk(stream, out, acc.Ncols, acc, beta, coeff);
+}
+
+template <typename T>
+hipError_t cov_conv_g_t(
+    hipStream_t stream,
+    const T* g,
+    int Nb,
+    int C,
+    int OH,
+    int OW,
+    float* out,
+    float beta,
+    float coeff) {
+  ConvGradAcc<T> acc{g, C, OH, OW, Nb * OH * OW, C};
+  return launch_syrk(stream, out, acc.Ncols, acc, beta, coeff);
+}
+
+// Explicit instantiations used by the binding.
+template hipError_t cov_linear_t<float>(hipStream_t, const float*, long, int, int, bool, float*, float, float);
+template hipError_t cov_linear_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, long, int, int, bool, float*, float, float);
+template hipError_t cov_linear_t<__half>(hipStream_t, const __half*, long, int, int, bool, float*, float, float);
+template hipError_t cov_conv_a_t<float>(hipStream_t, const float*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
+template hipError_t cov_conv_a_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
+template hipError_t cov_conv_a_t<__half>(hipStream_t, const __half*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
+template hipError_t cov_conv_g_t<float>(hipStream_t, const float*, int, int, int, int, float*, float, float);
+template hipError_t cov_conv_g_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, int, int, int, int, float*, float, float);
+template hipError_t cov_conv_g_t<__half>(hipStream_t, const __half*, int, int, int, int, float*, float, float);
+
+}  // namespace kfac

@@ -1,0 +1,100 @@
+"""GPU end-to-end tests: convergence + one ResNet-50 K-FAC step on MI355X."""
+
+from __future__ import annotations
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope='module', autouse=True)
+def _require_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip('no GPU')
+    from kfac_amd import ops
+
+    assert ops.extension_available(), 'HIP extension must be built'
+
+
+def test_lenet_converges_gpu() -> None:
+    from kfac_amd import KFACPreconditioner
+    from testing.models import LeNet
+
+    torch.manual_seed(42)
+    model = LeNet().cuda()
+    x = torch.randn(64, 1, 28, 28, device='cuda')
+    y = torch.randint(0, 10, (64,), device='cuda')
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    precon = KFACPreconditioner(
+        model, factor_update_steps=1, inv_update_steps=2, lr=0.01,
+    )
+    losses = []
+    for _ in range(15):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[0] > losses[-1], losses
+
+
+def test_gpu_step_matches_cpu_reference() -> None:
+    """Full per-layer pipeline numerics: GPU (HIP kernels) vs CPU (torch).
+
+    Same model, same data, one K-FAC step on both devices; the written
+    gradients must agree to fp32 tolerance.
+    """
+    from kfac_amd import KFACPreconditioner
+    from testing.models import LeNet
+
+    results = {}
+    for device in ('cpu', 'cuda'):
+        torch.manual_seed(123)
+        model = LeNet().to(device)
+        x = torch.randn(32, 1, 28, 28, device=device)
+        y = torch.randint(0, 10, (32,), device=device)
+        precon = KFACPreconditioner(
+            model, factor_update_steps=1, inv_update_steps=1, lr=0.01,
+        )
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        results[device] = {
+            name: p.grad.detach().cpu().clone()
+            for name, p in model.named_parameters()
+        }
+    for name in results['cpu']:
+        torch.testing.assert_close(
+            results['cuda'][name],
+            results['cpu'][name],
+            rtol=5e-3,
+            atol=5e-4,
+            msg=lambda m: f'{name}: {m}',
+        )
+
+
+def test_resnet50_bf16_step() -> None:
+    from kfac_amd import KFACPreconditioner
+    from kfac_amd.models import resnet50
+
+    torch.manual_seed(0)
+    model = resnet50().cuda()
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    precon = KFACPreconditioner(
+        model, factor_update_steps=1, inv_update_steps=1, lr=0.01,
+    )
+    x = torch.randn(8, 3, 224, 224, device='cuda')
+    y = torch.randint(0, 1000, (8,), device='cuda')
+    for _ in range(2):
+        opt.zero_grad()
+        with torch.autocast('cuda', dtype=torch.bfloat16):
+            loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
