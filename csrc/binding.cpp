@@ -199,7 +199,9 @@ void cov_conv_g(
 }
 
 // Shared tail of the eigen precondition chain:
-//   given v2 (already divided/multiplied), out = QG @ (v2 @ QA^T).
+//   given v2 (already divided/multiplied), out = (QG @ v2) @ QA^T.
+// Association matches the reference (eigen.py:385: qg @ v2 @ qa.t() is
+// left-associative) so numerics line up with the fp32 torch reference.
 torch::Tensor eigen_tail(
     const torch::Tensor& v2,
     const torch::Tensor& qa,
@@ -208,15 +210,15 @@ torch::Tensor eigen_tail(
   int m = (int)v2.size(0);
   int n = (int)v2.size(1);
   auto t2 = torch::empty_like(v2);
-  // t2 = v2 @ QA^T : [m,n] x [n,n]^T
+  // t2 = QG @ v2 : [m,m] x [m,n]
   CHECK_OK(kfac::gemm_f32(
-      stream, t2.data_ptr<float>(), v2.data_ptr<float>(),
-      qa.data_ptr<float>(), m, n, n, false, true, 0, nullptr, nullptr, 0.f));
+      stream, t2.data_ptr<float>(), qg.data_ptr<float>(),
+      v2.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr, 0.f));
   auto out = torch::empty_like(v2);
-  // out = QG @ t2 : [m,m] x [m,n]
+  // out = t2 @ QA^T : [m,n] x [n,n]^T
   CHECK_OK(kfac::gemm_f32(
-      stream, out.data_ptr<float>(), qg.data_ptr<float>(),
-      t2.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr, 0.f));
+      stream, out.data_ptr<float>(), t2.data_ptr<float>(),
+      qa.data_ptr<float>(), m, n, n, false, true, 0, nullptr, nullptr, 0.f));
   return out;
 }
 
@@ -236,15 +238,15 @@ torch::Tensor precond_eigen_fused(
   TORCH_CHECK(qa.size(0) == n && qg.size(0) == m, "shape mismatch");
   auto stream = current_stream(grad);
   auto t1 = torch::empty_like(g32);
-  // t1 = grad @ QA
+  // t1 = QG^T @ grad (reference association: (qg.t() @ grad) @ qa)
   CHECK_OK(kfac::gemm_f32(
-      stream, t1.data_ptr<float>(), g32.data_ptr<float>(),
-      qa.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f));
+      stream, t1.data_ptr<float>(), qg.data_ptr<float>(),
+      g32.data_ptr<float>(), m, n, m, true, false, 0, nullptr, nullptr, 0.f));
   auto v2 = torch::empty_like(g32);
-  // v2 = (QG^T @ t1) * dgda   (epilogue-fused elementwise)
+  // v2 = (t1 @ QA) * dgda   (epilogue-fused elementwise)
   CHECK_OK(kfac::gemm_f32(
-      stream, v2.data_ptr<float>(), qg.data_ptr<float>(),
-      t1.data_ptr<float>(), m, n, m, true, false, 1, dgda.data_ptr<float>(),
+      stream, v2.data_ptr<float>(), t1.data_ptr<float>(),
+      qa.data_ptr<float>(), m, n, n, false, false, 1, dgda.data_ptr<float>(),
       nullptr, 0.f));
   return eigen_tail(v2, qa, qg, stream).to(dtype);
 }
@@ -267,14 +269,15 @@ torch::Tensor precond_eigen(
   int n = (int)g32.size(1);
   auto stream = current_stream(grad);
   auto t1 = torch::empty_like(g32);
+  // t1 = QG^T @ grad
   CHECK_OK(kfac::gemm_f32(
-      stream, t1.data_ptr<float>(), g32.data_ptr<float>(),
-      qa.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f));
+      stream, t1.data_ptr<float>(), qg.data_ptr<float>(),
+      g32.data_ptr<float>(), m, n, m, true, false, 0, nullptr, nullptr, 0.f));
   auto v2 = torch::empty_like(g32);
-  // v2 = (QG^T @ t1) / (outer(dg, da) + damping)
+  // v2 = (t1 @ QA) / (outer(dg, da) + damping)
   CHECK_OK(kfac::gemm_f32(
-      stream, v2.data_ptr<float>(), qg.data_ptr<float>(),
-      t1.data_ptr<float>(), m, n, m, true, false, 2, dg.data_ptr<float>(),
+      stream, v2.data_ptr<float>(), t1.data_ptr<float>(),
+      qa.data_ptr<float>(), m, n, n, false, false, 2, dg.data_ptr<float>(),
       da.data_ptr<float>(), (float)damping));
   return eigen_tail(v2, qa, qg, stream).to(dtype);
 }

@@ -153,10 +153,20 @@ def precond_eigen(
         ext = _require_ext('precond_eigen')
         if ext is not None:
             if dgda is not None:
-                return ext.precond_eigen_fused(grad.contiguous(), qa, qg, dgda)
+                return ext.precond_eigen_fused(
+                    grad.contiguous(),
+                    qa.contiguous(),
+                    qg.contiguous(),
+                    dgda.contiguous(),
+                )
             assert da is not None and dg is not None
             return ext.precond_eigen(
-                grad.contiguous(), qa, qg, dg, da, float(damping),
+                grad.contiguous(),
+                qa.contiguous(),
+                qg.contiguous(),
+                dg.contiguous(),
+                da.contiguous(),
+                float(damping),
             )
     return ref.precond_eigen(
         grad, qa, qg, dgda=dgda, da=da, dg=dg, damping=damping,
@@ -172,7 +182,9 @@ def precond_inverse(
     if grad.is_cuda:
         ext = _require_ext('precond_inverse')
         if ext is not None:
-            return ext.precond_inverse(grad.contiguous(), a_inv, g_inv)
+            return ext.precond_inverse(
+                grad.contiguous(), a_inv.contiguous(), g_inv.contiguous(),
+            )
     return ref.precond_inverse(grad, a_inv, g_inv)
 
 

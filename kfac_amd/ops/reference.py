@@ -159,7 +159,9 @@ def eigh(x: torch.Tensor, *, clamp: bool = True) -> tuple[torch.Tensor, torch.Te
     d, q = torch.linalg.eigh(x.to(torch.float32))
     if clamp:
         d = torch.clamp(d, min=0.0)
-    return d, q
+    # rocSOLVER can return the eigenvector matrix as a transposed view;
+    # downstream kernels and collectives need contiguous storage.
+    return d.contiguous(), q.contiguous()
 
 
 def inv_damped(x: torch.Tensor, damping: float) -> torch.Tensor:
