@@ -31,8 +31,15 @@ _EXT_TRIED = False
 
 
 def _load_ext() -> Any:
-    """Import the in-tree HIP extension (kfac_amd/_kfaccore*.so)."""
+    """Import the in-tree HIP extension (kfac_amd/_kfaccore*.so).
+
+    KFAC_AMD_FORCE_EAGER=1 disables the extension entirely — used only to
+    benchmark the torch-eager baseline (the reference implementation's
+    op-for-op algorithm) on the same hardware.
+    """
     global _EXT, _EXT_TRIED
+    if os.environ.get('KFAC_AMD_FORCE_EAGER', '0') == '1':
+        return None
     if not _EXT_TRIED:
         _EXT_TRIED = True
         try:
@@ -52,7 +59,10 @@ def extension_available() -> bool:
 def _require_ext(op: str) -> Any:
     ext = _load_ext()
     if ext is None:
-        if os.environ.get('KFAC_AMD_ALLOW_EAGER', '0') == '1':
+        if (
+            os.environ.get('KFAC_AMD_ALLOW_EAGER', '0') == '1'
+            or os.environ.get('KFAC_AMD_FORCE_EAGER', '0') == '1'
+        ):
             return None
         raise RuntimeError(
             f'kfac_amd op {op!r} called on a GPU tensor but the HIP '

@@ -99,19 +99,31 @@ def test_cov_conv_g(dtype) -> None:
     torch.testing.assert_close(out, expected, rtol=tol, atol=tol)
 
 
-@pytest.mark.parametrize('m,n', [(64, 65), (100, 200), (1000, 2049), (16, 16)])
+@pytest.mark.parametrize('m,n', [(6, 26), (64, 65), (100, 200), (1000, 2049), (16, 16)])
 def test_precond_eigen_fused(m, n) -> None:
-    from kfac_amd.ops import reference as ref
+    """Validate against an fp64 reference with realistic (orthogonal) bases.
 
+    Random dense qa/qg amplify fp32 cancellation ~1000x; real K-FAC bases
+    are orthogonal eigenvector matrices, so test with those.
+    """
     torch.manual_seed(3)
     grad = torch.randn(m, n, device='cuda')
-    qa = torch.randn(n, n, device='cuda')
-    qg = torch.randn(m, m, device='cuda')
+    sa = torch.randn(n, n, device='cuda')
+    qa = torch.linalg.eigh(sa + sa.t())[1].contiguous()
+    sg = torch.randn(m, m, device='cuda')
+    qg = torch.linalg.eigh(sg + sg.t())[1].contiguous()
     dgda = torch.rand(m, n, device='cuda') + 0.5
-    expected = ref.precond_eigen(grad, qa, qg, dgda=dgda)
+    # fp64 ground truth
+    e64 = (
+        qg.double()
+        @ ((qg.double().t() @ grad.double() @ qa.double()) * dgda.double())
+        @ qa.double().t()
+    )
     out = _ext().precond_eigen_fused(grad, qa, qg, dgda)
     torch.cuda.synchronize()
-    torch.testing.assert_close(out, expected, rtol=2e-4, atol=2e-4)
+    torch.testing.assert_close(
+        out.double(), e64, rtol=1e-3, atol=1e-4 * float(e64.abs().max()),
+    )
 
 
 def test_precond_eigen_dgda_bf16_grad() -> None:

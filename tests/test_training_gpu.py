@@ -49,12 +49,15 @@ def test_gpu_step_matches_cpu_reference() -> None:
     from kfac_amd import KFACPreconditioner
     from testing.models import LeNet
 
+    torch.manual_seed(123)
+    x0 = torch.randn(32, 1, 28, 28)
+    y0 = torch.randint(0, 10, (32,))
     results = {}
     for device in ('cpu', 'cuda'):
         torch.manual_seed(123)
         model = LeNet().to(device)
-        x = torch.randn(32, 1, 28, 28, device=device)
-        y = torch.randint(0, 10, (32,), device=device)
+        x = x0.to(device)
+        y = y0.to(device)
         precon = KFACPreconditioner(
             model, factor_update_steps=1, inv_update_steps=1, lr=0.01,
         )
