@@ -22,11 +22,20 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 namespace kfac {
 
 constexpr int BT = 64;   // C tile edge
-constexpr int BK = 32;   // m-slice staged per iteration
+constexpr int BK = 32;   // m-slice staged per iteration (f32 path)
 constexpr int LDS_STRIDE = BT + 1;  // break bank alignment for b32 reads
+
+// bf16 path: K=32-deep MFMA, 64-deep m-slices, [col][m] LDS layout so each
+// lane's 8-element fragment is one 16-byte ds_read_b128.
+constexpr int BKB = 64;
+constexpr int BSTR = BKB + 8;  // row stride 144 B: 16-lane groups hit 16
+                               // distinct banks ((144/4)*j mod 64, gcd 4)
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
 // ---------------------------------------------------------------- accessors
 
@@ -43,6 +52,12 @@ struct LinearAcc {
     if (m >= M || i >= Ncols) return 0.0f;
     if (i < K) return to_f32(a[(long)m * lda + i]);
     return 1.0f;  // fused bias-ones column
+  }
+
+  // 16 consecutive-m values at a fixed column (bf16-path staging).
+  __device__ __forceinline__ void load16(int m, int i, float* out) const {
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = load(m + e, i);
   }
 };
 
@@ -72,6 +87,50 @@ struct ConvPatchAcc {
     if (h < 0 || h >= H || w < 0 || w >= W) return 0.0f;
     return to_f32(x[((long)(n * C + c) * H + h) * W + w]);
   }
+
+  // 16 consecutive-m (= consecutive ow) values at a fixed patch column:
+  // one div/mod decomposition, then incremental ow/h walk. Consecutive m
+  // is stride-sw in w, so loads are HBM-contiguous for stride-1 convs.
+  __device__ __forceinline__ void load16(int m, int i, float* out) const {
+    if (i >= K || m >= M) {
+      // bias-ones column (i == K < Ncols) or out-of-range padding
+      float fill = (i >= K && i < Ncols && m < M) ? 1.0f : 0.0f;
+#pragma unroll
+      for (int e = 0; e < 16; ++e) out[e] = (m + e < M) ? fill : 0.0f;
+      return;
+    }
+    int s = i % kw;
+    int t2 = i / kw;
+    int r = t2 % kh;
+    int c = t2 / kh;
+    int ow = m % OW;
+    int t = m / OW;
+    int oh = t % OH;
+    int n = t / OH;
+    int h = oh * sh - ph + r;
+    int w = ow * sw - pw + s;
+    const T* base = x + ((long)(n * C + c) * H) * W;
+#pragma unroll
+    for (int e = 0; e < 16; ++e) {
+      bool ok = (m + e < M) && h >= 0 && h < H && w >= 0 && w < W;
+      out[e] = ok ? to_f32(base[(long)h * W + w]) : 0.0f;
+      // advance to the next output position
+      ++ow;
+      w += sw;
+      if (ow == OW) {
+        ow = 0;
+        w = -pw + s;
+        ++oh;
+        h += sh;
+        if (oh == OH) {
+          oh = 0;
+          h = -ph + r;
+          ++n;
+          base += (long)C * H * W;
+        }
+      }
+    }
+  }
 };
 
 template <typename T>
@@ -89,6 +148,29 @@ struct ConvGradAcc {
     int oh = t % OH;
     int n = t / OH;
     return to_f32(g[((long)(n * C + j) * OH + oh) * OW + ow]);
+  }
+
+  // 16 consecutive-m values of one channel: contiguous in (oh, ow) except
+  // at image boundaries -> one decomposition + pointer walk.
+  __device__ __forceinline__ void load16(int m, int j, float* out) const {
+    if (m >= M || j >= Ncols) {
+#pragma unroll
+      for (int e = 0; e < 16; ++e) out[e] = 0.0f;
+      return;
+    }
+    int sp = m % (OH * OW);  // position within the image plane
+    int n = m / (OH * OW);
+    const T* p = g + ((long)(n * C + j) * OH * OW) + sp;
+#pragma unroll
+    for (int e = 0; e < 16; ++e) {
+      out[e] = (m + e < M) ? to_f32(*p) : 0.0f;
+      ++sp;
+      ++p;
+      if (sp == OH * OW) {
+        sp = 0;
+        p += (long)(C - 1) * OH * OW;
+      }
+    }
   }
 };
 
@@ -192,6 +274,101 @@ __global__ __launch_bounds__(256) void syrk_kernel(
   }
 }
 
+// bf16 variant: inputs staged to LDS as bf16 in [col][m] layout so each
+// lane's 8-element MFMA fragment is one 16-byte ds_read_b128; compute on
+// mfma_f32_16x16x32_bf16 (fp32 accumulate), ~16x the f32-MFMA rate.
+template <typename Acc>
+__device__ __forceinline__ void stage_tile_bf16(
+    const Acc& acc,
+    __bf16 (*lds)[BSTR],
+    int m0,
+    int col0,
+    int tid) {
+  // 64 cols x 64 m-values, 256 threads: each thread stages one 16-deep
+  // m-run of one column (4 threads per column).
+  const int i = tid >> 2;
+  const int k0 = (tid & 3) * 16;
+  float vals[16];
+  acc.load16(m0 + k0, col0 + i, vals);
+#pragma unroll
+  for (int e = 0; e < 16; ++e) {
+    lds[i][k0 + e] = (__bf16)vals[e];
+  }
+}
+
+template <typename AccL, typename AccR>
+__global__ __launch_bounds__(256) void syrk_kernel_bf16(
+    float* __restrict__ out,
+    int N,
+    AccL accl,
+    AccR accr,
+    int m_per_split,
+    float coeff,
+    bool same_tile_ok) {
+  const int ti = blockIdx.x;
+  const int tj = blockIdx.y;
+  if (tj < ti) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+
+  const int i0 = ti * BT;
+  const int j0 = tj * BT;
+  const int m_begin = blockIdx.z * m_per_split;
+  const int m_end = min(accl.M, m_begin + m_per_split);
+  if (m_begin >= m_end) return;
+
+  __shared__ __bf16 lds_l[BT][BSTR];
+  __shared__ __bf16 lds_r[BT][BSTR];
+  const bool diag = same_tile_ok && (ti == tj);
+
+  f32x4 acc[2][2] = {};
+
+  for (int m0 = m_begin; m0 < m_end; m0 += BKB) {
+    stage_tile_bf16(accl, lds_l, m0, i0, tid);
+    if (!diag) {
+      stage_tile_bf16(accr, lds_r, m0, j0, tid);
+    }
+    __syncthreads();
+    auto rbuf = diag ? lds_l : lds_r;
+#pragma unroll
+    for (int kt = 0; kt < BKB; kt += 32) {
+      const int kfrag = kt + (lane >> 4) * 8;
+      bf16x8 a0 = *(const bf16x8*)&lds_l[wr * 32 + (lane & 15)][kfrag];
+      bf16x8 a1 = *(const bf16x8*)&lds_l[wr * 32 + 16 + (lane & 15)][kfrag];
+      bf16x8 b0 = *(const bf16x8*)&rbuf[wc * 32 + (lane & 15)][kfrag];
+      bf16x8 b1 = *(const bf16x8*)&rbuf[wc * 32 + 16 + (lane & 15)][kfrag];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const bool mirror = (ti != tj);
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i0 + wr * 32 + fi * 16 + (lane >> 4) * 4 + r;
+        int col = j0 + wc * 32 + fj * 16 + (lane & 15);
+        if (row < N && col < N) {
+          float v = coeff * acc[fi][fj][r];
+          atomicAdd(&out[(long)row * N + col], v);
+          if (mirror) {
+            atomicAdd(&out[(long)col * N + row], v);
+          }
+        }
+      }
+    }
+  }
+}
+
 __global__ void scale_kernel(float* out, long n, float beta) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) {
@@ -208,7 +385,7 @@ static int pick_splits(int M, int n_tiles) {
   return min(max_splits, want);
 }
 
-template <typename AccL>
+template <bool BF16, typename AccL>
 static hipError_t launch_syrk(
     hipStream_t stream,
     float* out,
@@ -223,12 +400,25 @@ static hipError_t launch_syrk(
   int nt = ceil_div(N, BT);
   int n_tiles = nt * (nt + 1) / 2;
   int splits = pick_splits(acc.M, n_tiles);
-  int m_per_split = ceil_div(ceil_div(acc.M, splits), BK) * BK;
+  constexpr int kslab = BF16 ? BKB : BK;
+  int m_per_split = ceil_div(ceil_div(acc.M, splits), kslab) * kslab;
   splits = ceil_div(acc.M, m_per_split);
   dim3 grid(nt, nt, splits);
-  syrk_kernel<AccL, AccL><<<grid, 256, 0, stream>>>(
-      out, N, acc, acc, m_per_split, coeff, true);
+  if constexpr (BF16) {
+    syrk_kernel_bf16<AccL, AccL><<<grid, 256, 0, stream>>>(
+        out, N, acc, acc, m_per_split, coeff, true);
+  } else {
+    syrk_kernel<AccL, AccL><<<grid, 256, 0, stream>>>(
+        out, N, acc, acc, m_per_split, coeff, true);
+  }
   return hipGetLastError();
+}
+
+// bf16 inputs take the bf16-MFMA kernel (no precision loss: the data is
+// already bf16); fp32/fp16 inputs take the exact-f32 MFMA kernel.
+template <typename T>
+constexpr bool use_bf16_mfma() {
+  return std::is_same<T, __hip_bfloat16>::value;
 }
 
 template <typename T>
@@ -243,7 +433,7 @@ hipError_t cov_linear_t(
     float beta,
     float coeff) {
   LinearAcc<T> acc{a, lda, M, K, K + (bias ? 1 : 0)};
-  return launch_syrk(stream, out, acc.Ncols, acc, beta, coeff);
+  return launch_syrk<use_bf16_mfma<T>()>(stream, out, acc.Ncols, acc, beta, coeff);
 }
 
 template <typename T>
@@ -270,7 +460,7 @@ hipError_t cov_conv_a_t(
   ConvPatchAcc<T> acc{x,  C,  H,  W,  OH, OW, kh,
                       kw, sh, sw, ph, pw, Nb * OH * OW,
                       K,  K + (bias ? 1 : 0)};
-  return launch_syrk(stream, out, acc.Ncols, acc, beta, coeff);
+  return launch_syrk<use_bf16_mfma<T>()>(stream, out, acc.Ncols, acc, beta, coeff);
 }
 
 template <typename T>
@@ -285,7 +475,7 @@ hipError_t cov_conv_g_t(
     float beta,
     float coeff) {
   ConvGradAcc<T> acc{g, C, OH, OW, Nb * OH * OW, C};
-  return launch_syrk(stream, out, acc.Ncols, acc, beta, coeff);
+  return launch_syrk<use_bf16_mfma<T>()>(stream, out, acc.Ncols, acc, beta, coeff);
 }
 
 // Explicit instantiations used by the binding.
