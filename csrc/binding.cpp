@@ -22,6 +22,7 @@ hipError_t cov_conv_a_t(hipStream_t, const T*, int, int, int, int, int, int, int
 template <typename T>
 hipError_t cov_conv_g_t(hipStream_t, const T*, int, int, int, int, float*, float, float);
 hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float);
+hipError_t precond_grouped_f32(hipStream_t, const void*, int, int);
 template <typename T>
 hipError_t kl_clip_accum_t(hipStream_t, float*, const T*, const T*, long);
 hipError_t triu_pack_f32(hipStream_t, float*, const float*, int);
@@ -282,6 +283,89 @@ torch::Tensor precond_eigen(
   return eigen_tail(v2, qa, qg, stream).to(dtype);
 }
 
+// Mirrors kfac::PrecondDesc in gemm.hip (10 x 8-byte fields).
+struct PrecondDescHost {
+  int64_t m, n;
+  const float* grad;
+  const float* qa;
+  const float* qg;
+  const float* dgda;
+  float* s1;
+  float* s2;
+  float* out;
+  int64_t tile_off;
+};
+static_assert(sizeof(PrecondDescHost) == 80, "descriptor layout");
+
+std::vector<torch::Tensor> precond_eigen_grouped(
+    std::vector<torch::Tensor> grads,
+    std::vector<torch::Tensor> qas,
+    std::vector<torch::Tensor> qgs,
+    std::vector<torch::Tensor> dgdas) {
+  const int L = (int)grads.size();
+  TORCH_CHECK(L > 0, "empty layer list");
+  TORCH_CHECK(
+      (int)qas.size() == L && (int)qgs.size() == L && (int)dgdas.size() == L,
+      "list length mismatch");
+  auto dev_opts =
+      torch::TensorOptions().device(grads[0].device()).dtype(torch::kFloat32);
+
+  std::vector<torch::Tensor> g32(L);
+  std::vector<torch::ScalarType> dtypes(L);
+  int64_t total = 0;
+  int64_t tiles = 0;
+  std::vector<int64_t> offsets(L);
+  std::vector<int64_t> tile_offs(L);
+  for (int l = 0; l < L; ++l) {
+    check_gpu_contig(qas[l], "qa");
+    check_gpu_contig(qgs[l], "qg");
+    check_gpu_contig(dgdas[l], "dgda");
+    TORCH_CHECK(grads[l].dim() == 2, "grad must be 2D");
+    dtypes[l] = grads[l].scalar_type();
+    g32[l] = grads[l].to(torch::kFloat32).contiguous();
+    int64_t m = g32[l].size(0);
+    int64_t n = g32[l].size(1);
+    TORCH_CHECK(qgs[l].size(0) == m && qas[l].size(0) == n, "shape mismatch");
+    offsets[l] = total;
+    tile_offs[l] = tiles;
+    total += m * n;
+    tiles += (int64_t)((m + 63) / 64) * ((n + 63) / 64);
+  }
+  auto s1 = torch::empty({total}, dev_opts);
+  auto s2 = torch::empty({total}, dev_opts);
+  auto outbuf = torch::empty({total}, dev_opts);
+
+  auto desc_cpu = torch::empty(
+      {L * (int64_t)(sizeof(PrecondDescHost) / 8)},
+      torch::TensorOptions().dtype(torch::kInt64).pinned_memory(true));
+  auto* d = (PrecondDescHost*)desc_cpu.data_ptr<int64_t>();
+  for (int l = 0; l < L; ++l) {
+    d[l].m = g32[l].size(0);
+    d[l].n = g32[l].size(1);
+    d[l].grad = g32[l].data_ptr<float>();
+    d[l].qa = qas[l].data_ptr<float>();
+    d[l].qg = qgs[l].data_ptr<float>();
+    d[l].dgda = dgdas[l].data_ptr<float>();
+    d[l].s1 = s1.data_ptr<float>() + offsets[l];
+    d[l].s2 = s2.data_ptr<float>() + offsets[l];
+    d[l].out = outbuf.data_ptr<float>() + offsets[l];
+    d[l].tile_off = tile_offs[l];
+  }
+  auto desc_dev = desc_cpu.to(grads[0].device(), /*non_blocking=*/true);
+  auto stream = current_stream(grads[0]);
+  CHECK_OK(kfac::precond_grouped_f32(
+      stream, desc_dev.data_ptr<int64_t>(), L, (int)tiles));
+
+  std::vector<torch::Tensor> outs(L);
+  for (int l = 0; l < L; ++l) {
+    auto view = outbuf
+                    .narrow(0, offsets[l], d[l].m * d[l].n)
+                    .view({d[l].m, d[l].n});
+    outs[l] = (dtypes[l] == torch::kFloat32) ? view : view.to(dtypes[l]);
+  }
+  return outs;
+}
+
 torch::Tensor precond_inverse(
     torch::Tensor grad,
     torch::Tensor a_inv,
@@ -377,6 +461,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("precond_eigen_fused", &precond_eigen_fused, "Kronecker precondition (prediv)");
   m.def("precond_eigen", &precond_eigen, "Kronecker precondition (dg/da)");
   m.def("precond_inverse", &precond_inverse, "G^-1 grad A^-1");
+  m.def(
+      "precond_eigen_grouped",
+      &precond_eigen_grouped,
+      "whole precondition chain for all layers in 4 launches");
   m.def("kl_clip_accum", &kl_clip_accum, "device-side kl-clip accumulation");
   m.def("triu_pack", &triu_pack, "pack upper triangle");
   m.def("triu_unpack", &triu_unpack, "unpack upper triangle");

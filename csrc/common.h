@@ -21,7 +21,9 @@ namespace kfac {
 
 constexpr int kWave = 64;  // CDNA wavefront width
 
-inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
+__host__ __device__ inline int ceil_div(int a, int b) {
+  return (a + b - 1) / b;
+}
 
 // fp32x4 accumulator for mfma_f32_16x16x4_f32 (4 AGPRs per lane).
 typedef __attribute__((ext_vector_type(4))) float f32x4;

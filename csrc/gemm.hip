@@ -49,30 +49,31 @@ __device__ __forceinline__ void stage_gemm(
   }
 }
 
+// One 64x64 output tile of C = op(A) @ op(B) (+ epilogue). Shared between
+// the standalone GEMM kernel and the grouped (all-layers-one-launch)
+// precondition kernel.
 template <Epilogue EPI>
-__global__ __launch_bounds__(256) void gemm_kernel(
+__device__ __forceinline__ void gemm_tile_body(
     float* __restrict__ c,
     const float* __restrict__ a,
     const float* __restrict__ b,
     int M,
     int N,
     int K,
-    bool ta,  // use A^T (A physical is [K_phys x M_phys] interpreted so that
-              // op(A)[m][k]; ta=false: A[m][k] (ld=K); ta=true: A[k][m] (ld=M)
+    bool ta,  // op(A)[m][k]; ta=false: A[m][k] (ld=K); ta=true: A[k][m] (ld=M)
     bool tb,  // op(B)[k][n]; tb=false: B[k][n] (ld=N); tb=true: B[n][k] (ld=K)
     const float* __restrict__ e1,  // MUL: dgda [M x N]; DIV_OUTER: dg [M]
     const float* __restrict__ e2,  // DIV_OUTER: da [N]
-    float damping) {
+    float damping,
+    int i0,
+    int j0,
+    float (*lds_a)[GLDS],
+    float (*lds_b)[GLDS]) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wr = wave >> 1;
   const int wc = wave & 1;
-  const int i0 = blockIdx.x * GBT;  // row block of C
-  const int j0 = blockIdx.y * GBT;  // col block of C
-
-  __shared__ float lds_a[GBK][GLDS];
-  __shared__ float lds_b[GBK][GLDS];
 
   f32x4 acc[2][2] = {};
 
@@ -121,6 +122,99 @@ __global__ __launch_bounds__(256) void gemm_kernel(
       }
     }
   }
+}
+
+template <Epilogue EPI>
+__global__ __launch_bounds__(256) void gemm_kernel(
+    float* __restrict__ c,
+    const float* __restrict__ a,
+    const float* __restrict__ b,
+    int M,
+    int N,
+    int K,
+    bool ta,
+    bool tb,
+    const float* __restrict__ e1,
+    const float* __restrict__ e2,
+    float damping) {
+  __shared__ float lds_a[GBK][GLDS];
+  __shared__ float lds_b[GBK][GLDS];
+  gemm_tile_body<EPI>(
+      c, a, b, M, N, K, ta, tb, e1, e2, damping, (int)blockIdx.x * GBT,
+      (int)blockIdx.y * GBT, lds_a, lds_b);
+}
+
+// ---------------------------------------------------- grouped precondition
+//
+// The whole eigen precondition chain for EVERY layer in 4 kernel launches
+// (the reference launches ~8 torch ops per layer, eigen.py:374-385; the
+// per-layer v1 loop is launch-bound on 50+ layer models). Stages:
+//   1: s1  = QG^T @ grad
+//   2: s2  = (s1 @ QA) * dgda
+//   3: s1  = QG @ s2
+//   4: out = s1 @ QA^T
+// Every stage's output is (m, n) so one tile table serves all stages.
+
+struct PrecondDesc {
+  long m, n;
+  const float* grad;
+  const float* qa;
+  const float* qg;
+  const float* dgda;
+  float* s1;
+  float* s2;
+  float* out;
+  long tile_off;  // first global tile index of this layer
+};
+
+template <int STAGE>
+__global__ __launch_bounds__(256) void grouped_precond_kernel(
+    const PrecondDesc* __restrict__ desc,
+    int n_layers) {
+  const int tile = blockIdx.x;
+  int l = 0;
+  while (l + 1 < n_layers && tile >= (int)desc[l + 1].tile_off) ++l;
+  const PrecondDesc d = desc[l];
+  const int m = (int)d.m;
+  const int n = (int)d.n;
+  const int local = tile - (int)d.tile_off;
+  const int ntj = ceil_div(n, GBT);
+  const int i0 = (local / ntj) * GBT;
+  const int j0 = (local % ntj) * GBT;
+
+  __shared__ float lds_a[GBK][GLDS];
+  __shared__ float lds_b[GBK][GLDS];
+
+  if constexpr (STAGE == 1) {
+    gemm_tile_body<Epilogue::NONE>(
+        d.s1, d.qg, d.grad, m, n, m, true, false, nullptr, nullptr, 0.f, i0,
+        j0, lds_a, lds_b);
+  } else if constexpr (STAGE == 2) {
+    gemm_tile_body<Epilogue::MUL>(
+        d.s2, d.s1, d.qa, m, n, n, false, false, d.dgda, nullptr, 0.f, i0,
+        j0, lds_a, lds_b);
+  } else if constexpr (STAGE == 3) {
+    gemm_tile_body<Epilogue::NONE>(
+        d.s1, d.qg, d.s2, m, n, m, false, false, nullptr, nullptr, 0.f, i0,
+        j0, lds_a, lds_b);
+  } else {
+    gemm_tile_body<Epilogue::NONE>(
+        d.out, d.s1, d.qa, m, n, n, false, true, nullptr, nullptr, 0.f, i0,
+        j0, lds_a, lds_b);
+  }
+}
+
+hipError_t precond_grouped_f32(
+    hipStream_t stream,
+    const void* desc_dev,
+    int n_layers,
+    int total_tiles) {
+  auto desc = (const PrecondDesc*)desc_dev;
+  grouped_precond_kernel<1><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  grouped_precond_kernel<2><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  grouped_precond_kernel<3><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  grouped_precond_kernel<4><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  return hipGetLastError();
 }
 
 hipError_t gemm_f32(

@@ -276,31 +276,23 @@ class BaseKFACPreconditioner:
         self._tdc.flush_allreduce_buckets()
 
         if self.steps % self.inv_update_steps == 0:
-            for name, layer in reversed(list(self._layers.values())):
-                if get_rank() == self._assignment.inv_worker(name, 'A'):
-                    layer.compute_a_inv(damping=self.damping)
-                if (
-                    self._assignment.broadcast_inverses()
-                    and self._assignment.is_grad_worker(name)
-                ):
-                    layer.broadcast_a_inv(
-                        src=self._assignment.inv_worker(name, 'A'),
-                        group=self._assignment.grad_worker_group(name),
-                    )
-                if get_rank() == self._assignment.inv_worker(name, 'G'):
-                    layer.compute_g_inv(damping=self.damping)
-                if (
-                    self._assignment.broadcast_inverses()
-                    and self._assignment.is_grad_worker(name)
-                ):
-                    layer.broadcast_g_inv(
-                        src=self._assignment.inv_worker(name, 'G'),
-                        group=self._assignment.grad_worker_group(name),
-                    )
+            self._compute_local_inverses()
+            if self._assignment.broadcast_inverses():
+                for name, layer in reversed(list(self._layers.values())):
+                    if self._assignment.is_grad_worker(name):
+                        layer.broadcast_a_inv(
+                            src=self._assignment.inv_worker(name, 'A'),
+                            group=self._assignment.grad_worker_group(name),
+                        )
+                        layer.broadcast_g_inv(
+                            src=self._assignment.inv_worker(name, 'G'),
+                            group=self._assignment.grad_worker_group(name),
+                        )
             self._tdc.flush_allreduce_buckets()
 
+        grouped = self._grouped_precondition()
         for name, layer in reversed(list(self._layers.values())):
-            if self._assignment.is_grad_worker(name):
+            if not grouped and self._assignment.is_grad_worker(name):
                 layer.preconditioned_grad(damping=self.damping)
             if self._assignment.broadcast_gradients():
                 layer.broadcast_grad(
@@ -316,6 +308,135 @@ class BaseKFACPreconditioner:
 
         self._steps += 1
         self._mini_steps = defaultdict(int)
+
+    def _grouped_precondition(self) -> bool:
+        """Fast path: the precondition chain for every local layer in 4
+        grouped kernel launches (vs the reference's ~8 torch launches per
+        layer, eigen.py:374-385). Returns False if any layer is not
+        eligible (CPU, non-eigen, non-prediv, missing extension) — the
+        caller then runs the per-layer path.
+        """
+        from kfac_amd import ops
+        from kfac_amd.layers.eigen import KFACEigenLayer
+
+        if not ops.extension_available():
+            return False
+        work = []
+        for name, layer in reversed(list(self._layers.values())):
+            if not self._assignment.is_grad_worker(name):
+                continue
+            if (
+                not isinstance(layer, KFACEigenLayer)
+                or not layer.prediv_eigenvalues
+            ):
+                return False
+            qa, qg, dgda = layer.qa, layer.qg, layer.dgda
+            if qa is None or qg is None or dgda is None:
+                return False
+            if not (qa.is_cuda and qa.dtype == torch.float32):
+                return False
+            work.append((layer, layer.module.get_grad(), qa, qg, dgda))
+        if not work:
+            return True
+        outs = ops.precond_eigen_grouped(
+            [w[1] for w in work],
+            [w[2] for w in work],
+            [w[3] for w in work],
+            [w[4] for w in work],
+        )
+        for (layer, *_), out in zip(work, outs):
+            layer.grad = out
+        return True
+
+    def _compute_local_inverses(self) -> None:
+        """Compute second-order state for the layers assigned to this rank.
+
+        MI355X redesign of the reference's per-layer loop
+        (base_preconditioner.py:340-362): same-size factors assigned to
+        this rank are STACKED and eigendecomposed in one batched rocSOLVER
+        call — measured 3.1x faster than the per-layer loop for the
+        ResNet-50 factor-size distribution (profiles/eigh_strategies.md).
+        Falls back to per-layer compute for non-eigen or non-symmetric
+        layers.
+        """
+        from kfac_amd.layers.eigen import KFACEigenLayer
+
+        rank = get_rank()
+        eigen_a: list[KFACEigenLayer] = []
+        eigen_g: list[KFACEigenLayer] = []
+        other: list[tuple[str, KFACBaseLayer]] = []
+        for name, layer in reversed(list(self._layers.values())):
+            a_mine = rank == self._assignment.inv_worker(name, 'A')
+            g_mine = rank == self._assignment.inv_worker(name, 'G')
+            if not (a_mine or g_mine):
+                continue
+            if isinstance(layer, KFACEigenLayer) and layer.symmetric_factors:
+                if a_mine:
+                    eigen_a.append(layer)
+                if g_mine:
+                    eigen_g.append(layer)
+            else:
+                other.append((name, layer))
+
+        damping = self.damping
+        self._batched_eigh(eigen_a, 'a')
+        self._batched_eigh(eigen_g, 'g')
+        # prediv fusion: dgda on the G worker (requires colocated factors)
+        for layer in eigen_g:
+            if layer.prediv_eigenvalues:
+                da = layer.da
+                dg = layer.dg
+                assert da is not None and dg is not None
+                layer.dgda = 1 / (torch.outer(dg, da) + damping)
+                layer.dg = None
+                layer.da = None
+        for name, layer in other:
+            if rank == self._assignment.inv_worker(name, 'A'):
+                layer.compute_a_inv(damping=damping)
+            if rank == self._assignment.inv_worker(name, 'G'):
+                layer.compute_g_inv(damping=damping)
+
+    @staticmethod
+    def _batched_eigh(layers: list[Any], which: str) -> None:
+        """Group same-size factors, eigendecompose each group in one call."""
+        from collections import defaultdict
+
+        groups: dict[tuple, list[Any]] = defaultdict(list)
+        for layer in layers:
+            factor = layer.a_factor if which == 'a' else layer.g_factor
+            if not isinstance(factor, torch.Tensor):
+                raise RuntimeError(
+                    f'Cannot eigendecompose {which.upper()} before it has '
+                    'been computed',
+                )
+            groups[(factor.shape[0], factor.device, factor.dtype)].append(layer)
+        for (n, _dev, _dt), group in groups.items():
+            if len(group) == 1:
+                layer = group[0]
+                if which == 'a':
+                    layer.compute_a_inv()
+                else:
+                    layer.compute_g_inv_no_prediv()
+                continue
+            stack = torch.stack(
+                [
+                    (layer.a_factor if which == 'a' else layer.g_factor).to(
+                        torch.float32,
+                    )
+                    for layer in group
+                ],
+            )
+            d, q = torch.linalg.eigh(stack)
+            d = torch.clamp(d, min=0.0)
+            for i, layer in enumerate(group):
+                qv = q[i].to(layer.inv_dtype).contiguous()
+                dv = d[i].to(layer.inv_dtype).contiguous()
+                if which == 'a':
+                    layer.qa = qv
+                    layer.da = dv
+                else:
+                    layer.qg = qv
+                    layer.dg = dv
 
     def reset_batch(self) -> None:
         """Drop accumulated factor contributions from the current batch."""

@@ -175,8 +175,9 @@ class KFACEigenLayer(KFACBaseLayer):
         self.qa = qa.to(self.inv_dtype)
         self.da = torch.clamp(da.to(self.inv_dtype), min=0.0)
 
-    def compute_g_inv(self, damping: float = 0.001) -> None:
-        """Eigendecompose G; optionally fuse dGdA (reference eigen.py:323-348)."""
+    def compute_g_inv_no_prediv(self) -> None:
+        """Eigendecompose G without the dGdA fusion (batched-inverse path:
+        the preconditioner applies prediv after both factors are done)."""
         g = self.g_factor
         if not isinstance(g, torch.Tensor):
             raise RuntimeError('Cannot eigendecompose G before G has been computed')
@@ -187,6 +188,10 @@ class KFACEigenLayer(KFACBaseLayer):
             dg, qg = dgc.real, qgc.real
         self.qg = qg.to(self.inv_dtype)
         self.dg = torch.clamp(dg.to(self.inv_dtype), min=0.0)
+
+    def compute_g_inv(self, damping: float = 0.001) -> None:
+        """Eigendecompose G; optionally fuse dGdA (reference eigen.py:323-348)."""
+        self.compute_g_inv_no_prediv()
         if self.prediv_eigenvalues:
             da = self.da
             assert da is not None

@@ -183,6 +183,22 @@ def precond_eigen(
     )
 
 
+def precond_eigen_grouped(
+    grads: list[torch.Tensor],
+    qas: list[torch.Tensor],
+    qgs: list[torch.Tensor],
+    dgdas: list[torch.Tensor],
+) -> list[torch.Tensor]:
+    """Whole eigen precondition chain for all layers in 4 kernel launches.
+
+    GPU-only (requires the HIP extension); the per-layer path is the
+    fallback for CPU or non-prediv configurations.
+    """
+    ext = _require_ext('precond_eigen_grouped')
+    assert ext is not None
+    return ext.precond_eigen_grouped(grads, qas, qgs, dgdas)
+
+
 def precond_inverse(
     grad: torch.Tensor,
     a_inv: torch.Tensor,

@@ -199,3 +199,22 @@ def test_ops_dispatch_routes_to_ext() -> None:
     ref.cov_linear(a, bias=True, out=expected, beta=0.0, coeff=1.0 / 128)
     torch.cuda.synchronize()
     torch.testing.assert_close(out, expected, rtol=1e-5, atol=1e-5)
+
+
+def test_precond_eigen_grouped_matches_per_layer() -> None:
+    """Grouped 4-launch chain == per-layer fused chain for mixed shapes."""
+    torch.manual_seed(11)
+    shapes = [(64, 147), (256, 2304), (1000, 2049), (512, 513), (64, 64)]
+    grads, qas, qgs, dgdas = [], [], [], []
+    for m, n in shapes:
+        grads.append(torch.randn(m, n, device='cuda'))
+        sa = torch.randn(n, n, device='cuda')
+        qas.append(torch.linalg.eigh(sa + sa.t())[1].contiguous())
+        sg = torch.randn(m, m, device='cuda')
+        qgs.append(torch.linalg.eigh(sg + sg.t())[1].contiguous())
+        dgdas.append(torch.rand(m, n, device='cuda') + 0.5)
+    outs = _ext().precond_eigen_grouped(grads, qas, qgs, dgdas)
+    for i in range(len(shapes)):
+        expected = _ext().precond_eigen_fused(grads[i], qas[i], qgs[i], dgdas[i])
+        torch.cuda.synchronize()
+        torch.testing.assert_close(outs[i], expected, rtol=1e-5, atol=1e-5)
