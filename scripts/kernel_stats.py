@@ -26,18 +26,26 @@ def main(pattern: str, out_path: str | None = None) -> None:
         lines.append(f'tables: {tables}')
     else:
         t = cand[0]
-        cols = [r[1] for r in db.execute(f'PRAGMA table_info({t})')]
-        lines.append(f'table {t}: {cols}')
-        # find the string/name join
+
+        def tbl(prefix: str) -> str:
+            match = [x for x in tables if x.startswith(prefix)]
+            return match[0] if match else prefix
+
+        ksym = tbl('rocpd_info_kernel_symbol')
+        sstr = tbl('rocpd_string')
         try:
+            kcols = [r[1] for r in db.execute(f'PRAGMA table_info({ksym})')]
+            name_col = (
+                'display_name' if 'display_name' in kcols else 'kernel_name'
+            )
             q = f"""
             SELECT s.string AS name,
                    COUNT(*) AS calls,
                    SUM(k.end - k.start) / 1e6 AS total_ms,
                    AVG(k.end - k.start) / 1e3 AS avg_us
             FROM {t} k
-            JOIN rocpd_info_kernel_symbol ks ON k.kernel_id = ks.id
-            JOIN rocpd_string s ON ks.display_name = s.id
+            JOIN {ksym} ks ON k.kernel_id = ks.id
+            JOIN {sstr} s ON ks.{name_col} = s.id
             GROUP BY s.string ORDER BY total_ms DESC LIMIT 30
             """
             for row in db.execute(q):
