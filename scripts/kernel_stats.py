@@ -9,56 +9,76 @@ import sys
 
 def main(pattern: str, out_path: str | None = None) -> None:
     paths = sorted(glob.glob(pattern))
+    lines: list[str] = []
     if not paths:
-        print(f'no db matching {pattern}')
+        lines.append(f'no db matching {pattern}')
+        _emit(lines, out_path)
         return
     db = sqlite3.connect(paths[-1])
-    # rocpd schema: kernel dispatches with start/end timestamps.
     tables = [
         r[0]
         for r in db.execute(
             "SELECT name FROM sqlite_master WHERE type='table'",
         )
     ]
-    lines = []
-    cand = [t for t in tables if 'kernel_dispatch' in t]
-    if not cand:
+
+    def tbl(prefix: str) -> str | None:
+        match = [x for x in tables if x.startswith(prefix)]
+        return match[0] if match else None
+
+    kd = tbl('rocpd_kernel_dispatch')
+    ksym = tbl('rocpd_info_kernel_symbol')
+    sstr = tbl('rocpd_string')
+    if kd is None:
         lines.append(f'tables: {tables}')
-    else:
-        t = cand[0]
+        _emit(lines, out_path)
+        return
+    n = db.execute(f'SELECT COUNT(*) FROM {kd}').fetchone()[0]
+    lines.append(f'dispatches: {n}')
+    if ksym is not None:
+        kcols = [r[1] for r in db.execute(f'PRAGMA table_info({ksym})')]
+        lines.append(f'{ksym} cols: {kcols}')
+        sample = db.execute(f'SELECT * FROM {ksym} LIMIT 2').fetchall()
+        lines.append(f'sample: {str(sample)[:300]}')
+    done = False
+    if ksym is not None and sstr is not None:
+        kcols = [r[1] for r in db.execute(f'PRAGMA table_info({ksym})')]
+        for name_col in ('display_name', 'kernel_name', 'formatted_kernel_name'):
+            if name_col not in kcols:
+                continue
+            try:
+                q = f"""
+                SELECT s.string, COUNT(*), SUM(k.end - k.start)/1e6,
+                       AVG(k.end - k.start)/1e3
+                FROM {kd} k
+                JOIN {ksym} ks ON k.kernel_id = ks.id
+                JOIN {sstr} s ON ks.{name_col} = s.id
+                GROUP BY s.string ORDER BY 3 DESC LIMIT 30
+                """
+                rows = db.execute(q).fetchall()
+                if rows:
+                    lines.append(f'--- join via {name_col} ---')
+                    for r in rows:
+                        lines.append(
+                            f'{r[2]:10.2f} ms {r[1]:6d} calls '
+                            f'{r[3]:9.1f} us/call  {r[0][:90]}',
+                        )
+                    done = True
+                    break
+            except Exception as e:
+                lines.append(f'join {name_col} failed: {e}')
+    if not done:
+        # fall back: group by kernel_id only
+        q = f"""
+        SELECT k.kernel_id, COUNT(*), SUM(k.end - k.start)/1e6
+        FROM {kd} k GROUP BY k.kernel_id ORDER BY 3 DESC LIMIT 20
+        """
+        for r in db.execute(q):
+            lines.append(f'{r[2]:10.2f} ms {r[1]:6d} calls  kernel_id={r[0]}')
+    _emit(lines, out_path)
 
-        def tbl(prefix: str) -> str:
-            match = [x for x in tables if x.startswith(prefix)]
-            return match[0] if match else prefix
 
-        ksym = tbl('rocpd_info_kernel_symbol')
-        sstr = tbl('rocpd_string')
-        try:
-            kcols = [r[1] for r in db.execute(f'PRAGMA table_info({ksym})')]
-            name_col = (
-                'display_name' if 'display_name' in kcols else 'kernel_name'
-            )
-            q = f"""
-            SELECT s.string AS name,
-                   COUNT(*) AS calls,
-                   SUM(k.end - k.start) / 1e6 AS total_ms,
-                   AVG(k.end - k.start) / 1e3 AS avg_us
-            FROM {t} k
-            JOIN {ksym} ks ON k.kernel_id = ks.id
-            JOIN {sstr} s ON ks.{name_col} = s.id
-            GROUP BY s.string ORDER BY total_ms DESC LIMIT 30
-            """
-            for row in db.execute(q):
-                name = row[0][:80]
-                lines.append(
-                    f'{row[2]:10.2f} ms  {row[1]:6d} calls  {row[3]:9.1f} us/call  {name}',
-                )
-        except Exception as e:
-            lines.append(f'join failed: {e}')
-            # dump schema of related tables
-            for tt in tables:
-                cols = [r[1] for r in db.execute(f'PRAGMA table_info({tt})')]
-                lines.append(f'{tt}: {cols}')
+def _emit(lines: list[str], out_path: str | None) -> None:
     text = '\n'.join(lines)
     print(text)
     if out_path:
