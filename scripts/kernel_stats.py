@@ -41,32 +41,25 @@ def main(pattern: str, out_path: str | None = None) -> None:
         sample = db.execute(f'SELECT * FROM {ksym} LIMIT 2').fetchall()
         lines.append(f'sample: {str(sample)[:300]}')
     done = False
-    if ksym is not None and sstr is not None:
-        kcols = [r[1] for r in db.execute(f'PRAGMA table_info({ksym})')]
-        for name_col in ('display_name', 'kernel_name', 'formatted_kernel_name'):
-            if name_col not in kcols:
-                continue
-            try:
-                q = f"""
-                SELECT s.string, COUNT(*), SUM(k.end - k.start)/1e6,
-                       AVG(k.end - k.start)/1e3
-                FROM {kd} k
-                JOIN {ksym} ks ON k.kernel_id = ks.id
-                JOIN {sstr} s ON ks.{name_col} = s.id
-                GROUP BY s.string ORDER BY 3 DESC LIMIT 30
-                """
-                rows = db.execute(q).fetchall()
-                if rows:
-                    lines.append(f'--- join via {name_col} ---')
-                    for r in rows:
-                        lines.append(
-                            f'{r[2]:10.2f} ms {r[1]:6d} calls '
-                            f'{r[3]:9.1f} us/call  {r[0][:90]}',
-                        )
-                    done = True
-                    break
-            except Exception as e:
-                lines.append(f'join {name_col} failed: {e}')
+    if ksym is not None:
+        try:
+            q = f"""
+            SELECT ks.display_name, COUNT(*), SUM(k.end - k.start)/1e6,
+                   AVG(k.end - k.start)/1e3
+            FROM {kd} k
+            JOIN {ksym} ks ON k.kernel_id = ks.id
+            GROUP BY ks.display_name ORDER BY 3 DESC LIMIT 30
+            """
+            rows = db.execute(q).fetchall()
+            if rows:
+                for r in rows:
+                    lines.append(
+                        f'{r[2]:10.2f} ms {r[1]:6d} calls '
+                        f'{r[3]:9.1f} us/call  {str(r[0])[:100]}',
+                    )
+                done = True
+        except Exception as e:
+            lines.append(f'join failed: {e}')
     if not done:
         # fall back: group by kernel_id only
         q = f"""
