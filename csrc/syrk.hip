@@ -59,6 +59,25 @@ struct LinearAcc {
 #pragma unroll
     for (int e = 0; e < 16; ++e) out[e] = load(m + e, i);
   }
+
+  __device__ __forceinline__ void load16_bf16(int m, int i, __bf16* out) const {
+    float tmp[16];
+    load16(m, i, tmp);
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = (__bf16)tmp[e];
+  }
+
+  // 16 consecutive COLUMNS of one row: contiguous 32 B of HBM.
+  __device__ __forceinline__ void load16cols_bf16(int m, int i0, __bf16* out) const {
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      if (m < M && i0 + 15 < K) {
+        __builtin_memcpy(out, a + (long)m * lda + i0, 16 * sizeof(__bf16));
+        return;
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = (__bf16)load(m, i0 + e);
+  }
 };
 
 template <typename T>
@@ -131,6 +150,36 @@ struct ConvPatchAcc {
       }
     }
   }
+
+  // bf16 staging fast path: for unit-stride interior runs the 16 values
+  // are 32 contiguous bytes of HBM -> two (possibly unaligned) 16-byte
+  // loads instead of 16 scalar gathers.
+  __device__ __forceinline__ void load16_bf16(int m, int i, __bf16* out) const {
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      if (i < K && m + 15 < M) {
+        int s = i % kw;
+        int t2 = i / kw;
+        int r = t2 % kh;
+        int c = t2 / kh;
+        int ow = m % OW;
+        int t = m / OW;
+        int oh = t % OH;
+        int h = oh * sh - ph + r;
+        int w = ow * sw - pw + s;
+        if (sw == 1 && ow + 15 < OW && h >= 0 && h < H && w >= 0 &&
+            w + 15 < W) {
+          int n = t / OH;
+          const T* p = x + ((long)(n * C + c) * H + h) * W + w;
+          __builtin_memcpy(out, p, 16 * sizeof(__bf16));
+          return;
+        }
+      }
+    }
+    float tmp[16];
+    load16(m, i, tmp);
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = (__bf16)tmp[e];
+  }
 };
 
 template <typename T>
@@ -171,6 +220,24 @@ struct ConvGradAcc {
         p += (long)(C - 1) * OH * OW;
       }
     }
+  }
+
+  __device__ __forceinline__ void load16_bf16(int m, int j, __bf16* out) const {
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      if (j < Ncols && m + 15 < M) {
+        int sp = m % (OH * OW);
+        if (sp + 15 < OH * OW) {
+          int n = m / (OH * OW);
+          const T* p = g + ((long)(n * C + j) * OH * OW) + sp;
+          __builtin_memcpy(out, p, 16 * sizeof(__bf16));
+          return;
+        }
+      }
+    }
+    float tmp[16];
+    load16(m, j, tmp);
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = (__bf16)tmp[e];
   }
 };
 
@@ -277,6 +344,25 @@ __global__ __launch_bounds__(256) void syrk_kernel(
 // bf16 variant: inputs staged to LDS as bf16 in [col][m] layout so each
 // lane's 8-element MFMA fragment is one 16-byte ds_read_b128; compute on
 // mfma_f32_16x16x32_bf16 (fp32 accumulate), ~16x the f32-MFMA rate.
+// Column-fast staging (LinearAcc): each thread reads 16 contiguous
+// columns of one m-row from HBM and scatters them down one LDS column.
+template <typename Acc>
+__device__ __forceinline__ void stage_tile_bf16_colfast(
+    const Acc& acc,
+    __bf16 (*lds)[BSTR],
+    int m0,
+    int col0,
+    int tid) {
+  const int k = tid >> 2;
+  const int i0 = (tid & 3) * 16;
+  __bf16 vals[16];
+  acc.load16cols_bf16(m0 + k, col0 + i0, vals);
+#pragma unroll
+  for (int e = 0; e < 16; ++e) {
+    lds[i0 + e][k] = vals[e];
+  }
+}
+
 template <typename Acc>
 __device__ __forceinline__ void stage_tile_bf16(
     const Acc& acc,
@@ -285,15 +371,14 @@ __device__ __forceinline__ void stage_tile_bf16(
     int col0,
     int tid) {
   // 64 cols x 64 m-values, 256 threads: each thread stages one 16-deep
-  // m-run of one column (4 threads per column).
+  // m-run of one column (4 threads per column). The 16 bf16 land as two
+  // 16-byte LDS writes (row stride 144 B keeps them aligned).
   const int i = tid >> 2;
   const int k0 = (tid & 3) * 16;
-  float vals[16];
-  acc.load16(m0 + k0, col0 + i, vals);
-#pragma unroll
-  for (int e = 0; e < 16; ++e) {
-    lds[i][k0 + e] = (__bf16)vals[e];
-  }
+  __bf16 vals[16];
+  acc.load16_bf16(m0 + k0, col0 + i, vals);
+  *(bf16x8*)&lds[i][k0] = *(const bf16x8*)&vals[0];
+  *(bf16x8*)&lds[i][k0 + 8] = *(const bf16x8*)&vals[8];
 }
 
 template <typename AccL, typename AccR>
@@ -327,9 +412,12 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16(
   f32x4 acc[2][2] = {};
 
   for (int m0 = m_begin; m0 < m_end; m0 += BKB) {
-    stage_tile_bf16(accl, lds_l, m0, i0, tid);
-    if (!diag) {
-      stage_tile_bf16(accr, lds_r, m0, j0, tid);
+    if constexpr (AccL::kLaneAlongCols) {
+      stage_tile_bf16_colfast(accl, lds_l, m0, i0, tid);
+      if (!diag) stage_tile_bf16_colfast(accr, lds_r, m0, j0, tid);
+    } else {
+      stage_tile_bf16(accl, lds_l, m0, i0, tid);
+      if (!diag) stage_tile_bf16(accr, lds_r, m0, j0, tid);
     }
     __syncthreads();
     auto rbuf = diag ? lds_l : lds_r;
@@ -379,9 +467,10 @@ __global__ void scale_kernel(float* out, long n, float beta) {
 // ---------------------------------------------------------------- launchers
 
 static int pick_splits(int M, int n_tiles) {
-  // Fill the chip: want >= ~1024 workgroups; each split handles >= 1 BK.
+  // Fill the chip (256 CUs, 1-2 blocks of 4 waves each) without
+  // multiplying the fp32-atomic epilogue traffic more than needed.
   int max_splits = max(1, M / (8 * BK));
-  int want = max(1, 1024 / max(1, n_tiles));
+  int want = max(1, 512 / max(1, n_tiles));
   return min(max_splits, want);
 }
 
