@@ -329,7 +329,8 @@ std::vector<torch::Tensor> precond_eigen_grouped(
     offsets[l] = total;
     tile_offs[l] = tiles;
     total += m * n;
-    tiles += (int64_t)((m + 63) / 64) * ((n + 63) / 64);
+    // must match GBT=128 in gemm.hip
+    tiles += (int64_t)((m + 127) / 128) * ((n + 127) / 128);
   }
   auto s1 = torch::empty({total}, dev_opts);
   auto s2 = torch::empty({total}, dev_opts);

@@ -15,7 +15,7 @@
 
 namespace kfac {
 
-constexpr int GBT = 64;
+constexpr int GBT = 128;
 constexpr int GBK = 32;
 constexpr int GLDS = GBT + 1;
 
@@ -49,9 +49,11 @@ __device__ __forceinline__ void stage_gemm(
   }
 }
 
-// One 64x64 output tile of C = op(A) @ op(B) (+ epilogue). Shared between
-// the standalone GEMM kernel and the grouped (all-layers-one-launch)
-// precondition kernel.
+// One 128x128 output tile of C = op(A) @ op(B) (+ epilogue).
+// 4 waves, each owning a 64x64 quadrant as 4x4 16x16 MFMA fragments
+// (64 f32 accumulators/lane); per 32-deep K slice each wave issues 128
+// mfma_f32_16x16x4_f32 against 16 LDS reads per substep — the MFMA:LDS
+// ratio that the 64x64 structure lacked (guide §5 step-2 ladder).
 template <Epilogue EPI>
 __device__ __forceinline__ void gemm_tile_body(
     float* __restrict__ c,
@@ -75,41 +77,42 @@ __device__ __forceinline__ void gemm_tile_body(
   const int wr = wave >> 1;
   const int wc = wave & 1;
 
-  f32x4 acc[2][2] = {};
+  f32x4 acc[4][4] = {};
 
   for (int k0 = 0; k0 < K; k0 += GBK) {
-    // op(A) slab: lds_a[k][i] = op(A)[i0+i][k0+k]
-    //   ta=false: A[m][k], physical ld = K -> element A[(i0+i)*K + (k0+k)]
-    //   ta=true : A[k][m], physical ld = M -> element A[(k0+k)*M + (i0+i)]
     stage_gemm(a, M, K, ta ? (long)M : (long)K, !ta, k0, i0, lds_a, tid);
-    // op(B) slab: lds_b[k][j] = op(B)[k0+k][j0+j]
-    //   tb=false: B[k][n], ld = N -> B[(k0+k)*N + (j0+j)]
-    //   tb=true : B[n][k], ld = K -> B[(j0+j)*K + (k0+k)]
     stage_gemm(b, N, K, tb ? (long)K : (long)N, tb, k0, j0, lds_b, tid);
     __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < GBK; kk += 4) {
       const int krow = kk + (lane >> 4);
-      float a0 = lds_a[krow][wr * 32 + (lane & 15)];
-      float a1 = lds_a[krow][wr * 32 + 16 + (lane & 15)];
-      float b0 = lds_b[krow][wc * 32 + (lane & 15)];
-      float b1 = lds_b[krow][wc * 32 + 16 + (lane & 15)];
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+      float av[4];
+      float bv[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        av[f] = lds_a[krow][wr * 64 + f * 16 + (lane & 15)];
+        bv[f] = lds_b[krow][wc * 64 + f * 16 + (lane & 15)];
+      }
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj) {
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              av[fi], bv[fj], acc[fi][fj], 0, 0, 0);
+        }
+      }
     }
     __syncthreads();
   }
 
 #pragma unroll
-  for (int fi = 0; fi < 2; ++fi) {
+  for (int fi = 0; fi < 4; ++fi) {
 #pragma unroll
-    for (int fj = 0; fj < 2; ++fj) {
+    for (int fj = 0; fj < 4; ++fj) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = i0 + wr * 32 + fi * 16 + (lane >> 4) * 4 + r;
-        int col = j0 + wc * 32 + fj * 16 + (lane & 15);
+        int row = i0 + wr * 64 + fi * 16 + (lane >> 4) * 4 + r;
+        int col = j0 + wc * 64 + fj * 16 + (lane & 15);
         if (row < M && col < N) {
           float v = acc[fi][fj][r];
           if constexpr (EPI == Epilogue::MUL) {

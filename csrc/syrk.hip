@@ -457,6 +457,133 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16(
   }
 }
 
+
+// 128x128-tile bf16 SYRK for large factors (N > 128): 4 waves each own a
+// 64x64 quadrant as 4x4 fragments (64 fp32 accumulators/lane) -> 32
+// MFMAs per 64-deep m-slice per wave against 4 load16 staging calls.
+constexpr int BTB = 128;
+
+template <typename Acc>
+__device__ __forceinline__ void stage_tile_bf16_big(
+    const Acc& acc,
+    __bf16 (*lds)[BSTR],
+    int m0,
+    int col0,
+    int tid) {
+  // 128 cols x 64 m, 256 threads: each thread stages two 16-deep m-runs
+  // of one column.
+  const int i = tid >> 1;
+  const int k0 = (tid & 1) * 32;
+  __bf16 vals[16];
+  acc.load16_bf16(m0 + k0, col0 + i, vals);
+  *(bf16x8*)&lds[i][k0] = *(const bf16x8*)&vals[0];
+  *(bf16x8*)&lds[i][k0 + 8] = *(const bf16x8*)&vals[8];
+  acc.load16_bf16(m0 + k0 + 16, col0 + i, vals);
+  *(bf16x8*)&lds[i][k0 + 16] = *(const bf16x8*)&vals[0];
+  *(bf16x8*)&lds[i][k0 + 24] = *(const bf16x8*)&vals[8];
+}
+
+template <typename Acc>
+__device__ __forceinline__ void stage_tile_bf16_big_colfast(
+    const Acc& acc,
+    __bf16 (*lds)[BSTR],
+    int m0,
+    int col0,
+    int tid) {
+  const int k = tid >> 2;
+  const int i0 = (tid & 3) * 32;
+  __bf16 vals[16];
+  acc.load16cols_bf16(m0 + k, col0 + i0, vals);
+#pragma unroll
+  for (int e = 0; e < 16; ++e) lds[i0 + e][k] = vals[e];
+  acc.load16cols_bf16(m0 + k, col0 + i0 + 16, vals);
+#pragma unroll
+  for (int e = 0; e < 16; ++e) lds[i0 + 16 + e][k] = vals[e];
+}
+
+template <typename AccL, typename AccR>
+__global__ __launch_bounds__(256) void syrk_kernel_bf16_big(
+    float* __restrict__ out,
+    int N,
+    AccL accl,
+    AccR accr,
+    int m_per_split,
+    float coeff,
+    bool same_tile_ok) {
+  const int ti = blockIdx.x;
+  const int tj = blockIdx.y;
+  if (tj < ti) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+
+  const int i0 = ti * BTB;
+  const int j0 = tj * BTB;
+  const int m_begin = blockIdx.z * m_per_split;
+  const int m_end = min(accl.M, m_begin + m_per_split);
+  if (m_begin >= m_end) return;
+
+  __shared__ __bf16 lds_l[BTB][BSTR];
+  __shared__ __bf16 lds_r[BTB][BSTR];
+  const bool diag = same_tile_ok && (ti == tj);
+
+  f32x4 acc[4][4] = {};
+
+  for (int m0 = m_begin; m0 < m_end; m0 += BKB) {
+    if constexpr (AccL::kLaneAlongCols) {
+      stage_tile_bf16_big_colfast(accl, lds_l, m0, i0, tid);
+      if (!diag) stage_tile_bf16_big_colfast(accr, lds_r, m0, j0, tid);
+    } else {
+      stage_tile_bf16_big(accl, lds_l, m0, i0, tid);
+      if (!diag) stage_tile_bf16_big(accr, lds_r, m0, j0, tid);
+    }
+    __syncthreads();
+    auto rbuf = diag ? lds_l : lds_r;
+#pragma unroll
+    for (int kt = 0; kt < BKB; kt += 32) {
+      const int kfrag = kt + (lane >> 4) * 8;
+      bf16x8 av[4];
+      bf16x8 bv[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        av[f] = *(const bf16x8*)&lds_l[wr * 64 + f * 16 + (lane & 15)][kfrag];
+        bv[f] = *(const bf16x8*)&rbuf[wc * 64 + f * 16 + (lane & 15)][kfrag];
+      }
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj) {
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av[fi], bv[fj], acc[fi][fj], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  const bool mirror = (ti != tj);
+#pragma unroll
+  for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 4; ++fj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i0 + wr * 64 + fi * 16 + (lane >> 4) * 4 + r;
+        int col = j0 + wc * 64 + fj * 16 + (lane & 15);
+        if (row < N && col < N) {
+          float v = coeff * acc[fi][fj][r];
+          atomicAdd(&out[(long)row * N + col], v);
+          if (mirror) {
+            atomicAdd(&out[(long)col * N + row], v);
+          }
+        }
+      }
+    }
+  }
+}
+
 __global__ void scale_kernel(float* out, long n, float beta) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) {
@@ -486,16 +613,23 @@ static hipError_t launch_syrk(
   int threads = 256;
   scale_kernel<<<(n2 + threads - 1) / threads, threads, 0, stream>>>(
       out, n2, beta);
-  int nt = ceil_div(N, BT);
+  constexpr int kslab = BF16 ? BKB : BK;
+  const bool big = BF16 && N > 128;
+  int bt = big ? BTB : BT;
+  int nt = ceil_div(N, bt);
   int n_tiles = nt * (nt + 1) / 2;
   int splits = pick_splits(acc.M, n_tiles);
-  constexpr int kslab = BF16 ? BKB : BK;
   int m_per_split = ceil_div(ceil_div(acc.M, splits), kslab) * kslab;
   splits = ceil_div(acc.M, m_per_split);
   dim3 grid(nt, nt, splits);
   if constexpr (BF16) {
-    syrk_kernel_bf16<AccL, AccL><<<grid, 256, 0, stream>>>(
-        out, N, acc, acc, m_per_split, coeff, true);
+    if (big) {
+      syrk_kernel_bf16_big<AccL, AccL><<<grid, 256, 0, stream>>>(
+          out, N, acc, acc, m_per_split, coeff, true);
+    } else {
+      syrk_kernel_bf16<AccL, AccL><<<grid, 256, 0, stream>>>(
+          out, N, acc, acc, m_per_split, coeff, true);
+    }
   } else {
     syrk_kernel<AccL, AccL><<<grid, 256, 0, stream>>>(
         out, N, acc, acc, m_per_split, coeff, true);
