@@ -17,15 +17,17 @@ namespace kfac {
 
 constexpr int GBT = 128;
 constexpr int GBK = 32;
-constexpr int GLDS = GBT + 1;
+constexpr int GLDS = GBT + 4;  // +4 keeps float4 LDS rows 16B-aligned
 
 enum class Epilogue : int { NONE = 0, MUL = 1, DIV_OUTER = 2 };
 
-// Stage op(A)'s [k, rows i0..i0+63] slab into lds[k][i].
-// src is f32; trans selects A[k][i] vs A[i][k].
+// Stage op(A)'s [k, rows i0..i0+GBT) slab into lds[k][i].
+// The thread->element mapping follows the PHYSICAL memory order so loads
+// coalesce either way: each thread moves 4 consecutive floats of the
+// source (one 16-byte load in the interior).
 __device__ __forceinline__ void stage_gemm(
     const float* __restrict__ src,
-    int rows,      // logical rows of op(src) slab dimension (i extent)
+    int rows,      // logical i extent of op(src)
     int ks,        // logical k extent
     long ld,       // leading dim of the PHYSICAL matrix
     bool trans,    // false: physical[k][i] = src[k*ld+i] feeds lds[k][i]
@@ -35,17 +37,42 @@ __device__ __forceinline__ void stage_gemm(
     float (*lds)[GLDS],
     int tid) {
 #pragma unroll
-  for (int e = 0; e < (GBK * GBT) / 256; ++e) {
-    int idx = tid + e * 256;
-    int k = idx / GBT;
-    int i = idx % GBT;
-    int gk = k0 + k;
-    int gi = i0 + i;
-    float v = 0.0f;
-    if (gk < ks && gi < rows) {
-      v = trans ? src[(long)gi * ld + gk] : src[(long)gk * ld + gi];
+  for (int e = 0; e < (GBK * GBT) / (256 * 4); ++e) {
+    const int c = tid + e * 256;
+    if (trans) {
+      // contiguous along k in memory: 8 threads cover one 32-k row
+      const int k = (c & 7) * 4;
+      const int i = c >> 3;
+      const int gi = i0 + i;
+      const long base = (long)gi * ld + (k0 + k);
+      float v[4];
+      if (gi < rows && k0 + k + 3 < ks) {
+        __builtin_memcpy(v, src + base, 16);
+      } else {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          v[q] = (gi < rows && k0 + k + q < ks) ? src[base + q] : 0.0f;
+        }
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) lds[k + q][i] = v[q];
+    } else {
+      // contiguous along i in memory: 32 threads cover one 128-i row
+      const int i = (c & 31) * 4;
+      const int k = c >> 5;
+      const int gk = k0 + k;
+      const long base = (long)gk * ld + (i0 + i);
+      float v[4];
+      if (gk < ks && i0 + i + 3 < rows) {
+        __builtin_memcpy(v, src + base, 16);
+      } else {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          v[q] = (gk < ks && i0 + i + q < rows) ? src[base + q] : 0.0f;
+        }
+      }
+      *(float4*)&lds[k][i] = make_float4(v[0], v[1], v[2], v[3]);
     }
-    lds[k][i] = v;
   }
 }
 
