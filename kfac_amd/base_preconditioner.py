@@ -261,6 +261,19 @@ class BaseKFACPreconditioner:
         Call after loss.backward() (grads already averaged by DDP) and
         before optimizer.step(). Reference base_preconditioner.py:310-382.
         """
+        # Join the covariance side stream: everything the hooks queued
+        # (fused cov kernels overlapping backward) is ordered before the
+        # rest of the step; input refs can then be released.
+        layers = list(self._layers.values())
+        if layers:
+            device = layers[0][1].module.device
+            if device.type == 'cuda':
+                from kfac_amd.streams import join_cov_stream
+
+                join_cov_stream(device)
+            for _, layer in layers:
+                layer.clear_pending()
+
         if (
             not self._update_factors_in_hook
             and self.steps % self.factor_update_steps == 0

@@ -83,6 +83,10 @@ class KFACBaseLayer:
 
         self.eps = 1e-10
         self.symmetric_factors = self.module.has_symmetric_factors()
+        # Tensors whose side-stream covariance kernels may still be in
+        # flight; held so autograd cannot free them early. Cleared by the
+        # preconditioner after joining the side stream.
+        self._pending_inputs: list[torch.Tensor] = []
 
         # Persistent accumulators for the current batch (fp32).
         self._a_batch: torch.Tensor | None = None
@@ -177,7 +181,16 @@ class KFACBaseLayer:
                 shape, dtype=torch.float32, device=a.device,
             )
         beta = 0.0 if self._a_count == 0 else 1.0
-        self.module.accumulate_a_factor(a, self._a_batch, beta, 1.0)
+        if a.is_cuda:
+            from kfac_amd.streams import cov_stream
+
+            s = cov_stream(a.device)
+            s.wait_stream(torch.cuda.current_stream(a.device))
+            with torch.cuda.stream(s):
+                self.module.accumulate_a_factor(a, self._a_batch, beta, 1.0)
+            self._pending_inputs.append(a)
+        else:
+            self.module.accumulate_a_factor(a, self._a_batch, beta, 1.0)
         self._a_count += 1
 
     def save_layer_grad_output(self, grad_output: tuple[torch.Tensor, ...]) -> None:
@@ -194,11 +207,43 @@ class KFACBaseLayer:
             )
         coeff = 1.0
         if self.grad_scaler is not None:
-            s = float(self.grad_scaler())
-            coeff = 1.0 / (s * s)
+            sc = float(self.grad_scaler())
+            coeff = 1.0 / (sc * sc)
         beta = 0.0 if self._g_count == 0 else 1.0
-        self.module.accumulate_g_factor(g, self._g_batch, beta, coeff)
+        if g.is_cuda:
+            from kfac_amd.streams import cov_stream
+
+            s = cov_stream(g.device)
+            s.wait_stream(torch.cuda.current_stream(g.device))
+            with torch.cuda.stream(s):
+                self.module.accumulate_g_factor(g, self._g_batch, beta, coeff)
+            self._pending_inputs.append(g)
+        else:
+            self.module.accumulate_g_factor(g, self._g_batch, beta, coeff)
         self._g_count += 1
+
+    def _factor_ctx(self) -> Any:
+        """Stream context for factor EMA/reduce ops.
+
+        While side-stream covariance kernels are in flight (hook path),
+        the EMA and the allreduce launch must be ordered after them on
+        the same side stream; once the preconditioner has joined the
+        stream (pending list cleared), plain current-stream execution is
+        correct.
+        """
+        if self._pending_inputs and self._pending_inputs[0].is_cuda:
+            from kfac_amd.streams import cov_stream
+
+            return torch.cuda.stream(
+                cov_stream(self._pending_inputs[0].device),
+            )
+        import contextlib
+
+        return contextlib.nullcontext()
+
+    def clear_pending(self) -> None:
+        """Release input refs after the cov stream has been joined."""
+        self._pending_inputs.clear()
 
     def reset_batch(self) -> None:
         """Drop accumulated batch contributions (buffers are kept)."""
@@ -214,14 +259,15 @@ class KFACBaseLayer:
         if self._a_count == 0 or self._a_batch is None:
             return
         w = (1.0 - alpha) / self._a_count
-        if self._a_factor is None:
-            f = self._a_batch.clone().mul_(w)
-            f.diagonal().add_(alpha)
-            self.a_factor = f.to(self.factor_dtype)
-        else:
-            a = self.a_factor
-            assert a is not None
-            a.mul_(alpha).add_(self._a_batch.to(a.dtype), alpha=w)
+        with self._factor_ctx():
+            if self._a_factor is None:
+                f = self._a_batch.clone().mul_(w)
+                f.diagonal().add_(alpha)
+                self.a_factor = f.to(self.factor_dtype)
+            else:
+                a = self.a_factor
+                assert a is not None
+                a.mul_(alpha).add_(self._a_batch.to(a.dtype), alpha=w)
         self._a_count = 0
 
     def update_g_factor(self, alpha: float = 0.95) -> None:
@@ -229,14 +275,15 @@ class KFACBaseLayer:
         if self._g_count == 0 or self._g_batch is None:
             return
         w = (1.0 - alpha) / self._g_count
-        if self._g_factor is None:
-            f = self._g_batch.clone().mul_(w)
-            f.diagonal().add_(alpha)
-            self.g_factor = f.to(self.factor_dtype)
-        else:
-            g = self.g_factor
-            assert g is not None
-            g.mul_(alpha).add_(self._g_batch.to(g.dtype), alpha=w)
+        with self._factor_ctx():
+            if self._g_factor is None:
+                f = self._g_batch.clone().mul_(w)
+                f.diagonal().add_(alpha)
+                self.g_factor = f.to(self.factor_dtype)
+            else:
+                g = self.g_factor
+                assert g is not None
+                g.mul_(alpha).add_(self._g_batch.to(g.dtype), alpha=w)
         self._g_count = 0
 
     # -- communication -----------------------------------------------------
@@ -252,23 +299,25 @@ class KFACBaseLayer:
         """Launch async allreduce-average of A over ``group``."""
         if self.a_factor is None:
             raise RuntimeError('a_factor is None, cannot reduce')
-        self.a_factor = self._allreduce_fn()(
-            self.a_factor,
-            average=True,
-            symmetric=self.symmetric_factors and self.symmetry_aware,
-            group=group,
-        )
+        with self._factor_ctx():
+            self.a_factor = self._allreduce_fn()(
+                self.a_factor,
+                average=True,
+                symmetric=self.symmetric_factors and self.symmetry_aware,
+                group=group,
+            )
 
     def reduce_g_factor(self, group: dist.ProcessGroup | None = None) -> None:
         """Launch async allreduce-average of G over ``group``."""
         if self.g_factor is None:
             raise RuntimeError('g_factor is None, cannot reduce')
-        self.g_factor = self._allreduce_fn()(
-            self.g_factor,
-            average=True,
-            symmetric=self.symmetric_factors and self.symmetry_aware,
-            group=group,
-        )
+        with self._factor_ctx():
+            self.g_factor = self._allreduce_fn()(
+                self.g_factor,
+                average=True,
+                symmetric=self.symmetric_factors and self.symmetry_aware,
+                group=group,
+            )
 
     def broadcast_grad(
         self,
