@@ -341,6 +341,7 @@ class BaseKFACPreconditioner:
             if (
                 not isinstance(layer, KFACEigenLayer)
                 or not layer.prediv_eigenvalues
+                or not getattr(layer, 'grouped_precondition', True)
             ):
                 return False
             qa, qg, dgda = layer.qa, layer.qg, layer.dgda

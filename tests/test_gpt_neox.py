@@ -1,0 +1,236 @@
+"""Tests for the tensor/pipeline-parallel (GPT-NeoX) subsystem."""
+
+from __future__ import annotations
+
+import sys
+import tempfile
+
+import pytest
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, '.')
+
+from kfac_amd.gpt_neox.topology import PipeModelDataTopology  # noqa: E402
+from kfac_amd.gpt_neox.assignment import GPTNeoXAssignment  # noqa: E402
+from kfac_amd.gpt_neox.mpu import get_group_with_rank  # noqa: E402
+from kfac_amd.gpt_neox.mpu import split_tensor_along_dim  # noqa: E402
+from testing.distributed import run_distributed  # noqa: E402
+
+
+@pytest.mark.parametrize('pp,mp,dp', [(1, 1, 4), (2, 1, 2), (1, 2, 2), (2, 2, 2)])
+def test_topology_properties(pp: int, mp: int, dp: int) -> None:
+    topo = PipeModelDataTopology(num_pp=pp, num_mp=mp, num_dp=dp)
+    world = pp * mp * dp
+    assert topo.world_size() == world
+    # coords roundtrip and are unique
+    coords = {topo.get_coord(r) for r in range(world)}
+    assert len(coords) == world
+    for r in range(world):
+        c = topo.get_coord(r)
+        assert topo.get_rank(c.pipe, c.data, c.model) == r
+    # axis groups partition the world
+    for axis, size in (('data', dp), ('model', mp), ('pipe', pp)):
+        groups = topo.get_axis_comm_lists(axis)
+        assert all(len(g) == size for g in groups)
+        flat = sorted(r for g in groups for r in g)
+        assert flat == list(range(world))
+
+
+def test_split_tensor() -> None:
+    t = torch.arange(12.0).reshape(2, 6)
+    parts = split_tensor_along_dim(t, 3, dim=-1, contiguous_split_chunks=True)
+    assert len(parts) == 3
+    assert all(p.shape == (2, 2) for p in parts)
+    torch.testing.assert_close(torch.cat(parts, dim=-1), t)
+    with pytest.raises(ValueError):
+        split_tensor_along_dim(t, 5, dim=-1)
+
+
+def test_get_group_with_rank() -> None:
+    groups = [[0, 1], [2, 3]]
+    assert get_group_with_rank(2, groups) == [2, 3]
+    with pytest.raises(ValueError):
+        get_group_with_rank(9, groups)
+
+
+def test_assignment_dp_only() -> None:
+    """pp=1, mp=1: everything reduces to MEM-OPT data parallelism."""
+    topo = PipeModelDataTopology(num_pp=1, num_mp=1, num_dp=4)
+    work = {f'l{i}': {'A': float(i + 1), 'G': float(i + 1)} for i in range(6)}
+    for rank in range(4):
+        asn = GPTNeoXAssignment(
+            work,
+            local_rank=rank,
+            topology=topo,
+            data_parallel_group=None,
+            model_parallel_group=None,
+        )
+        assert asn.broadcast_gradients()
+        assert not asn.broadcast_inverses()
+        for layer in asn.get_layers():
+            inv = asn.inv_worker(layer, 'A')
+            assert inv == asn.inv_worker(layer, 'G')
+            assert 0 <= inv < 4
+            # mp=1: the primary (factor gatherer) for this rank is itself;
+            # src grad worker is the inv worker (it is a dp peer).
+            assert asn.factor_worker(layer, 'A') == rank
+            assert asn.src_grad_worker(layer) == inv
+            assert asn.is_grad_worker(layer) == (rank == inv)
+
+
+def test_assignment_balances_across_pipe_peers() -> None:
+    topo = PipeModelDataTopology(num_pp=2, num_mp=1, num_dp=2)
+    # stage-0 ranks: 0,1 ; stage-1 ranks: 2,3
+    work = {f'l{i}': {'A': 1.0, 'G': 1.0} for i in range(4)}
+    asn = GPTNeoXAssignment(
+        work,
+        local_rank=0,
+        topology=topo,
+        data_parallel_group=None,
+        model_parallel_group=None,
+    )
+    workers = [asn.inv_worker(layer, 'A') for layer in asn.get_layers()]
+    # only stage-0 peers get stage-0 layers, evenly
+    assert sorted(workers) == [0, 0, 1, 1]
+
+
+def _mlp_training(tmpdir: str) -> None:
+    from kfac_amd.gpt_neox import GPTNeoXKFACPreconditioner
+    from kfac_amd.gpt_neox.topology import PipeModelDataTopology
+    from testing.gpt_neox import ParallelMLP
+
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    topo = PipeModelDataTopology(num_pp=1, num_mp=1, num_dp=world)
+    dp_group = dist.new_group(list(range(world)))
+    torch.manual_seed(0)
+    model = ParallelMLP()
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    precon = GPTNeoXKFACPreconditioner(
+        model,
+        topology=topo,
+        data_parallel_group=dp_group,
+        model_parallel_group=None,
+        factor_update_steps=1,
+        inv_update_steps=1,
+        lr=0.05,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    x = torch.randn(16, 10)
+    y = torch.randint(0, 4, (16,))
+    losses = []
+    for _ in range(10):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        for p in model.parameters():
+            dist.all_reduce(p.grad)
+            p.grad /= world
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[0] > losses[-1], losses
+
+    # sharded checkpoint: gather-based
+    sd = precon.state_dict()
+    assert set(sd['layers'].keys()) == {'dense_h_to_4h', 'dense_4h_to_h'}
+    precon.load_state_dict(sd, compute_inverses=True)
+
+    # factor-dir based
+    precon.factor_checkpoint_dir = f'{tmpdir}/factors'
+    sd2 = precon.state_dict()
+    assert 'layers' not in sd2
+    precon.load_factors_from_dir(compute_inverses=False)
+
+
+def test_gpt_neox_dp_training() -> None:
+    with tempfile.TemporaryDirectory() as td:
+        run_distributed(2, _mlp_training, td)
+
+
+def _tp_layer_protocol() -> None:
+    """mp=2: gather -> precondition on primary -> scatter equals the
+    single-process preconditioning of the full gradient."""
+    from kfac_amd.distributed import TorchDistributedCommunicator
+    from kfac_amd.gpt_neox.layer import GPTNeoXKFACEigenLayer
+    from kfac_amd.gpt_neox.modules import GPTNeoXLinearModuleHelper
+    from kfac_amd.ops import reference as ref
+    from testing.gpt_neox import RowParallelLinear
+
+    rank = dist.get_rank()
+    mp_group = dist.new_group([0, 1])
+    # dp size 1: every rank must call new_group for every rank set
+    dp_groups = [dist.new_group([r]) for r in range(2)]
+    dp_group = dp_groups[rank]
+    torch.manual_seed(5)
+
+    in_dim, out_dim, batch = 8, 6, 32
+    # full data, same on both ranks
+    x = torch.randn(batch, in_dim)
+    full_grad = torch.randn(out_dim, in_dim)
+    bias_grad = torch.randn(out_dim)
+
+    shard = in_dim // 2
+    module = RowParallelLinear(shard, out_dim, bias=True)
+    module.weight.grad = full_grad[:, rank * shard : (rank + 1) * shard].clone()
+    module.bias.grad = bias_grad.clone()
+
+    tdc = TorchDistributedCommunicator()
+    layer = GPTNeoXKFACEigenLayer(
+        GPTNeoXLinearModuleHelper(module, mp_group, parallelism='input'),
+        parallelism='input',
+        model_parallel_group=mp_group,
+        data_parallel_group=dp_group,
+        pipe_parallel_peer_group=mp_group,
+        primary_rank=0,
+        tdc=tdc,
+        prediv_eigenvalues=False,
+    )
+
+    # factor shapes account for the gathered input dim
+    assert layer.module.a_factor_shape == (in_dim + 1, in_dim + 1)
+    assert layer.module.g_factor_shape == (out_dim, out_dim)
+
+    # save input: each rank contributes its shard
+    layer.save_layer_input([x[:, rank * shard : (rank + 1) * shard]])
+    g_out = torch.randn(batch, out_dim)
+    layer.save_layer_grad_output((g_out,))
+    layer.update_a_factor(0.95)
+    layer.update_g_factor(0.95)
+    layer.reduce_a_factor()
+    layer.reduce_g_factor()
+    tdc.flush_allreduce_buckets()  # all ranks: launch pending buckets
+    if rank == 0:
+        layer.compute_a_inv(damping=1e-3)
+        layer.compute_g_inv(damping=1e-3)
+    layer.preconditioned_grad(damping=1e-3)
+
+    # single-process reference on the full gradient
+    if rank == 0:
+        grad = torch.cat([full_grad, bias_grad.view(-1, 1)], 1)
+        expected = ref.precond_eigen(
+            grad, layer.qa, layer.qg, da=layer.da, dg=layer.dg, damping=1e-3,
+        )
+    # each rank's shard of the result must match
+    result = layer.grad
+    assert result.shape == (out_dim, shard + 1)
+    if rank == 0:
+        shard_expected = torch.cat(
+            [expected[:, :shard], expected[:, -1:]], 1,
+        )
+        torch.testing.assert_close(result, shard_expected, rtol=1e-4, atol=1e-5)
+        # send expected to rank 1 for its shard check
+        dist.broadcast(expected, src=0, group=mp_group)
+    else:
+        expected = torch.empty(out_dim, in_dim + 1)
+        dist.broadcast(expected, src=0, group=mp_group)
+        shard_expected = torch.cat(
+            [expected[:, shard : 2 * shard], expected[:, -1:]], 1,
+        )
+        torch.testing.assert_close(result, shard_expected, rtol=1e-4, atol=1e-5)
+
+
+def test_tp_layer_gather_precondition_scatter() -> None:
+    run_distributed(2, _tp_layer_protocol)
