@@ -1,0 +1,293 @@
+"""Unit tests for aux subsystems: scheduler, hyperparams, tracing,
+registration, module helpers, preconditioner validation + checkpointing.
+
+Coverage model: reference tests/scheduler_test.py, hyperparams_test.py,
+tracing_test.py, layers/register_test.py, layers/modules_test.py,
+base_preconditioner_test.py.
+"""
+
+from __future__ import annotations
+
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, '.')
+
+from kfac_amd import KFACPreconditioner  # noqa: E402
+from kfac_amd.distributed import TorchDistributedCommunicator  # noqa: E402
+from kfac_amd.enums import AssignmentStrategy  # noqa: E402
+from kfac_amd.hyperparams import exp_decay_factor_averaging  # noqa: E402
+from kfac_amd.layers.eigen import KFACEigenLayer  # noqa: E402
+from kfac_amd.layers.modules import Conv2dModuleHelper  # noqa: E402
+from kfac_amd.layers.modules import LinearModuleHelper  # noqa: E402
+from kfac_amd.layers.register import any_match  # noqa: E402
+from kfac_amd.layers.register import get_flattened_modules  # noqa: E402
+from kfac_amd.layers.register import register_modules  # noqa: E402
+from kfac_amd.scheduler import LambdaParamScheduler  # noqa: E402
+from kfac_amd import tracing  # noqa: E402
+from testing.models import LeNet  # noqa: E402
+from testing.models import TinyModel  # noqa: E402
+
+
+# ---------------------------------------------------------------- hyperparams
+
+def test_exp_decay_factor_averaging() -> None:
+    fn = exp_decay_factor_averaging()
+    assert fn(0) == fn(1) == 0.0
+    assert fn(2) == 0.5
+    assert fn(100) == pytest.approx(0.95)  # capped at min_value
+    with pytest.raises(ValueError):
+        fn(-1)
+    with pytest.raises(ValueError):
+        exp_decay_factor_averaging(min_value=0)
+
+
+# ---------------------------------------------------------------- scheduler
+
+def test_lambda_param_scheduler() -> None:
+    model = TinyModel()
+    p = KFACPreconditioner(model, damping=0.01, lr=0.1)
+    sched = LambdaParamScheduler(
+        p,
+        damping_lambda=lambda step: 0.5,
+        lr_lambda=lambda step: 2.0,
+        factor_update_steps_lambda=lambda step: 3.0,
+    )
+    sched.step()
+    assert p.damping == pytest.approx(0.005)
+    assert p.lr == pytest.approx(0.2)
+    assert p.factor_update_steps == 3
+
+
+def test_scheduler_rejects_callable_params() -> None:
+    model = TinyModel()
+    p = KFACPreconditioner(model, damping=lambda s: 0.01)
+    with pytest.raises(ValueError):
+        LambdaParamScheduler(p, damping_lambda=lambda s: 0.5)
+
+
+def test_callable_hyperparams_resolve_with_step() -> None:
+    model = TinyModel()
+    p = KFACPreconditioner(
+        model,
+        damping=lambda s: 0.1 / (s + 1),
+        factor_update_steps=lambda s: 2,
+    )
+    assert p.damping == pytest.approx(0.1)
+    p._steps = 9
+    assert p.damping == pytest.approx(0.01)
+    assert p.factor_update_steps == 2
+
+
+# ---------------------------------------------------------------- tracing
+
+def test_trace_decorator() -> None:
+    tracing.clear_trace()
+
+    @tracing.trace()
+    def f(x: int) -> int:
+        return x * 2
+
+    assert f(3) == 6
+    assert f(4) == 8
+    t = tracing.get_trace()
+    (name,) = t.keys()
+    assert 'f' in name
+    assert t[name] >= 0
+    total = tracing.get_trace(average=False)[name]
+    assert total >= t[name]
+    tracing.log_trace()
+    tracing.clear_trace()
+    assert tracing.get_trace() == {}
+
+
+# ---------------------------------------------------------------- register
+
+def test_register_modules_counts() -> None:
+    tdc = TorchDistributedCommunicator()
+    layers = register_modules(LeNet(), KFACEigenLayer, skip_layers=[], tdc=tdc)
+    # 2 conv + 3 linear
+    assert len(layers) == 5
+
+
+def test_register_skip_by_name_and_class() -> None:
+    tdc = TorchDistributedCommunicator()
+    layers = register_modules(
+        LeNet(), KFACEigenLayer, skip_layers=['conv.*'], tdc=tdc,
+    )
+    assert len(layers) == 3
+    layers = register_modules(
+        LeNet(), KFACEigenLayer, skip_layers=['Conv2d'], tdc=tdc,
+    )
+    assert len(layers) == 3
+    layers = register_modules(
+        LeNet(), KFACEigenLayer, skip_layers=['.*'], tdc=tdc,
+    )
+    assert len(layers) == 0
+
+
+def test_register_skips_frozen() -> None:
+    model = LeNet()
+    model.fc1.weight.requires_grad_(False)
+    tdc = TorchDistributedCommunicator()
+    layers = register_modules(model, KFACEigenLayer, skip_layers=[], tdc=tdc)
+    assert len(layers) == 4
+
+
+def test_any_match_is_fullmatch() -> None:
+    assert any_match('conv1', ['conv1'])
+    assert not any_match('xconv1y', ['conv1'])
+    assert any_match('xconv1y', ['.*conv1.*'])
+
+
+def test_get_flattened_modules_leaves_only() -> None:
+    names = [n for n, _ in get_flattened_modules(LeNet())]
+    assert 'conv1' in names and 'fc3' in names
+    assert '' not in names  # root is not a leaf
+
+
+# ---------------------------------------------------------------- helpers
+
+def test_linear_helper_shapes_and_grads() -> None:
+    lin = torch.nn.Linear(7, 3)
+    h = LinearModuleHelper(lin)
+    assert h.a_factor_shape == (8, 8)
+    assert h.g_factor_shape == (3, 3)
+    lin.weight.grad = torch.randn(3, 7)
+    lin.bias.grad = torch.randn(3)
+    g = h.get_grad()
+    assert g.shape == (3, 8)
+    torch.testing.assert_close(g[:, -1], lin.bias.grad)
+    new = torch.randn(3, 8)
+    h.set_grad(new)
+    torch.testing.assert_close(lin.weight.grad, new[:, :-1])
+    torch.testing.assert_close(lin.bias.grad, new[:, -1])
+
+
+def test_conv_helper_shapes_and_grads() -> None:
+    conv = torch.nn.Conv2d(2, 4, 3, padding=1)
+    h = Conv2dModuleHelper(conv)
+    assert h.a_factor_shape == (2 * 9 + 1, 2 * 9 + 1)
+    assert h.g_factor_shape == (4, 4)
+    conv.weight.grad = torch.randn(4, 2, 3, 3)
+    conv.bias.grad = torch.randn(4)
+    g = h.get_grad()
+    assert g.shape == (4, 19)
+    h.set_grad(g)
+    torch.testing.assert_close(
+        conv.weight.grad, g[:, :-1].reshape(4, 2, 3, 3),
+    )
+
+
+# -------------------------------------------------------- preconditioner API
+
+def test_preconditioner_validation() -> None:
+    model = TinyModel()
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, factor_update_steps=0)
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, inv_update_steps=-1)
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, damping=0)
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, factor_decay=1.5)
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, kl_clip=0.0)
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, allreduce_bucket_cap_mb=-1)
+    with pytest.raises(ValueError):
+        KFACPreconditioner(
+            model,
+            compute_eigenvalue_outer_product=True,
+            colocate_factors=False,
+        )
+    with pytest.raises(ValueError):
+        KFACPreconditioner(model, grad_worker_fraction=2.0)
+
+
+def test_preconditioner_string_enums() -> None:
+    model = TinyModel()
+    p = KFACPreconditioner(
+        model,
+        assignment_strategy='memory',
+        compute_method='inverse',
+        compute_eigenvalue_outer_product=False,
+    )
+    assert p.assignment_strategy == AssignmentStrategy.MEMORY
+    assert repr(p)  # smoke: sorted-config repr
+
+
+def test_state_dict_roundtrip_and_resume() -> None:
+    torch.manual_seed(0)
+    model = TinyModel()
+    x = torch.randn(16, 10)
+    y = torch.randint(0, 3, (16,))
+    p = KFACPreconditioner(model, factor_update_steps=1, inv_update_steps=1)
+    loss = torch.nn.functional.cross_entropy(model(x), y)
+    loss.backward()
+    p.step()
+
+    sd = p.state_dict()
+    assert sd['steps'] == 1
+    assert 'damping' in sd and 'layers' in sd
+    assert set(next(iter(sd['layers'].values())).keys()) == {'A', 'G'}
+
+    model2 = TinyModel()
+    p2 = KFACPreconditioner(model2, factor_update_steps=1, inv_update_steps=1)
+    p2.load_state_dict(sd, compute_inverses=True)
+    assert p2.steps == 1
+    for (_, (n1, l1)), (_, (n2, l2)) in zip(
+        p._layers.items(), p2._layers.items(),
+    ):
+        torch.testing.assert_close(l1.a_factor, l2.a_factor)
+        torch.testing.assert_close(l1.g_factor, l2.g_factor)
+        # inverses recomputed on load
+        assert l2.qa is not None and l2.qg is not None
+
+
+def test_state_dict_without_factors() -> None:
+    model = TinyModel()
+    p = KFACPreconditioner(model)
+    sd = p.state_dict(include_factors=False)
+    assert 'layers' not in sd
+    p2 = KFACPreconditioner(TinyModel())
+    with pytest.warns(UserWarning):
+        p2.load_state_dict(sd, compute_inverses=True)
+
+
+def test_load_state_dict_layer_count_mismatch() -> None:
+    p = KFACPreconditioner(TinyModel())
+    sd = p.state_dict()
+    sd['layers'] = {'one': {'A': None, 'G': None}}
+    p2 = KFACPreconditioner(TinyModel())
+    with pytest.raises(ValueError):
+        p2.load_state_dict(sd)
+
+
+def test_memory_usage_accounting() -> None:
+    torch.manual_seed(0)
+    model = TinyModel()
+    p = KFACPreconditioner(model, factor_update_steps=1, inv_update_steps=1)
+    x = torch.randn(8, 10)
+    torch.nn.functional.cross_entropy(model(x), torch.randint(0, 3, (8,))).backward()
+    p.step()
+    usage = p.memory_usage()
+    assert usage['a_factors'] > 0
+    assert usage['g_factors'] > 0
+    assert usage['a_inverses'] > 0
+    assert usage['total'] == sum(v for k, v in usage.items() if k != 'total')
+
+
+def test_reset_batch() -> None:
+    torch.manual_seed(0)
+    model = TinyModel()
+    p = KFACPreconditioner(
+        model, factor_update_steps=1, update_factors_in_hook=False,
+    )
+    x = torch.randn(8, 10)
+    torch.nn.functional.cross_entropy(model(x), torch.randint(0, 3, (8,))).backward()
+    assert any(l._a_count > 0 for _, l in p._layers.values())
+    p.reset_batch()
+    assert all(l._a_count == 0 for _, l in p._layers.values())

@@ -1,0 +1,71 @@
+"""Example CLIs smoke tests (subprocess, synthetic data, CPU)."""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import tempfile
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(args: list[str], timeout: int = 240) -> subprocess.CompletedProcess:
+    return subprocess.run(
+        [sys.executable, *args],
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=timeout,
+    )
+
+
+def test_cifar_example_runs_and_checkpoints() -> None:
+    with tempfile.TemporaryDirectory() as td:
+        args = [
+            'examples/torch_cifar10_resnet.py',
+            '--epochs', '1',
+            '--max-steps-per-epoch', '3',
+            '--batch-size', '16',
+            '--val-batch-size', '16',
+            '--model', 'resnet20',
+            '--kfac-inv-update-steps', '2',
+            '--checkpoint-dir', td,
+            '--checkpoint-freq', '1',
+        ]
+        r = _run(args)
+        assert r.returncode == 0, r.stderr[-2000:]
+        assert os.path.exists(os.path.join(td, 'checkpoint_1.pth.tar'))
+        # resume path
+        r2 = _run(args)
+        assert r2.returncode == 0, r2.stderr[-2000:]
+
+
+def test_language_model_example_runs() -> None:
+    r = _run(
+        [
+            'examples/torch_language_model.py',
+            '--epochs', '1',
+            '--steps-per-epoch', '4',
+            '--batch-size', '4',
+            '--vocab', '256',
+        ],
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert 'ppl=' in r.stdout
+
+
+def test_cifar_example_no_kfac() -> None:
+    with tempfile.TemporaryDirectory() as td:
+        r = _run(
+            [
+                'examples/torch_cifar10_resnet.py',
+                '--epochs', '1',
+                '--max-steps-per-epoch', '2',
+                '--batch-size', '8',
+                '--model', 'resnet20',
+                '--kfac-inv-update-steps', '0',
+                '--checkpoint-dir', td,
+            ],
+        )
+        assert r.returncode == 0, r.stderr[-2000:]
