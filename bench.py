@@ -44,6 +44,12 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--factor-update-steps', type=int, default=10)
     p.add_argument('--inv-update-steps', type=int, default=100)
     p.add_argument('--no-kfac', action='store_true')
+    p.add_argument(
+        '--async-inverse',
+        type=int,
+        default=1,
+        help='pipeline eigendecompositions behind training steps (0=off)',
+    )
     return p.parse_args()
 
 
@@ -54,6 +60,7 @@ def main() -> None:
     local_rank = int(os.environ.get('LOCAL_RANK', '0'))
 
     assert torch.cuda.is_available(), 'bench.py requires a GPU'
+    local_rank = local_rank % torch.cuda.device_count()
     torch.cuda.set_device(local_rank)
     device = torch.device('cuda', local_rank)
 
@@ -117,6 +124,7 @@ def main() -> None:
             allreduce_bucket_cap_mb=25.0,
             compute_eigenvalue_outer_product=True,
             skip_layers=[],
+            inv_update_async=bool(args.async_inverse),
         )
 
     bs = args.batch_size
@@ -186,6 +194,7 @@ def main() -> None:
                 'strategy': strategy_name,
                 'factor_update_steps': args.factor_update_steps,
                 'inv_update_steps': args.inv_update_steps,
+                'async_inverse': bool(args.async_inverse),
                 'precond_step_ms_mean': (
                     sum(precond_times) / len(precond_times) * 1000.0
                     if precond_times

@@ -86,6 +86,8 @@ def main() -> None:
     if world > 1:
         dist.init_process_group(args.backend or ('nccl' if use_cuda else 'gloo'))
     if use_cuda:
+        # modulo so N-rank smokes run on fewer GPUs (e.g. 2 ranks, 1 GPU)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
     device = torch.device('cuda', local_rank) if use_cuda else torch.device('cpu')
     torch.manual_seed(args.seed)

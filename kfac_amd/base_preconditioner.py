@@ -51,6 +51,8 @@ class BaseKFACPreconditioner:
         lr: Callable[[int], float] | float = 0.1,
         accumulation_steps: int = 1,
         update_factors_in_hook: bool = True,
+        inv_update_async: bool = False,
+        inv_async_delay: int = 15,
         defaults: dict[str, Any] | None = None,
         loglevel: int = logging.DEBUG,
     ) -> None:
@@ -97,6 +99,10 @@ class BaseKFACPreconditioner:
         self._lr = lr
         self._tdc = tdc
         self._update_factors_in_hook = update_factors_in_hook
+
+        self._inv_update_async = inv_update_async
+        self._inv_async_delay = inv_async_delay
+        self._async_job: Any = None
 
         self._steps = 0
         self._mini_steps: dict[str, int] = defaultdict(int)
@@ -289,19 +295,18 @@ class BaseKFACPreconditioner:
         self._tdc.flush_allreduce_buckets()
 
         if self.steps % self.inv_update_steps == 0:
-            self._compute_local_inverses()
-            if self._assignment.broadcast_inverses():
-                for name, layer in reversed(list(self._layers.values())):
-                    if self._assignment.is_grad_worker(name):
-                        layer.broadcast_a_inv(
-                            src=self._assignment.inv_worker(name, 'A'),
-                            group=self._assignment.grad_worker_group(name),
-                        )
-                        layer.broadcast_g_inv(
-                            src=self._assignment.inv_worker(name, 'G'),
-                            group=self._assignment.grad_worker_group(name),
-                        )
-            self._tdc.flush_allreduce_buckets()
+            if self._async_job is not None:
+                self._finish_async_inverses()
+            if self._can_async_inverses():
+                self._launch_async_inverses()
+            else:
+                self._compute_local_inverses()
+                self._broadcast_inverses()
+        elif (
+            self._async_job is not None
+            and self.steps >= self._async_job['due']
+        ):
+            self._finish_async_inverses()
 
         grouped = self._grouped_precondition()
         for name, layer in reversed(list(self._layers.values())):
@@ -361,6 +366,148 @@ class BaseKFACPreconditioner:
         for (layer, *_), out in zip(work, outs):
             layer.grad = out
         return True
+
+    # -- async inverse pipeline --------------------------------------------
+    #
+    # rocSOLVER's eigendecomposition is host-launch-bound (~48k small
+    # kernels per phase), so even on a side stream the *host* is busy for
+    # the whole phase. The async path runs the batched eigendecomposition
+    # in a worker THREAD on its own HIP stream, overlapping the next
+    # ``inv_async_delay`` training steps; all ranks swap in the new
+    # second-order state (and issue the inverse broadcasts) at exactly
+    # boundary+delay so collectives stay matched. Preconditioning between
+    # boundary and swap uses the previous inverses — well within K-FAC's
+    # by-design staleness (inverses are already inv_update_steps old).
+
+    def _broadcast_inverses(self) -> None:
+        if self._assignment.broadcast_inverses():
+            for name, layer in reversed(list(self._layers.values())):
+                if self._assignment.is_grad_worker(name):
+                    layer.broadcast_a_inv(
+                        src=self._assignment.inv_worker(name, 'A'),
+                        group=self._assignment.grad_worker_group(name),
+                    )
+                    layer.broadcast_g_inv(
+                        src=self._assignment.inv_worker(name, 'G'),
+                        group=self._assignment.grad_worker_group(name),
+                    )
+        self._tdc.flush_allreduce_buckets()
+
+    def _can_async_inverses(self) -> bool:
+        """Async is uniform across ranks: depends only on config + types."""
+        from kfac_amd.layers.eigen import KFACEigenLayer
+
+        if not self._inv_update_async or self._steps == 0:
+            return False
+        layers = list(self._layers.values())
+        if not layers:
+            return False
+        if layers[0][1].module.device.type != 'cuda':
+            return False
+        return all(
+            isinstance(layer, KFACEigenLayer) and layer.symmetric_factors
+            for _, layer in layers
+        )
+
+    def _launch_async_inverses(self) -> None:
+        import threading
+
+        from kfac_amd.layers.eigen import KFACEigenLayer
+
+        rank = get_rank()
+        damping = self.damping
+        work: list[tuple[KFACEigenLayer, torch.Tensor | None, torch.Tensor | None]] = []
+        for name, layer in reversed(list(self._layers.values())):
+            a_mine = rank == self._assignment.inv_worker(name, 'A')
+            g_mine = rank == self._assignment.inv_worker(name, 'G')
+            if not (a_mine or g_mine):
+                continue
+            a = layer.a_factor.detach().clone() if a_mine else None
+            g = layer.g_factor.detach().clone() if g_mine else None
+            work.append((layer, a, g))
+
+        device = list(self._layers.values())[0][1].module.device
+        ready = torch.cuda.Event()
+        ready.record()
+        stream = torch.cuda.Stream(device=device)
+        job: dict[str, Any] = {
+            'due': self._steps
+            + max(1, min(self._inv_async_delay, self.inv_update_steps - 1)),
+            'results': None,
+            'error': None,
+        }
+
+        def worker() -> None:
+            try:
+                from collections import defaultdict as dd
+
+                with torch.cuda.stream(stream):
+                    stream.wait_event(ready)
+                    results: dict[Any, dict[str, torch.Tensor | None]] = {}
+                    for which in ('a', 'g'):
+                        groups: dict[int, list[tuple[Any, torch.Tensor]]] = dd(list)
+                        for layer, a, g in work:
+                            fac = a if which == 'a' else g
+                            if fac is not None:
+                                groups[fac.shape[0]].append((layer, fac))
+                        for n, items in groups.items():
+                            if len(items) == 1:
+                                d, q = torch.linalg.eigh(
+                                    items[0][1].to(torch.float32),
+                                )
+                                d = torch.clamp(d, min=0.0).unsqueeze(0)
+                                q = q.unsqueeze(0)
+                            else:
+                                stack = torch.stack(
+                                    [f.to(torch.float32) for _, f in items],
+                                )
+                                d, q = torch.linalg.eigh(stack)
+                                d = torch.clamp(d, min=0.0)
+                            for i, (layer, _) in enumerate(items):
+                                res = results.setdefault(layer, {})
+                                res[f'q{which}'] = (
+                                    q[i].to(layer.inv_dtype).contiguous()
+                                )
+                                res[f'd{which}'] = (
+                                    d[i].to(layer.inv_dtype).contiguous()
+                                )
+                    for layer, res in results.items():
+                        if layer.prediv_eigenvalues and 'dg' in res and 'da' in res:
+                            res['dgda'] = 1 / (
+                                torch.outer(res['dg'], res['da']) + damping
+                            )
+                            res['dg'] = None
+                            res['da'] = None
+                stream.synchronize()
+                job['results'] = results
+            except Exception as e:  # pragma: no cover - surfaced at join
+                job['error'] = e
+
+        thread = threading.Thread(target=worker, daemon=True)
+        job['thread'] = thread
+        thread.start()
+        self._async_job = job
+
+    def _finish_async_inverses(self) -> None:
+        job = self._async_job
+        assert job is not None
+        job['thread'].join()
+        self._async_job = None
+        if job['error'] is not None:
+            raise job['error']
+        for layer, res in job['results'].items():
+            if 'qa' in res:
+                layer.qa = res['qa']
+                layer.da = res['da'] if 'da' in res else layer.da
+            if 'qg' in res:
+                layer.qg = res['qg']
+                if layer.prediv_eigenvalues:
+                    layer.dgda = res['dgda']
+                    layer.dg = None
+                    layer.da = None
+                else:
+                    layer.dg = res['dg']
+        self._broadcast_inverses()
 
     def _compute_local_inverses(self) -> None:
         """Compute second-order state for the layers assigned to this rank.

@@ -74,6 +74,8 @@ class KFACPreconditioner(BaseKFACPreconditioner):
         inv_dtype: torch.dtype = torch.float32,
         skip_layers: list[str] | None = None,
         update_factors_in_hook: bool = True,
+        inv_update_async: bool = False,
+        inv_async_delay: int = 15,
         loglevel: int = logging.DEBUG,
     ) -> None:
         """Init KFACPreconditioner.
@@ -87,6 +89,12 @@ class KFACPreconditioner(BaseKFACPreconditioner):
           gfx950) so there is no perf reason to store bf16 factors.
         - ``allreduce_bucket_cap_mb`` buckets are flat fp32 buffers sized
           for the 7-link xGMI fan-out.
+        - ``inv_update_async=True`` pipelines the eigendecomposition
+          phase: a worker thread computes the batched eigendecompositions
+          on a side HIP stream while training continues with the previous
+          second-order state; every rank swaps in the new state exactly
+          ``inv_async_delay`` steps after the boundary (collective-safe).
+          Off by default for strict reference-semantics parity.
         """
         if allreduce_bucket_cap_mb < 0:
             raise ValueError('allreduce_bucket_cap_mb must be >= 0')
@@ -254,6 +262,8 @@ class KFACPreconditioner(BaseKFACPreconditioner):
             accumulation_steps=accumulation_steps,
             assignment=assignment,
             update_factors_in_hook=update_factors_in_hook,
+            inv_update_async=inv_update_async,
+            inv_async_delay=inv_async_delay,
             defaults=defaults,
             tdc=self.tdc,
             loglevel=loglevel,

@@ -101,3 +101,42 @@ def test_resnet50_bf16_step() -> None:
     assert torch.isfinite(loss)
     for p in model.parameters():
         assert torch.isfinite(p).all()
+
+
+def test_async_inverse_pipeline() -> None:
+    """Async eigendecomposition pipeline trains and swaps correctly."""
+    from kfac_amd import KFACPreconditioner
+    from testing.models import LeNet
+
+    torch.manual_seed(42)
+    model = LeNet().cuda()
+    x = torch.randn(64, 1, 28, 28, device='cuda')
+    y = torch.randint(0, 10, (64,), device='cuda')
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    precon = KFACPreconditioner(
+        model,
+        factor_update_steps=1,
+        inv_update_steps=5,
+        lr=0.01,
+        inv_update_async=True,
+        inv_async_delay=2,
+    )
+    losses = []
+    qa_versions = []
+    for i in range(20):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+        layer = next(iter(precon._layers.values()))[1]
+        qa_versions.append(
+            None if layer.qa is None else layer.qa.data_ptr(),
+        )
+    assert losses[0] > losses[-1], losses
+    # the eigen state must have been swapped at least twice after step 0
+    assert len({v for v in qa_versions if v is not None}) >= 3
+    # no job left hanging mid-train
+    if precon._async_job is not None:
+        precon._finish_async_inverses()
