@@ -173,6 +173,18 @@ struct ConvPatchAcc {
           __builtin_memcpy(out, p, 16 * sizeof(__bf16));
           return;
         }
+        if (sw == 2 && ow + 15 < OW && h >= 0 && h < H && w >= 0 &&
+            w + 31 < W) {
+          // stride-2 interior: 16 values span 31 contiguous elements ->
+          // two 32-byte wide loads, keep every other element.
+          int n = t / OH;
+          const T* p = x + ((long)(n * C + c) * H + h) * W + w;
+          __bf16 tmp[32];
+          __builtin_memcpy(tmp, p, 32 * sizeof(__bf16));
+#pragma unroll
+          for (int e = 0; e < 16; ++e) out[e] = tmp[2 * e];
+          return;
+        }
       }
     }
     float tmp[16];

@@ -521,10 +521,13 @@ class BaseKFACPreconditioner:
         layers.
         """
         from kfac_amd.layers.eigen import KFACEigenLayer
+        from kfac_amd.layers.inverse import KFACInverseLayer
 
         rank = get_rank()
         eigen_a: list[KFACEigenLayer] = []
         eigen_g: list[KFACEigenLayer] = []
+        inverse_a: list[KFACInverseLayer] = []
+        inverse_g: list[KFACInverseLayer] = []
         other: list[tuple[str, KFACBaseLayer]] = []
         for name, layer in reversed(list(self._layers.values())):
             a_mine = rank == self._assignment.inv_worker(name, 'A')
@@ -536,6 +539,13 @@ class BaseKFACPreconditioner:
                     eigen_a.append(layer)
                 if g_mine:
                     eigen_g.append(layer)
+            elif (
+                isinstance(layer, KFACInverseLayer) and layer.symmetric_factors
+            ):
+                if a_mine:
+                    inverse_a.append(layer)
+                if g_mine:
+                    inverse_g.append(layer)
             else:
                 other.append((name, layer))
 
@@ -551,11 +561,57 @@ class BaseKFACPreconditioner:
                 layer.dgda = 1 / (torch.outer(dg, da) + damping)
                 layer.dg = None
                 layer.da = None
+        self._batched_cholesky_inverse(inverse_a, 'a', damping)
+        self._batched_cholesky_inverse(inverse_g, 'g', damping)
         for name, layer in other:
             if rank == self._assignment.inv_worker(name, 'A'):
                 layer.compute_a_inv(damping=damping)
             if rank == self._assignment.inv_worker(name, 'G'):
                 layer.compute_g_inv(damping=damping)
+
+    @staticmethod
+    def _batched_cholesky_inverse(
+        layers: list[Any],
+        which: str,
+        damping: float,
+    ) -> None:
+        """Group same-size factors; one batched Cholesky inverse per group.
+
+        The explicit-inverse method's analog of the batched eigh path:
+        (F + damping I)^-1 via batched rocSOLVER potrf/potri.
+        """
+        from collections import defaultdict
+
+        groups: dict[tuple, list[Any]] = defaultdict(list)
+        for layer in layers:
+            factor = layer.a_factor if which == 'a' else layer.g_factor
+            if not isinstance(factor, torch.Tensor):
+                raise RuntimeError(
+                    f'Cannot invert {which.upper()} before it has been '
+                    'computed',
+                )
+            groups[(factor.shape[0], factor.device, factor.dtype)].append(layer)
+        for (n, _dev, _dt), group in groups.items():
+            stack = torch.stack(
+                [
+                    (layer.a_factor if which == 'a' else layer.g_factor).to(
+                        torch.float32,
+                    )
+                    for layer in group
+                ],
+            )
+            stack.diagonal(dim1=-2, dim2=-1).add_(damping)
+            try:
+                chol = torch.linalg.cholesky(stack)
+                inv = torch.cholesky_inverse(chol)
+            except Exception:
+                inv = torch.linalg.inv(stack)
+            for i, layer in enumerate(group):
+                result = inv[i].to(layer.inv_dtype).contiguous()
+                if which == 'a':
+                    layer.a_inv = result
+                else:
+                    layer.g_inv = result
 
     @staticmethod
     def _batched_eigh(layers: list[Any], which: str) -> None:
