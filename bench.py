@@ -45,6 +45,12 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--inv-update-steps', type=int, default=100)
     p.add_argument('--no-kfac', action='store_true')
     p.add_argument(
+        '--compute-method',
+        type=str,
+        default='eigen',
+        choices=['eigen', 'inverse'],
+    )
+    p.add_argument(
         '--async-inverse',
         type=int,
         default=1,
@@ -122,7 +128,8 @@ def main() -> None:
             grad_worker_fraction=strategy,
             accumulation_steps=1,
             allreduce_bucket_cap_mb=25.0,
-            compute_eigenvalue_outer_product=True,
+            compute_method=args.compute_method,
+            compute_eigenvalue_outer_product=args.compute_method == 'eigen',
             skip_layers=[],
             inv_update_async=bool(args.async_inverse),
         )
@@ -195,6 +202,7 @@ def main() -> None:
                 'factor_update_steps': args.factor_update_steps,
                 'inv_update_steps': args.inv_update_steps,
                 'async_inverse': bool(args.async_inverse),
+                'compute_method': args.compute_method,
                 'precond_step_ms_mean': (
                     sum(precond_times) / len(precond_times) * 1000.0
                     if precond_times
