@@ -20,6 +20,8 @@ hipError_t cov_linear_t(hipStream_t, const T*, long, int, int, bool, float*, flo
 template <typename T>
 hipError_t cov_conv_a_t(hipStream_t, const T*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
 template <typename T>
+hipError_t im2col_t(hipStream_t, const T*, T*, int, int, int, int, int, int, int, int, int, int, int);
+template <typename T>
 hipError_t cov_conv_g_t(hipStream_t, const T*, int, int, int, int, float*, float, float);
 hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float);
 hipError_t precond_grouped_f32(hipStream_t, const void*, int, int);
@@ -143,25 +145,38 @@ void cov_conv_a(
   double coeff = coeff_scale / ((double)m * (double)s * (double)s);
   auto stream = current_stream(x);
   float* outp = out.data_ptr<float>();
+  // Materialize the patch matrix once (gather VALU cost paid once), then
+  // run the flat-matrix SYRK with wide coalesced staging. The scratch
+  // lives in the caching allocator; bytes are trivial vs 288 GB HBM3E.
+  const int K_pad = (K + 7) / 8 * 8;
+  auto scratch = torch::empty({m, (long)K_pad}, x.options());
   dispatch_dtype(
       x.scalar_type(),
       [&] {
-        CHECK_OK(kfac::cov_conv_a_t<float>(
-            stream, x.data_ptr<float>(), Nb, C, H, W, (int)kh, (int)kw,
-            (int)sh, (int)sw, (int)ph, (int)pw, bias, outp, (float)beta,
-            (float)coeff));
-      },
-      [&] {
-        CHECK_OK(kfac::cov_conv_a_t<__hip_bfloat16>(
-            stream, (const __hip_bfloat16*)x.data_ptr(), Nb, C, H, W, (int)kh,
-            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, bias, outp,
+        CHECK_OK(kfac::im2col_t<float>(
+            stream, x.data_ptr<float>(), scratch.data_ptr<float>(), Nb, C, H,
+            W, (int)kh, (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, K_pad));
+        CHECK_OK(kfac::cov_linear_t<float>(
+            stream, scratch.data_ptr<float>(), K_pad, (int)m, K, bias, outp,
             (float)beta, (float)coeff));
       },
       [&] {
-        CHECK_OK(kfac::cov_conv_a_t<__half>(
-            stream, (const __half*)x.data_ptr(), Nb, C, H, W, (int)kh,
-            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, bias, outp,
-            (float)beta, (float)coeff));
+        CHECK_OK(kfac::im2col_t<__hip_bfloat16>(
+            stream, (const __hip_bfloat16*)x.data_ptr(),
+            (__hip_bfloat16*)scratch.data_ptr(), Nb, C, H, W, (int)kh,
+            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, K_pad));
+        CHECK_OK(kfac::cov_linear_t<__hip_bfloat16>(
+            stream, (const __hip_bfloat16*)scratch.data_ptr(), K_pad, (int)m,
+            K, bias, outp, (float)beta, (float)coeff));
+      },
+      [&] {
+        CHECK_OK(kfac::im2col_t<__half>(
+            stream, (const __half*)x.data_ptr(),
+            (__half*)scratch.data_ptr(), Nb, C, H, W, (int)kh, (int)kw,
+            (int)sh, (int)sw, (int)ph, (int)pw, K_pad));
+        CHECK_OK(kfac::cov_linear_t<__half>(
+            stream, (const __half*)scratch.data_ptr(), K_pad, (int)m, K,
+            bias, outp, (float)beta, (float)coeff));
       });
 }
 

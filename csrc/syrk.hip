@@ -603,6 +603,85 @@ __global__ void scale_kernel(float* out, long n, float beta) {
   }
 }
 
+// -------------------------------------------------------------- im2col
+//
+// Why materialize at all? PMC on the fused-gather SYRK showed 523 VALU
+// per MFMA: the per-element patch address math re-runs for every
+// (tile-row, tile-col) pair touching the element (~2x tiles per dim), so
+// gather VALU — not HBM — dominated. Materializing runs the gather ONCE
+// (coalesced 16-byte writes along K) and the SYRK then stages from the
+// flat matrix with plain wide loads; the extra HBM round trip is cheap
+// on 8 TB/s HBM3E relative to the n^2 MFMA work.
+template <typename T>
+__global__ __launch_bounds__(256) void im2col_kernel(
+    T* __restrict__ dst,  // [M][K_pad]
+    ConvPatchAcc<T> acc,
+    int K_pad) {
+  const int k8 = K_pad / 8;
+  const long total = (long)acc.M * k8;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int m = (int)(idx / k8);
+    const int k0 = (int)(idx % k8) * 8;
+    // one m decomposition per 8 elements
+    const int ow = m % acc.OW;
+    int t = m / acc.OW;
+    const int oh = t % acc.OH;
+    const int n = t / acc.OH;
+    T vals[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int k = k0 + e;
+      T v = T(0.0f);
+      if (k < acc.K) {
+        const int s = k % acc.kw;
+        const int t2 = k / acc.kw;
+        const int r = t2 % acc.kh;
+        const int c = t2 / acc.kh;
+        const int h = oh * acc.sh - acc.ph + r;
+        const int w = ow * acc.sw - acc.pw + s;
+        if (h >= 0 && h < acc.H && w >= 0 && w < acc.W) {
+          v = acc.x[((long)(n * acc.C + c) * acc.H + h) * acc.W + w];
+        }
+      }
+      vals[e] = v;
+    }
+    __builtin_memcpy(&dst[(long)m * K_pad + k0], vals, 8 * sizeof(T));
+  }
+}
+
+template <typename T>
+hipError_t im2col_t(
+    hipStream_t stream,
+    const T* x,
+    T* dst,
+    int Nb,
+    int C,
+    int H,
+    int W,
+    int kh,
+    int kw,
+    int sh,
+    int sw,
+    int ph,
+    int pw,
+    int K_pad) {
+  int OH = (H + 2 * ph - kh) / sh + 1;
+  int OW = (W + 2 * pw - kw) / sw + 1;
+  int K = C * kh * kw;
+  ConvPatchAcc<T> acc{x,  C,  H,  W,  OH, OW, kh,
+                      kw, sh, sw, ph, pw, Nb * OH * OW,
+                      K,  K};
+  long total = (long)acc.M * (K_pad / 8);
+  int blocks = (int)min((total + 255) / 256, (long)2048);
+  im2col_kernel<T><<<blocks, 256, 0, stream>>>(dst, acc, K_pad);
+  return hipGetLastError();
+}
+
+template hipError_t im2col_t<float>(hipStream_t, const float*, float*, int, int, int, int, int, int, int, int, int, int, int);
+template hipError_t im2col_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
+template hipError_t im2col_t<__half>(hipStream_t, const __half*, __half*, int, int, int, int, int, int, int, int, int, int, int);
+
 // ---------------------------------------------------------------- launchers
 
 static int pick_splits(int M, int n_tiles) {
