@@ -10,6 +10,8 @@
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
+#include <rocblas/rocblas.h>
+#include <rocsolver/rocsolver.h>
 
 namespace kfac {
 
@@ -444,6 +446,65 @@ void kl_clip_accum(
       });
 }
 
+// Batched Jacobi symmetric eigensolver via rocSOLVER syevj: a few large
+// batched kernels per sweep instead of syevd's ~50k tiny tridiagonalization
+// launches — faster for K-FAC's many same-size factors and far friendlier
+// to the async-inverse worker thread (launch-lock contention).
+std::tuple<torch::Tensor, torch::Tensor> eigh_jacobi(
+    torch::Tensor stack,
+    double abstol,
+    int64_t max_sweeps) {
+  check_gpu_contig(stack, "stack");
+  TORCH_CHECK(
+      stack.dim() == 3 && stack.size(1) == stack.size(2),
+      "stack must be (B, n, n)");
+  TORCH_CHECK(stack.scalar_type() == torch::kFloat32, "fp32 only");
+  const int B = (int)stack.size(0);
+  const int n = (int)stack.size(1);
+
+  static rocblas_handle handle = nullptr;
+  if (handle == nullptr) {
+    TORCH_CHECK(
+        rocblas_create_handle(&handle) == rocblas_status_success,
+        "rocblas_create_handle failed");
+  }
+  auto stream = current_stream(stack);
+  rocblas_set_stream(handle, stream);
+
+  // rocSOLVER overwrites A with the eigenvectors (column-major -> the
+  // row-major view holds V^T).
+  auto a = stack.clone();
+  auto w = torch::empty({B, (long)n}, stack.options());
+  auto residual = torch::empty({B}, stack.options());
+  auto opts_i =
+      torch::TensorOptions().device(stack.device()).dtype(torch::kInt32);
+  auto n_sweeps = torch::empty({B}, opts_i);
+  auto info = torch::empty({B}, opts_i);
+
+  auto status = rocsolver_ssyevj_strided_batched(
+      handle,
+      rocblas_esort_ascending,
+      rocblas_evect_original,
+      rocblas_fill_upper,
+      n,
+      a.data_ptr<float>(),
+      n,
+      (rocblas_stride)n * n,
+      (float)abstol,
+      residual.data_ptr<float>(),
+      (rocblas_int)max_sweeps,
+      n_sweeps.data_ptr<int>(),
+      w.data_ptr<float>(),
+      (rocblas_stride)n,
+      info.data_ptr<int>(),
+      B);
+  TORCH_CHECK(
+      status == rocblas_status_success, "rocsolver syevj failed: ", status);
+  // V^T (row-major view) -> V with eigenvectors as columns.
+  auto q = a.transpose(1, 2).contiguous();
+  return {w, q};
+}
+
 torch::Tensor triu_pack(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.dim() == 2 && x.size(0) == x.size(1), "x must be square");
@@ -482,6 +543,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       &precond_eigen_grouped,
       "whole precondition chain for all layers in 4 launches");
   m.def("kl_clip_accum", &kl_clip_accum, "device-side kl-clip accumulation");
+  m.def(
+      "eigh_jacobi",
+      &eigh_jacobi,
+      "batched Jacobi symmetric eigendecomposition (rocSOLVER syevj)",
+      pybind11::arg("stack"),
+      pybind11::arg("abstol") = 0.0,
+      pybind11::arg("max_sweeps") = 100);
   m.def("triu_pack", &triu_pack, "pack upper triangle");
   m.def("triu_unpack", &triu_unpack, "unpack upper triangle");
 }

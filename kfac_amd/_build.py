@@ -71,6 +71,8 @@ def build(force: bool = False, arch: str = 'gfx950', verbose: bool = True) -> st
         *[f'-L{p}' for p in lib_dirs],
         '-ltorch',
         '-ltorch_hip',
+        '-lrocsolver',
+        '-lrocblas',
         '-lc10',
         '-lc10_hip',
         '-ltorch_python',

@@ -107,5 +107,55 @@ def main() -> None:
         print(f'single n={n}: {time.perf_counter() - t0:.4f}s')
 
 
-if __name__ == '__main__':
+if __name__ == '__main__' and '--jacobi' not in sys.argv:
     main()
+
+
+def bench_jacobi() -> None:
+    """Compare batched syevd vs syevj on the real group distribution."""
+    import time as _t
+
+    from kfac_amd import _kfaccore
+
+    sizes = factor_sizes()
+    from collections import Counter
+
+    dist = Counter(sizes)
+    print('jacobi-vs-syevd per group:')
+    total_d = total_j = total_jl = 0.0
+    for n, count in sorted(dist.items(), reverse=True):
+        a = torch.randn(count, n, n, device='cuda')
+        stack = 0.95 * torch.eye(n, device='cuda').expand(count, n, n).clone()
+        stack += 0.05 * (a @ a.transpose(1, 2)) / n
+
+        def t(fn):
+            fn()
+            torch.cuda.synchronize()
+            t0 = _t.perf_counter()
+            fn()
+            torch.cuda.synchronize()
+            return _t.perf_counter() - t0
+
+        td = t(lambda: torch.linalg.eigh(stack))
+        tj = t(lambda: _kfaccore.eigh_jacobi(stack, 0.0, 100))
+        tjl = t(lambda: _kfaccore.eigh_jacobi(stack, 1e-5, 20))
+        total_d += td
+        total_j += tj
+        total_jl += tjl
+        # accuracy of the loose variant
+        w, q = _kfaccore.eigh_jacobi(stack, 1e-5, 20)
+        recon = q @ torch.diag_embed(w) @ q.transpose(1, 2)
+        err = (recon - stack).abs().max().item()
+        print(
+            f'  n={n:5d} x{count:2d}: syevd {td*1000:8.2f} ms | '
+            f'syevj {tj*1000:8.2f} | syevj-loose {tjl*1000:8.2f} '
+            f'(recon err {err:.2e})',
+        )
+    print(
+        f'TOTAL: syevd {total_d*1000:.1f} ms, syevj {total_j*1000:.1f} ms, '
+        f'syevj-loose {total_jl*1000:.1f} ms',
+    )
+
+
+if __name__ == '__main__' and '--jacobi' in sys.argv:
+    bench_jacobi()
