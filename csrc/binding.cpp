@@ -20,8 +20,6 @@ enum class Dtype : int { F32 = 0, BF16 = 1, F16 = 2 };
 template <typename T>
 hipError_t cov_linear_t(hipStream_t, const T*, long, int, int, bool, float*, float, float);
 template <typename T>
-hipError_t cov_conv_a_t(hipStream_t, const T*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
-template <typename T>
 hipError_t im2col_t(hipStream_t, const T*, T*, int, int, int, int, int, int, int, int, int, int, int);
 template <typename T>
 hipError_t cov_conv_g_t(hipStream_t, const T*, int, int, int, int, float*, float, float);

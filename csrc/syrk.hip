@@ -751,33 +751,6 @@ hipError_t cov_linear_t(
 }
 
 template <typename T>
-hipError_t cov_conv_a_t(
-    hipStream_t stream,
-    const T* x,
-    int Nb,
-    int C,
-    int H,
-    int W,
-    int kh,
-    int kw,
-    int sh,
-    int sw,
-    int ph,
-    int pw,
-    bool bias,
-    float* out,
-    float beta,
-    float coeff) {
-  int OH = (H + 2 * ph - kh) / sh + 1;
-  int OW = (W + 2 * pw - kw) / sw + 1;
-  int K = C * kh * kw;
-  ConvPatchAcc<T> acc{x,  C,  H,  W,  OH, OW, kh,
-                      kw, sh, sw, ph, pw, Nb * OH * OW,
-                      K,  K + (bias ? 1 : 0)};
-  return launch_syrk<use_bf16_mfma<T>()>(stream, out, acc.Ncols, acc, beta, coeff);
-}
-
-template <typename T>
 hipError_t cov_conv_g_t(
     hipStream_t stream,
     const T* g,
@@ -796,9 +769,6 @@ hipError_t cov_conv_g_t(
 template hipError_t cov_linear_t<float>(hipStream_t, const float*, long, int, int, bool, float*, float, float);
 template hipError_t cov_linear_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, long, int, int, bool, float*, float, float);
 template hipError_t cov_linear_t<__half>(hipStream_t, const __half*, long, int, int, bool, float*, float, float);
-template hipError_t cov_conv_a_t<float>(hipStream_t, const float*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
-template hipError_t cov_conv_a_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
-template hipError_t cov_conv_a_t<__half>(hipStream_t, const __half*, int, int, int, int, int, int, int, int, int, int, bool, float*, float, float);
 template hipError_t cov_conv_g_t<float>(hipStream_t, const float*, int, int, int, int, float*, float, float);
 template hipError_t cov_conv_g_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, int, int, int, int, float*, float, float);
 template hipError_t cov_conv_g_t<__half>(hipStream_t, const __half*, int, int, int, int, float*, float, float);

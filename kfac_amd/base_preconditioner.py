@@ -627,7 +627,7 @@ class BaseKFACPreconditioner:
                     'been computed',
                 )
             groups[(factor.shape[0], factor.device, factor.dtype)].append(layer)
-        for (n, _dev, _dt), group in groups.items():
+        for (n, dev, _dt), group in groups.items():
             if len(group) == 1:
                 layer = group[0]
                 if which == 'a':
@@ -643,7 +643,18 @@ class BaseKFACPreconditioner:
                     for layer in group
                 ],
             )
-            d, q = torch.linalg.eigh(stack)
+            if n <= 64 and dev.type == 'cuda':
+                # rocSOLVER's batched Jacobi beats syevd only for tiny
+                # factors (profiles/eigh_strategies.md); use it there.
+                from kfac_amd import ops as _ops
+
+                ext = _ops._load_ext()
+                if ext is not None:
+                    d, q = ext.eigh_jacobi(stack, 0.0, 100)
+                else:
+                    d, q = torch.linalg.eigh(stack)
+            else:
+                d, q = torch.linalg.eigh(stack)
             d = torch.clamp(d, min=0.0)
             for i, layer in enumerate(group):
                 qv = q[i].to(layer.inv_dtype).contiguous()
