@@ -28,6 +28,7 @@ hipError_t precond_grouped_f32(hipStream_t, const void*, int, int);
 template <typename T>
 hipError_t kl_clip_accum_t(hipStream_t, float*, const T*, const T*, long);
 hipError_t triu_pack_f32(hipStream_t, float*, const float*, int);
+hipError_t syevj_small_f32(hipStream_t, const float*, float*, float*, int, int, int, float);
 hipError_t triu_unpack_f32(hipStream_t, float*, const float*, int);
 
 }  // namespace kfac
@@ -503,6 +504,29 @@ std::tuple<torch::Tensor, torch::Tensor> eigh_jacobi(
   return {w, q};
 }
 
+// Hand-written LDS-resident batched Jacobi eigensolver (n <= 128): one
+// kernel launch per group; eigenvalues UNSORTED (K-FAC is order-
+// invariant) and clamped >= 0.
+std::tuple<torch::Tensor, torch::Tensor> syevj_small(
+    torch::Tensor stack,
+    int64_t max_sweeps,
+    double tol) {
+  check_gpu_contig(stack, "stack");
+  TORCH_CHECK(
+      stack.dim() == 3 && stack.size(1) == stack.size(2),
+      "stack must be (B, n, n)");
+  TORCH_CHECK(stack.scalar_type() == torch::kFloat32, "fp32 only");
+  const int B = (int)stack.size(0);
+  const int n = (int)stack.size(1);
+  TORCH_CHECK(n <= 128, "syevj_small supports n <= 128");
+  auto w = torch::empty({B, (long)n}, stack.options());
+  auto v = torch::empty_like(stack);
+  CHECK_OK(kfac::syevj_small_f32(
+      current_stream(stack), stack.data_ptr<float>(), w.data_ptr<float>(),
+      v.data_ptr<float>(), n, B, (int)max_sweeps, (float)tol));
+  return {w, v};
+}
+
 torch::Tensor triu_pack(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.dim() == 2 && x.size(0) == x.size(1), "x must be square");
@@ -541,6 +565,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       &precond_eigen_grouped,
       "whole precondition chain for all layers in 4 launches");
   m.def("kl_clip_accum", &kl_clip_accum, "device-side kl-clip accumulation");
+  m.def(
+      "syevj_small",
+      &syevj_small,
+      "LDS-resident batched Jacobi eigensolver (n <= 128, unsorted)",
+      pybind11::arg("stack"),
+      pybind11::arg("max_sweeps") = 30,
+      pybind11::arg("tol") = 1e-7);
   m.def(
       "eigh_jacobi",
       &eigh_jacobi,

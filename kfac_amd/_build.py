@@ -19,6 +19,7 @@ OUT = os.path.join(REPO_ROOT, 'kfac_amd', '_kfaccore.so')
 SOURCES = [
     os.path.join(CSRC, 'syrk.hip'),
     os.path.join(CSRC, 'gemm.hip'),
+    os.path.join(CSRC, 'eigh.hip'),
     os.path.join(CSRC, 'binding.cpp'),
 ]
 HEADERS = [os.path.join(CSRC, 'common.h')]

@@ -643,14 +643,16 @@ class BaseKFACPreconditioner:
                     for layer in group
                 ],
             )
-            if n <= 64 and dev.type == 'cuda':
-                # rocSOLVER's batched Jacobi beats syevd only for tiny
-                # factors (profiles/eigh_strategies.md); use it there.
+            if n <= 128 and dev.type == 'cuda':
+                # hand-written LDS-resident batched Jacobi: one launch
+                # per group vs rocSOLVER syevd's thousands
+                # (csrc/eigh.hip; eigenvalue order is irrelevant to the
+                # Kronecker preconditioner).
                 from kfac_amd import ops as _ops
 
                 ext = _ops._load_ext()
                 if ext is not None:
-                    d, q = ext.eigh_jacobi(stack, 0.0, 100)
+                    d, q = ext.syevj_small(stack, 30, 1e-7)
                 else:
                     d, q = torch.linalg.eigh(stack)
             else:
