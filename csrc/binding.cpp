@@ -518,7 +518,7 @@ std::tuple<torch::Tensor, torch::Tensor> syevj_small(
   TORCH_CHECK(stack.scalar_type() == torch::kFloat32, "fp32 only");
   const int B = (int)stack.size(0);
   const int n = (int)stack.size(1);
-  TORCH_CHECK(n <= 128, "syevj_small supports n <= 128");
+  TORCH_CHECK(n <= 64, "syevj_small supports n <= 64");
   auto w = torch::empty({B, (long)n}, stack.options());
   auto v = torch::empty_like(stack);
   CHECK_OK(kfac::syevj_small_f32(
@@ -568,7 +568,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def(
       "syevj_small",
       &syevj_small,
-      "LDS-resident batched Jacobi eigensolver (n <= 128, unsorted)",
+      "LDS-resident batched Jacobi eigensolver (n <= 64, unsorted)",
       pybind11::arg("stack"),
       pybind11::arg("max_sweeps") = 30,
       pybind11::arg("tol") = 1e-7);

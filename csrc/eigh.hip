@@ -1,97 +1,94 @@
 // Batched symmetric eigensolver for small factors (K5 of SURVEY §2.4).
 //
-// Two-sided cyclic Jacobi, one workgroup per matrix, A and the
-// eigenvector accumulator V both LDS-resident (NMAX=64: 33 KB -> 4
-// blocks/CU; NMAX=128: 132 KB -> 1 block/CU). Round-robin tournament
-// ordering gives NMAX/2 independent rotations per round; each round is
-// three barrier phases (rotation params -> row updates -> col+V updates)
-// so no two rotations ever touch the same element in a phase.
+// Two-sided cyclic Jacobi, ONE WAVE per matrix (n <= 64), A and the
+// eigenvector accumulator V LDS-resident (33 KB -> 4 blocks/CU). A
+// single wave executes phases in lockstep, so the inter-phase barriers
+// are single-wave s_barriers (near-free) instead of multi-wave
+// rendezvous — the multi-wave variant measured barrier-latency-bound.
+// Round-robin tournament pairs are precomputed into LDS each round so
+// the update loops are free of div/mod.
 //
-// Why hand-written: rocSOLVER's syevd costs ~50k host-launched kernels
-// per batch (launch-bound for small n) and its syevj is 18x slower than
-// syevd (profiles/eigh_strategies.md); this kernel is ONE launch per
-// group and leaves the eigenvalues unsorted (K-FAC's Kronecker
-// preconditioner is order-invariant as long as d and Q columns match).
+// Why hand-written: rocSOLVER syevd is host-launch-bound for small n
+// (~50k kernels per inverse phase) and its own syevj is 18x slower than
+// syevd (profiles/eigh_strategies.md). Eigenvalues come out UNSORTED
+// (K-FAC\'s Kronecker preconditioner is order-invariant) and clamped
+// >= 0 as the eigen layer requires.
 
 #include "common.h"
 
 namespace kfac {
 
-// round-robin pair schedule: round r (0..m-2), slot j (0..m/2-1)
-__device__ __forceinline__ void rr_pair(int m, int r, int j, int* p, int* q) {
-  if (j == 0) {
-    *p = m - 1;
-    *q = r;
-  } else {
-    *p = (r + j) % (m - 1);
-    *q = (r - j + (m - 1)) % (m - 1);
-  }
-}
+constexpr int JN = 64;   // matrix capacity
+constexpr int JLD = JN + 1;
 
-template <int NMAX>
-__global__ __launch_bounds__(256) void syevj_small_kernel(
+__global__ __launch_bounds__(64) void syevj_wave_kernel(
     const float* __restrict__ a_stack,  // [batch][n][n]
     float* __restrict__ w_out,          // [batch][n]
     float* __restrict__ v_out,          // [batch][n][n] (columns = vectors)
     int n,
     int max_sweeps,
     float tol) {
-  constexpr int LD = NMAX + 1;
-  __shared__ float As[NMAX][LD];
-  __shared__ float Vs[NMAX][LD];
-  __shared__ float cs[NMAX / 2];
-  __shared__ float ss[NMAX / 2];
-  __shared__ float red[4];
+  __shared__ float As[JN][JLD];
+  __shared__ float Vs[JN][JLD];
+  __shared__ float cs[JN / 2];
+  __shared__ float ss[JN / 2];
+  __shared__ unsigned char ps[JN / 2];
+  __shared__ unsigned char qs[JN / 2];
   __shared__ int converged;
 
-  const int tid = threadIdx.x;
+  const int lane = threadIdx.x;
   const long mat = blockIdx.x;
   const float* A = a_stack + mat * (long)n * n;
 
-  // load (zero-pad to NMAX; padding stays diagonal so it never mixes)
-  for (int idx = tid; idx < NMAX * NMAX; idx += 256) {
-    int i = idx / NMAX;
-    int j = idx % NMAX;
+  for (int idx = lane; idx < JN * JN; idx += 64) {
+    int i = idx >> 6;
+    int j = idx & 63;
     As[i][j] = (i < n && j < n) ? A[(long)i * n + j] : 0.0f;
     Vs[i][j] = (i == j) ? 1.0f : 0.0f;
   }
   __syncthreads();
 
-  const int m = NMAX;
+  constexpr int m = JN;
   for (int sweep = 0; sweep < max_sweeps; ++sweep) {
-    // convergence: off(A)^2 <= tol^2 * diag(A)^2
     float off2 = 0.0f;
-    for (int idx = tid; idx < n * n; idx += 256) {
+    float d2 = 0.0f;
+    for (int idx = lane; idx < n * n; idx += 64) {
       int i = idx / n;
       int j = idx % n;
       float v = As[i][j];
-      if (i != j) off2 += v * v;
+      if (i != j) {
+        off2 += v * v;
+      } else {
+        d2 += v * v;
+      }
     }
 #pragma unroll
     for (int o = 32; o > 0; o >>= 1) {
       off2 += __shfl_down(off2, o, 64);
+      d2 += __shfl_down(d2, o, 64);
     }
-    if ((tid & 63) == 0) {
-      red[tid >> 6] = off2;
-    }
-    __syncthreads();
-    if (tid == 0) {
-      float o_total = red[0] + red[1] + red[2] + red[3];
-      float d2 = 0.0f;
-      for (int i = 0; i < n; ++i) d2 += As[i][i] * As[i][i];
-      converged = (o_total <= tol * tol * (d2 + 1e-30f)) ? 1 : 0;
+    if (lane == 0) {
+      converged = (off2 <= tol * tol * (d2 + 1e-30f)) ? 1 : 0;
     }
     __syncthreads();
     if (converged) break;
-    __syncthreads();
 
     for (int r = 0; r < m - 1; ++r) {
-      // phase 1: rotation parameters from the untouched matrix
-      for (int j = tid; j < m / 2; j += 256) {
+      // phase 1: pair table + rotation params (lanes 0..31)
+      if (lane < m / 2) {
+        int j = lane;
         int p, q;
-        rr_pair(m, r, j, &p, &q);
+        if (j == 0) {
+          p = m - 1;
+          q = r;
+        } else {
+          p = (r + j) % (m - 1);
+          q = (r - j + (m - 1)) % (m - 1);
+        }
         int lo = min(p, q);
         int hi = max(p, q);
+        ps[j] = (unsigned char)lo;
+        qs[j] = (unsigned char)hi;
         float apq = As[lo][hi];
         float c = 1.0f;
         float s = 0.0f;
@@ -106,14 +103,13 @@ __global__ __launch_bounds__(256) void syevj_small_kernel(
         ss[j] = s;
       }
       __syncthreads();
-      // phase 2: row updates (rows are partitioned across pairs)
-      for (int idx = tid; idx < (m / 2) * m; idx += 256) {
-        int j = idx / m;
-        int k = idx % m;
-        int p, q;
-        rr_pair(m, r, j, &p, &q);
-        int lo = min(p, q);
-        int hi = max(p, q);
+      // phase 2: row updates (rows partitioned across pairs)
+#pragma unroll 4
+      for (int idx = lane; idx < (m / 2) * m; idx += 64) {
+        int j = idx >> 6;   // pair slot (m = 64)
+        int k = idx & 63;
+        int lo = ps[j];
+        int hi = qs[j];
         float c = cs[j];
         float s = ss[j];
         float alo = As[lo][k];
@@ -123,13 +119,12 @@ __global__ __launch_bounds__(256) void syevj_small_kernel(
       }
       __syncthreads();
       // phase 3: column updates of A and V (columns partitioned)
-      for (int idx = tid; idx < (m / 2) * m; idx += 256) {
-        int j = idx / m;
-        int k = idx % m;
-        int p, q;
-        rr_pair(m, r, j, &p, &q);
-        int lo = min(p, q);
-        int hi = max(p, q);
+#pragma unroll 4
+      for (int idx = lane; idx < (m / 2) * m; idx += 64) {
+        int j = idx >> 6;
+        int k = idx & 63;
+        int lo = ps[j];
+        int hi = qs[j];
         float c = cs[j];
         float s = ss[j];
         float alo = As[k][lo];
@@ -145,13 +140,12 @@ __global__ __launch_bounds__(256) void syevj_small_kernel(
     }
   }
 
-  // eigenvalues (diag, clamped >= 0 as K-FAC requires) and vectors
   float* W = w_out + mat * (long)n;
   float* V = v_out + mat * (long)n * n;
-  for (int i = tid; i < n; i += 256) {
+  for (int i = lane; i < n; i += 64) {
     W[i] = fmaxf(As[i][i], 0.0f);
   }
-  for (int idx = tid; idx < n * n; idx += 256) {
+  for (int idx = lane; idx < n * n; idx += 64) {
     int i = idx / n;
     int j = idx % n;
     V[(long)i * n + j] = Vs[i][j];
@@ -167,15 +161,11 @@ hipError_t syevj_small_f32(
     int batch,
     int max_sweeps,
     float tol) {
-  if (n <= 64) {
-    syevj_small_kernel<64><<<batch, 256, 0, stream>>>(
-        a_stack, w_out, v_out, n, max_sweeps, tol);
-  } else if (n <= 128) {
-    syevj_small_kernel<128><<<batch, 256, 0, stream>>>(
-        a_stack, w_out, v_out, n, max_sweeps, tol);
-  } else {
+  if (n > JN) {
     return hipErrorInvalidValue;
   }
+  syevj_wave_kernel<<<batch, 64, 0, stream>>>(
+      a_stack, w_out, v_out, n, max_sweeps, tol);
   return hipGetLastError();
 }
 

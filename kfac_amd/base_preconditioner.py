@@ -643,7 +643,7 @@ class BaseKFACPreconditioner:
                     for layer in group
                 ],
             )
-            if n <= 128 and dev.type == 'cuda':
+            if n <= 64 and dev.type == 'cuda':
                 # hand-written LDS-resident batched Jacobi: one launch
                 # per group vs rocSOLVER syevd's thousands
                 # (csrc/eigh.hip; eigenvalue order is irrelevant to the

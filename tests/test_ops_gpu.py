@@ -220,7 +220,7 @@ def test_precond_eigen_grouped_matches_per_layer() -> None:
         torch.testing.assert_close(outs[i], expected, rtol=1e-5, atol=1e-5)
 
 
-@pytest.mark.parametrize('n,batch', [(64, 12), (128, 6), (100, 4), (37, 3)])
+@pytest.mark.parametrize('n,batch', [(64, 12), (48, 6), (37, 3)])
 def test_syevj_small_vs_torch(n, batch) -> None:
     """Hand-written batched Jacobi vs torch.linalg.eigh ground truth."""
     torch.manual_seed(13)
