@@ -1,24 +1,27 @@
 // Batched symmetric eigensolver for small factors (K5 of SURVEY §2.4).
 //
-// Two-sided cyclic Jacobi, ONE WAVE per matrix (n <= 64), A and the
-// eigenvector accumulator V LDS-resident (33 KB -> 4 blocks/CU). A
-// single wave executes phases in lockstep, so the inter-phase barriers
-// are single-wave s_barriers (near-free) instead of multi-wave
-// rendezvous — the multi-wave variant measured barrier-latency-bound.
-// Round-robin tournament pairs are precomputed into LDS each round so
-// the update loops are free of div/mod.
+// Two-sided cyclic Jacobi, ONE WAVE per matrix (n <= 64), LDS-resident.
+// Each phase reads one LDS buffer and writes the other (ping-pong): with
+// a single array the compiler must assume every store may alias the next
+// iteration's load and serializes the whole phase on lgkmcnt(0) — the
+// aliased variant measured ~6.3 us per phase, latency-bound. Ping-pong
+// makes all 64 load pairs of a phase independent so they pipeline.
+// Round-robin tournament pairs give 32 independent rotations per round;
+// rows are partitioned in the row phase and columns in the column phase,
+// and since the 32 pairs cover all 64 indices both phases fully rewrite
+// the destination buffer.
 //
 // Why hand-written: rocSOLVER syevd is host-launch-bound for small n
-// (~50k kernels per inverse phase) and its own syevj is 18x slower than
-// syevd (profiles/eigh_strategies.md). Eigenvalues come out UNSORTED
-// (K-FAC\'s Kronecker preconditioner is order-invariant) and clamped
-// >= 0 as the eigen layer requires.
+// (~50k kernels per inverse phase) and rocSOLVER's own syevj is 18x
+// slower than syevd (profiles/eigh_strategies.md). Eigenvalues come out
+// UNSORTED (the Kronecker preconditioner is order-invariant) and
+// clamped >= 0 as the eigen layer requires.
 
 #include "common.h"
 
 namespace kfac {
 
-constexpr int JN = 64;   // matrix capacity
+constexpr int JN = 64;  // matrix capacity (padded with zeros)
 constexpr int JLD = JN + 1;
 
 __global__ __launch_bounds__(64) void syevj_wave_kernel(
@@ -28,8 +31,8 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
     int n,
     int max_sweeps,
     float tol) {
-  __shared__ float As[JN][JLD];
-  __shared__ float Vs[JN][JLD];
+  __shared__ float Abuf[2][JN][JLD];
+  __shared__ float Vbuf[2][JN][JLD];
   __shared__ float cs[JN / 2];
   __shared__ float ss[JN / 2];
   __shared__ unsigned char ps[JN / 2];
@@ -40,11 +43,13 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
   const long mat = blockIdx.x;
   const float* A = a_stack + mat * (long)n * n;
 
+  int acur = 0;  // which A buffer holds the current matrix
+  int vcur = 0;
   for (int idx = lane; idx < JN * JN; idx += 64) {
     int i = idx >> 6;
     int j = idx & 63;
-    As[i][j] = (i < n && j < n) ? A[(long)i * n + j] : 0.0f;
-    Vs[i][j] = (i == j) ? 1.0f : 0.0f;
+    Abuf[0][i][j] = (i < n && j < n) ? A[(long)i * n + j] : 0.0f;
+    Vbuf[0][i][j] = (i == j) ? 1.0f : 0.0f;
   }
   __syncthreads();
 
@@ -52,14 +57,14 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
   for (int sweep = 0; sweep < max_sweeps; ++sweep) {
     float off2 = 0.0f;
     float d2 = 0.0f;
-    for (int idx = lane; idx < n * n; idx += 64) {
-      int i = idx / n;
-      int j = idx % n;
-      float v = As[i][j];
-      if (i != j) {
-        off2 += v * v;
-      } else {
-        d2 += v * v;
+    for (int i = 0; i < n; ++i) {
+      if (lane < n) {
+        float v = Abuf[acur][i][lane];
+        if (i != lane) {
+          off2 += v * v;
+        } else {
+          d2 += v * v;
+        }
       }
     }
 #pragma unroll
@@ -74,6 +79,8 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
     if (converged) break;
 
     for (int r = 0; r < m - 1; ++r) {
+      auto Ain = Abuf[acur];
+      auto Amid = Abuf[acur ^ 1];
       // phase 1: pair table + rotation params (lanes 0..31)
       if (lane < m / 2) {
         int j = lane;
@@ -89,11 +96,11 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
         int hi = max(p, q);
         ps[j] = (unsigned char)lo;
         qs[j] = (unsigned char)hi;
-        float apq = As[lo][hi];
+        float apq = Ain[lo][hi];
         float c = 1.0f;
         float s = 0.0f;
         if (fabsf(apq) > 1e-30f) {
-          float tau = (As[hi][hi] - As[lo][lo]) / (2.0f * apq);
+          float tau = (Ain[hi][hi] - Ain[lo][lo]) / (2.0f * apq);
           float t = copysignf(1.0f, tau) /
                     (fabsf(tau) + sqrtf(1.0f + tau * tau));
           c = rsqrtf(1.0f + t * t);
@@ -103,52 +110,52 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
         ss[j] = s;
       }
       __syncthreads();
-      // phase 2: row updates (rows partitioned across pairs)
-#pragma unroll 4
-      for (int idx = lane; idx < (m / 2) * m; idx += 64) {
-        int j = idx >> 6;   // pair slot (m = 64)
-        int k = idx & 63;
+      // phase 2: row updates, Ain -> Amid (pairs cover every row)
+#pragma unroll 8
+      for (int j = 0; j < m / 2; ++j) {
         int lo = ps[j];
         int hi = qs[j];
         float c = cs[j];
         float s = ss[j];
-        float alo = As[lo][k];
-        float ahi = As[hi][k];
-        As[lo][k] = c * alo - s * ahi;
-        As[hi][k] = s * alo + c * ahi;
+        float alo = Ain[lo][lane];
+        float ahi = Ain[hi][lane];
+        Amid[lo][lane] = c * alo - s * ahi;
+        Amid[hi][lane] = s * alo + c * ahi;
       }
       __syncthreads();
-      // phase 3: column updates of A and V (columns partitioned)
-#pragma unroll 4
-      for (int idx = lane; idx < (m / 2) * m; idx += 64) {
-        int j = idx >> 6;
-        int k = idx & 63;
+      // phase 3: column updates, Amid -> Ain (reuse as output buffer) and
+      // V ping-pong (pairs cover every column)
+      auto Vin = Vbuf[vcur];
+      auto Vout = Vbuf[vcur ^ 1];
+#pragma unroll 8
+      for (int j = 0; j < m / 2; ++j) {
         int lo = ps[j];
         int hi = qs[j];
         float c = cs[j];
         float s = ss[j];
-        float alo = As[k][lo];
-        float ahi = As[k][hi];
-        As[k][lo] = c * alo - s * ahi;
-        As[k][hi] = s * alo + c * ahi;
-        float vlo = Vs[k][lo];
-        float vhi = Vs[k][hi];
-        Vs[k][lo] = c * vlo - s * vhi;
-        Vs[k][hi] = s * vlo + c * vhi;
+        float alo = Amid[lane][lo];
+        float ahi = Amid[lane][hi];
+        Ain[lane][lo] = c * alo - s * ahi;
+        Ain[lane][hi] = s * alo + c * ahi;
+        float vlo = Vin[lane][lo];
+        float vhi = Vin[lane][hi];
+        Vout[lane][lo] = c * vlo - s * vhi;
+        Vout[lane][hi] = s * vlo + c * vhi;
       }
+      vcur ^= 1;
       __syncthreads();
     }
   }
 
   float* W = w_out + mat * (long)n;
   float* V = v_out + mat * (long)n * n;
-  for (int i = lane; i < n; i += 64) {
-    W[i] = fmaxf(As[i][i], 0.0f);
+  if (lane < n) {
+    W[lane] = fmaxf(Abuf[acur][lane][lane], 0.0f);
   }
   for (int idx = lane; idx < n * n; idx += 64) {
     int i = idx / n;
     int j = idx % n;
-    V[(long)i * n + j] = Vs[i][j];
+    V[(long)i * n + j] = Vbuf[vcur][i][j];
   }
 }
 
