@@ -570,8 +570,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       &syevj_small,
       "LDS-resident batched Jacobi eigensolver (n <= 64, unsorted)",
       pybind11::arg("stack"),
-      pybind11::arg("max_sweeps") = 30,
-      pybind11::arg("tol") = 1e-7);
+      pybind11::arg("max_sweeps") = 20,
+      pybind11::arg("tol") = 1e-5);
   m.def(
       "eigh_jacobi",
       &eigh_jacobi,

@@ -33,10 +33,6 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
     float tol) {
   __shared__ float Abuf[2][JN][JLD];
   __shared__ float Vbuf[2][JN][JLD];
-  __shared__ float cs[JN / 2];
-  __shared__ float ss[JN / 2];
-  __shared__ unsigned char ps[JN / 2];
-  __shared__ unsigned char qs[JN / 2];
   __shared__ int converged;
 
   const int lane = threadIdx.x;
@@ -81,9 +77,15 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
     for (int r = 0; r < m - 1; ++r) {
       auto Ain = Abuf[acur];
       auto Amid = Abuf[acur ^ 1];
-      // phase 1: pair table + rotation params (lanes 0..31)
-      if (lane < m / 2) {
-        int j = lane;
+      // phase 1: lane j < 32 computes its pair + rotation into REGISTERS;
+      // phases 2/3 fetch them with __shfl (wave-synchronous, no LDS
+      // round trip, no barrier).
+      int mylo = 0;
+      int myhi = 0;
+      float myc = 1.0f;
+      float mys = 0.0f;
+      {
+        int j = lane & 31;
         int p, q;
         if (j == 0) {
           p = m - 1;
@@ -92,47 +94,40 @@ __global__ __launch_bounds__(64) void syevj_wave_kernel(
           p = (r + j) % (m - 1);
           q = (r - j + (m - 1)) % (m - 1);
         }
-        int lo = min(p, q);
-        int hi = max(p, q);
-        ps[j] = (unsigned char)lo;
-        qs[j] = (unsigned char)hi;
-        float apq = Ain[lo][hi];
-        float c = 1.0f;
-        float s = 0.0f;
+        mylo = min(p, q);
+        myhi = max(p, q);
+        float apq = Ain[mylo][myhi];
         if (fabsf(apq) > 1e-30f) {
-          float tau = (Ain[hi][hi] - Ain[lo][lo]) / (2.0f * apq);
+          float tau = (Ain[myhi][myhi] - Ain[mylo][mylo]) / (2.0f * apq);
           float t = copysignf(1.0f, tau) /
                     (fabsf(tau) + sqrtf(1.0f + tau * tau));
-          c = rsqrtf(1.0f + t * t);
-          s = t * c;
+          myc = rsqrtf(1.0f + t * t);
+          mys = t * myc;
         }
-        cs[j] = c;
-        ss[j] = s;
       }
-      __syncthreads();
+      __builtin_amdgcn_wave_barrier();
       // phase 2: row updates, Ain -> Amid (pairs cover every row)
-#pragma unroll 8
+#pragma unroll
       for (int j = 0; j < m / 2; ++j) {
-        int lo = ps[j];
-        int hi = qs[j];
-        float c = cs[j];
-        float s = ss[j];
+        int lo = __shfl(mylo, j, 64);
+        int hi = __shfl(myhi, j, 64);
+        float c = __shfl(myc, j, 64);
+        float s = __shfl(mys, j, 64);
         float alo = Ain[lo][lane];
         float ahi = Ain[hi][lane];
         Amid[lo][lane] = c * alo - s * ahi;
         Amid[hi][lane] = s * alo + c * ahi;
       }
       __syncthreads();
-      // phase 3: column updates, Amid -> Ain (reuse as output buffer) and
-      // V ping-pong (pairs cover every column)
+      // phase 3: column updates, Amid -> Ain, and V ping-pong
       auto Vin = Vbuf[vcur];
       auto Vout = Vbuf[vcur ^ 1];
-#pragma unroll 8
+#pragma unroll
       for (int j = 0; j < m / 2; ++j) {
-        int lo = ps[j];
-        int hi = qs[j];
-        float c = cs[j];
-        float s = ss[j];
+        int lo = __shfl(mylo, j, 64);
+        int hi = __shfl(myhi, j, 64);
+        float c = __shfl(myc, j, 64);
+        float s = __shfl(mys, j, 64);
         float alo = Amid[lane][lo];
         float ahi = Amid[lane][hi];
         Ain[lane][lo] = c * alo - s * ahi;

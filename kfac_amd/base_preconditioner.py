@@ -652,7 +652,7 @@ class BaseKFACPreconditioner:
 
                 ext = _ops._load_ext()
                 if ext is not None:
-                    d, q = ext.syevj_small(stack, 30, 1e-7)
+                    d, q = ext.syevj_small(stack, 20, 1e-5)
                 else:
                     d, q = torch.linalg.eigh(stack)
             else:

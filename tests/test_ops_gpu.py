@@ -227,7 +227,7 @@ def test_syevj_small_vs_torch(n, batch) -> None:
     a = torch.randn(batch, n, n, device='cuda')
     stack = 0.95 * torch.eye(n, device='cuda').expand(batch, n, n).clone()
     stack = stack + 0.05 * (a @ a.transpose(1, 2)) / n
-    w, v = _ext().syevj_small(stack, 30, 1e-7)
+    w, v = _ext().syevj_small(stack, 20, 1e-5)
     torch.cuda.synchronize()
     # reconstruction: V diag(w) V^T == A
     recon = v @ torch.diag_embed(w) @ v.transpose(1, 2)
@@ -250,6 +250,6 @@ def test_syevj_small_clamps_negative() -> None:
     a = torch.diag(
         torch.linspace(-1.0, 1.0, n, device='cuda'),
     ).unsqueeze(0).contiguous()
-    w, v = _ext().syevj_small(a, 30, 1e-7)
+    w, v = _ext().syevj_small(a, 20, 1e-5)
     torch.cuda.synchronize()
     assert (w >= 0).all()
