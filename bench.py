@@ -33,7 +33,13 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--steps', type=int, default=100)
     p.add_argument('--warmup', type=int, default=20)
     p.add_argument('--batch-size', type=int, default=64, help='per-GPU batch')
-    p.add_argument('--model', type=str, default='resnet50')
+    p.add_argument(
+        '--model',
+        type=str,
+        default='resnet50',
+        choices=['resnet50', 'resnet101', 'resnet152', 'gptneox125m'],
+    )
+    p.add_argument('--seq-len', type=int, default=2048)
     p.add_argument(
         '--strategy',
         type=str,
@@ -77,7 +83,8 @@ def main() -> None:
     from kfac_amd import KFACPreconditioner
     from kfac_amd import ops
     from kfac_amd.enums import DistributedStrategy
-    from kfac_amd.models import resnet50, resnet101, resnet152
+    from kfac_amd.models import gptneox_125m, resnet50, resnet101, resnet152
+    from kfac_amd.models.gptneox import KFAC_SKIP_LAYERS
 
     if os.environ.get('KFAC_AMD_FORCE_EAGER', '0') != '1':
         assert ops.extension_available(), (
@@ -85,9 +92,17 @@ def main() -> None:
         )
 
     torch.manual_seed(1234 + rank)
-    model_fn = {'resnet50': resnet50, 'resnet101': resnet101, 'resnet152': resnet152}[args.model]
+    is_lm = args.model == 'gptneox125m'
+    model_fn = {
+        'resnet50': resnet50,
+        'resnet101': resnet101,
+        'resnet152': resnet152,
+        'gptneox125m': gptneox_125m,
+    }[args.model]
     model = model_fn().to(device)
     model.train()
+    if is_lm and args.batch_size == 64:
+        args.batch_size = 8  # 8 x 2048 tokens per GPU
 
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(
@@ -130,13 +145,18 @@ def main() -> None:
             allreduce_bucket_cap_mb=25.0,
             compute_method=args.compute_method,
             compute_eigenvalue_outer_product=args.compute_method == 'eigen',
-            skip_layers=[],
+            skip_layers=KFAC_SKIP_LAYERS if is_lm else [],
             inv_update_async=bool(args.async_inverse),
         )
 
     bs = args.batch_size
-    x = torch.randn(bs, 3, 224, 224, device=device)
-    y = torch.randint(0, 1000, (bs,), device=device)
+    if is_lm:
+        vocab = 50304
+        x = torch.randint(0, vocab, (bs, args.seq_len), device=device)
+        y = torch.randint(0, vocab, (bs * args.seq_len,), device=device)
+    else:
+        x = torch.randn(bs, 3, 224, 224, device=device)
+        y = torch.randint(0, 1000, (bs,), device=device)
     criterion = torch.nn.CrossEntropyLoss()
 
     precond_times: list[float] = []
@@ -144,7 +164,11 @@ def main() -> None:
     def one_step(timed: bool) -> None:
         optimizer.zero_grad(set_to_none=True)
         with torch.autocast('cuda', dtype=torch.bfloat16):
-            loss = criterion(model(x), y)
+            out = model(x)
+            if is_lm:
+                loss = criterion(out.view(-1, out.size(-1)), y)
+            else:
+                loss = criterion(out, y)
         loss.backward()
         if precon is not None:
             t0 = time.perf_counter() if timed else 0.0
@@ -174,14 +198,21 @@ def main() -> None:
         elapsed = float(t.item())
 
     global_batch = bs * world
-    images_per_sec = global_batch * args.steps / elapsed
+    if is_lm:
+        value = global_batch * args.seq_len * args.steps / elapsed
+        metric = 'tokens/sec (whole node) GPT-NeoX-125M + K-FAC precond'
+        unit = 'tokens/sec'
+    else:
+        value = global_batch * args.steps / elapsed
+        metric = 'images/sec (whole node) ResNet-50 + K-FAC precond'
+        unit = 'images/sec'
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
         result = {
-            'metric': 'images/sec (whole node) ResNet-50 + K-FAC precond',
-            'value': images_per_sec,
-            'unit': 'images/sec',
+            'metric': metric,
+            'value': value,
+            'unit': unit,
             'n_gpus': world,
             'steps': args.steps,
             'warmup': args.warmup,
@@ -194,7 +225,7 @@ def main() -> None:
             'config': {
                 'model': args.model,
                 'global_batch': global_batch,
-                'seq_len': None,
+                'seq_len': args.seq_len if is_lm else None,
                 'parallelism': f'dp{world}',
                 'image_size': 224,
                 'kfac': not args.no_kfac,

@@ -10,6 +10,8 @@ from kfac_amd.models.cifar_resnet import cifar_resnet
 from kfac_amd.models.resnet import resnet50
 from kfac_amd.models.resnet import resnet101
 from kfac_amd.models.resnet import resnet152
+from kfac_amd.models.gptneox import gptneox_125m
+from kfac_amd.models.gptneox import GPTNeoXModel
 from kfac_amd.models.transformer import TransformerModel
 
 __all__ = [
@@ -18,4 +20,6 @@ __all__ = [
     'resnet152',
     'cifar_resnet',
     'TransformerModel',
+    'GPTNeoXModel',
+    'gptneox_125m',
 ]
