@@ -13,6 +13,10 @@ sys.path.insert(0, '.')
 from kfac_amd.models import resnet50  # noqa: E402
 
 
+def lm_shapes() -> list[tuple[int, int]]:
+    return [(3072, 769), (768, 3073)] * 12
+
+
 def shapes() -> list[tuple[int, int]]:
     out = []
     for m in resnet50().modules():
@@ -38,7 +42,9 @@ def timeit(fn, reps=10) -> float:
 def main() -> None:
     from kfac_amd import _kfaccore
 
-    ss = shapes()
+    import sys as _sys
+
+    ss = lm_shapes() if '--lm' in _sys.argv else shapes()
     grads, qas, qgs, dgdas = [], [], [], []
     for m, n in ss:
         grads.append(torch.randn(m, n, device='cuda'))
