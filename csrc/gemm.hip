@@ -16,7 +16,7 @@
 namespace kfac {
 
 constexpr int GBT = 128;
-constexpr int GBK = 64;
+constexpr int GBK = 32;
 constexpr int GLDS = GBT + 4;  // +4 keeps float4 LDS rows 16B-aligned
 
 enum class Epilogue : int { NONE = 0, MUL = 1, DIV_OUTER = 2 };

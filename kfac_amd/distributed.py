@@ -180,6 +180,8 @@ class TorchDistributedCommunicator:
         self._open: dict[tuple, _Bucket] = {}
         self._pending: dict[tuple, list[_BucketFuture]] = defaultdict(list)
         self._inflight: list[_Bucket] = []
+        # group handle + averaging flag per bucket key
+        self._groups: dict[tuple, tuple] = {}
 
     # -- plain collectives -------------------------------------------------
 
@@ -287,7 +289,6 @@ class TorchDistributedCommunicator:
         if bucket is None:
             bucket = _Bucket(self._cap_bytes)
             self._open[key] = bucket
-            self._groups = getattr(self, '_groups', {})
             self._groups[key] = (group, average)
         bucket.append(payload)
         self._pending[key].append(fut)
@@ -304,14 +305,12 @@ class TorchDistributedCommunicator:
         self._pending[key] = []
 
     def _flush_key(self, key: tuple) -> None:
-        groups = getattr(self, '_groups', {})
-        if key in self._open and key in groups:
-            group, average = groups[key]
+        if key in self._open and key in self._groups:
+            group, average = self._groups[key]
             self._launch(key, group, average)
 
     def flush_allreduce_buckets(self) -> None:
         """Launch every open bucket (trailing partial buckets)."""
-        groups = getattr(self, '_groups', {})
         for key in list(self._open.keys()):
-            group, average = groups[key]
+            group, average = self._groups[key]
             self._launch(key, group, average)
