@@ -177,9 +177,12 @@ class KFACBaseLayer:
         a = input_[0]
         if self._a_batch is None:
             shape = self.module.a_factor_shape
-            self._a_batch = torch.zeros(
-                shape, dtype=torch.float32, device=a.device,
+            bdt = (
+                torch.float64
+                if self.factor_dtype == torch.float64
+                else torch.float32
             )
+            self._a_batch = torch.zeros(shape, dtype=bdt, device=a.device)
         beta = 0.0 if self._a_count == 0 else 1.0
         if a.is_cuda:
             from kfac_amd.streams import cov_stream
@@ -202,9 +205,12 @@ class KFACBaseLayer:
         g = grad_output[0]
         if self._g_batch is None:
             shape = self.module.g_factor_shape
-            self._g_batch = torch.zeros(
-                shape, dtype=torch.float32, device=g.device,
+            bdt = (
+                torch.float64
+                if self.factor_dtype == torch.float64
+                else torch.float32
             )
+            self._g_batch = torch.zeros(shape, dtype=bdt, device=g.device)
         coeff = 1.0
         if self.grad_scaler is not None:
             sc = float(self.grad_scaler())

@@ -170,7 +170,7 @@ class KFACEigenLayer(KFACBaseLayer):
         if self.symmetric_factors:
             da, qa = ops.eigh(a, clamp=False)
         else:
-            dac, qac = torch.linalg.eig(a.to(torch.float32))
+            dac, qac = torch.linalg.eig(a)
             da, qa = dac.real, qac.real
         self.qa = qa.to(self.inv_dtype)
         self.da = torch.clamp(da.to(self.inv_dtype), min=0.0)
@@ -184,7 +184,7 @@ class KFACEigenLayer(KFACBaseLayer):
         if self.symmetric_factors:
             dg, qg = ops.eigh(g, clamp=False)
         else:
-            dgc, qgc = torch.linalg.eig(g.to(torch.float32))
+            dgc, qgc = torch.linalg.eig(g)
             dg, qg = dgc.real, qgc.real
         self.qg = qg.to(self.inv_dtype)
         self.dg = torch.clamp(dg.to(self.inv_dtype), min=0.0)

@@ -38,7 +38,7 @@ def cov_linear(
     a = a.reshape(-1, a.shape[-1])
     if bias:
         a = append_bias_ones(a)
-    a32 = a.to(torch.float32)
+    a32 = a.to(out.dtype)
     cov = a32.t() @ a32
     cov = (cov + cov.t()).mul_(0.5 * coeff)
     if beta == 0.0:
@@ -129,7 +129,7 @@ def precond_eigen(
     v1 = QG^T @ grad @ QA; v2 = v1 * dGdA (or v1 / (outer(dG, dA)+damping));
     out = QG @ v2 @ QA^T, cast back to grad dtype.
     """
-    g32 = grad.to(torch.float32)
+    g32 = grad.to(qa.dtype)
     v1 = qg.t() @ g32 @ qa
     if dgda is not None:
         v2 = v1 * dgda
@@ -145,10 +145,8 @@ def precond_inverse(
     g_inv: torch.Tensor,
 ) -> torch.Tensor:
     """Explicit-inverse preconditioning: G^-1 @ grad @ A^-1 (inverse.py:215-234)."""
-    g32 = grad.to(torch.float32)
-    return (g_inv.to(torch.float32) @ g32 @ a_inv.to(torch.float32)).to(
-        grad.dtype,
-    )
+    g32 = grad.to(g_inv.dtype)
+    return (g_inv @ g32 @ a_inv.to(g_inv.dtype)).to(grad.dtype)
 
 
 def eigh(x: torch.Tensor, *, clamp: bool = True) -> tuple[torch.Tensor, torch.Tensor]:
@@ -156,7 +154,8 @@ def eigh(x: torch.Tensor, *, clamp: bool = True) -> tuple[torch.Tensor, torch.Te
 
     Returns (d, Q) with eigenvalues ascending; eigenvalues clamped >= 0.
     """
-    d, q = torch.linalg.eigh(x.to(torch.float32))
+    dt = torch.float64 if x.dtype == torch.float64 else torch.float32
+    d, q = torch.linalg.eigh(x.to(dt))
     if clamp:
         d = torch.clamp(d, min=0.0)
     # rocSOLVER can return the eigenvector matrix as a transposed view;
@@ -169,17 +168,14 @@ def inv_damped(x: torch.Tensor, damping: float) -> torch.Tensor:
 
     Falls back to LU inverse if the damped matrix is not positive definite.
     """
-    x32 = x.to(torch.float32)
-    d = torch.diagonal(x32)
-    xd = x32.clone()
+    dt = torch.float64 if x.dtype == torch.float64 else torch.float32
+    xd = x.to(dt).clone()
     torch.diagonal(xd).add_(damping)
     try:
         chol = torch.linalg.cholesky(xd)
         return torch.cholesky_inverse(chol)
     except Exception:  # singular / not PD: LU fallback
         return torch.linalg.inv(xd)
-    finally:
-        del d
 
 
 def triu_pack(x: torch.Tensor) -> torch.Tensor:
