@@ -7,6 +7,7 @@ from __future__ import annotations
 
 import sys
 
+import pytest
 import torch
 import torch.distributed as dist
 
@@ -175,3 +176,30 @@ def test_bucketed_wait_without_flush() -> None:
 
 def test_allreduce_world1() -> None:
     _allreduce_world1_passthrough()
+
+
+def _bucket_invariants_body() -> None:
+    from kfac_amd.distributed import _Bucket
+
+    b = _Bucket(1024)
+    t = torch.ones(4)
+    b.append(t)
+    b.communicate(None, 1.0)
+    with pytest.raises(RuntimeError):
+        b.communicate(None, 1.0)
+    with pytest.raises(RuntimeError):
+        b.append(torch.ones(2))
+    b.wait_and_unpack()
+
+    b2 = _Bucket(1024)
+    b2.append(torch.ones(4))
+    with pytest.raises(RuntimeError):
+        b2.wait_and_unpack()
+
+
+def test_bucket_invariants() -> None:
+    """Misuse of a bucket raises (single-flight discipline,
+    reference distributed.py:89-95,180-188)."""
+    from testing.distributed import run_distributed
+
+    run_distributed(1, _bucket_invariants_body)
