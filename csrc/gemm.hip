@@ -195,6 +195,9 @@ struct PrecondDesc {
   float* s2;
   float* out;
   long tile_off;  // first global tile index of this layer
+  // fully-fused path (gather/scatter directly on module gradients)
+  float* wgrad;   // weight grad storage [m][n - has_bias] (fp32)
+  float* bgrad;   // bias grad [m] or nullptr
 };
 
 template <int STAGE>
@@ -232,6 +235,102 @@ __global__ __launch_bounds__(256) void grouped_precond_kernel(
         d.out, d.s1, d.qa, m, n, n, false, true, nullptr, nullptr, 0.f, i0,
         j0, lds_a, lds_b);
   }
+}
+
+// Gather module grads into the flat fp32 matrix (bias as last column),
+// or scatter the preconditioned result back scaled by *scale.
+template <bool SCATTER>
+__global__ __launch_bounds__(256) void grouped_grad_copy_kernel(
+    const PrecondDesc* __restrict__ desc,
+    int n_layers,
+    const float* __restrict__ scale) {
+  const int tile = blockIdx.x;
+  int l = 0;
+  while (l + 1 < n_layers && tile >= (int)desc[l + 1].tile_off) ++l;
+  const PrecondDesc d = desc[l];
+  const int m = (int)d.m;
+  const int n = (int)d.n;
+  const int local = tile - (int)d.tile_off;
+  const int ntj = ceil_div(n, GBT);
+  const int i0 = (local / ntj) * GBT;
+  const int j0 = (local % ntj) * GBT;
+  const int n_w = (d.bgrad != nullptr) ? n - 1 : n;
+  const float sc = SCATTER ? *scale : 0.0f;
+  for (int e = threadIdx.x; e < GBT * GBT; e += 256) {
+    const int row = i0 + e / GBT;
+    const int col = j0 + e % GBT;
+    if (row >= m || col >= n) continue;
+    if constexpr (SCATTER) {
+      const float v = d.out[(long)row * n + col] * sc;
+      if (col < n_w) {
+        d.wgrad[(long)row * n_w + col] = v;
+      } else {
+        d.bgrad[row] = v;
+      }
+    } else {
+      float v;
+      if (col < n_w) {
+        v = d.wgrad[(long)row * n_w + col];
+      } else {
+        v = d.bgrad[row];
+      }
+      // grad buffer doubles as stage-1 B operand
+      ((float*)d.grad)[(long)row * n + col] = v;
+    }
+  }
+}
+
+// scale = min(1, sqrt(kl_clip / |accum * lr^2|)); kl_clip <= 0 -> 1.
+__global__ void grad_scale_kernel(
+    float* __restrict__ scale,
+    const float* __restrict__ accum,
+    float kl_clip,
+    float lr) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    float s = 1.0f;
+    if (kl_clip > 0.0f) {
+      float vg = fabsf(*accum * lr * lr);
+      if (vg > 0.0f) {
+        s = fminf(1.0f, sqrtf(kl_clip / vg));
+      }
+    }
+    *scale = s;
+  }
+}
+
+__global__ void zero_kernel(float* p) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *p = 0.0f;
+}
+
+template <typename T>
+hipError_t kl_clip_accum_t(hipStream_t, float*, const T*, const T*, long);
+
+hipError_t precond_apply_grouped_f32(
+    hipStream_t stream,
+    const void* desc_dev,
+    int n_layers,
+    int total_tiles,
+    float* gbuf,     // flat gathered grads (== concatenated desc.grad)
+    float* outbuf,   // flat outputs (== concatenated desc.out)
+    long total_elems,
+    float* accum,    // 1-elem workspace
+    float* scale,    // 1-elem output
+    float kl_clip,
+    float lr) {
+  auto desc = (const PrecondDesc*)desc_dev;
+  grouped_grad_copy_kernel<false>
+      <<<total_tiles, 256, 0, stream>>>(desc, n_layers, nullptr);
+  grouped_precond_kernel<1><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  grouped_precond_kernel<2><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  grouped_precond_kernel<3><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  grouped_precond_kernel<4><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
+  zero_kernel<<<1, 1, 0, stream>>>(accum);
+  KFAC_HIP_CHECK(
+      kl_clip_accum_t<float>(stream, accum, outbuf, gbuf, total_elems));
+  grad_scale_kernel<<<1, 1, 0, stream>>>(scale, accum, kl_clip, lr);
+  grouped_grad_copy_kernel<true>
+      <<<total_tiles, 256, 0, stream>>>(desc, n_layers, scale);
+  return hipGetLastError();
 }
 
 hipError_t precond_grouped_f32(

@@ -199,6 +199,27 @@ def precond_eigen_grouped(
     return ext.precond_eigen_grouped(grads, qas, qgs, dgdas)
 
 
+def precond_apply_grouped(
+    weight_grads: list[torch.Tensor],
+    bias_grads: list[torch.Tensor],
+    qas: list[torch.Tensor],
+    qgs: list[torch.Tensor],
+    dgdas: list[torch.Tensor],
+    kl_clip: float,
+    lr: float,
+) -> torch.Tensor:
+    """Fused COMM-OPT precondition + kl-clip + in-place grad update.
+
+    ~9 kernel launches for the whole model; returns the applied scale
+    (1-elem device tensor). GPU-only.
+    """
+    ext = _require_ext('precond_apply_grouped')
+    assert ext is not None
+    return ext.precond_apply_grouped(
+        weight_grads, bias_grads, qas, qgs, dgdas, kl_clip, lr,
+    )
+
+
 def precond_inverse(
     grad: torch.Tensor,
     a_inv: torch.Tensor,

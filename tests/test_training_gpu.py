@@ -140,3 +140,40 @@ def test_async_inverse_pipeline() -> None:
     # no job left hanging mid-train
     if precon._async_job is not None:
         precon._finish_async_inverses()
+
+
+def test_fused_precondition_matches_general_path() -> None:
+    """COMM-OPT fused path == general per-layer path + kl-clip scaling."""
+    from kfac_amd import KFACPreconditioner
+    from testing.models import LeNet
+
+    torch.manual_seed(123)
+    x0 = torch.randn(32, 1, 28, 28)
+    y0 = torch.randint(0, 10, (32,))
+    results = {}
+    for fused in (True, False):
+        torch.manual_seed(123)
+        model = LeNet().cuda()
+        precon = KFACPreconditioner(
+            model, factor_update_steps=1, inv_update_steps=1, lr=0.01,
+        )
+        if not fused:
+            # force the general path
+            precon._fused_precondition_update = lambda: False
+        loss = torch.nn.functional.cross_entropy(
+            model(x0.cuda()), y0.cuda(),
+        )
+        loss.backward()
+        precon.step()
+        results[fused] = {
+            n: p.grad.detach().cpu().clone()
+            for n, p in model.named_parameters()
+        }
+    for name in results[True]:
+        torch.testing.assert_close(
+            results[True][name],
+            results[False][name],
+            rtol=1e-4,
+            atol=1e-6,
+            msg=lambda m: f'{name}: {m}',
+        )
