@@ -56,6 +56,8 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--seed', type=int, default=42)
     p.add_argument('--max-steps-per-epoch', type=int, default=None)
     p.add_argument('--backend', type=str, default=None, choices=['nccl', 'gloo'])
+    p.add_argument('--log-dir', type=str, default=None,
+                   help='TensorBoard log dir (requires tensorboard)')
     # K-FAC flags (parity with reference :148-237)
     p.add_argument('--kfac-inv-update-steps', type=int, default=10,
                    help='steps between inverse updates (0 disables K-FAC)')
@@ -125,6 +127,15 @@ def main() -> None:
     optimizer, preconditioner, schedulers = optimizers.get_optimizer(model, args)
     loss_func = LabelSmoothLoss(args.label_smoothing)
 
+    writer = None
+    if args.log_dir is not None:
+        try:
+            from torch.utils.tensorboard import SummaryWriter
+
+            writer = SummaryWriter(args.log_dir)
+        except ImportError:
+            print('tensorboard not installed; skipping TB logging')
+
     # resume from the newest checkpoint (reference :313-317)
     start_epoch = 0
     os.makedirs(args.checkpoint_dir, exist_ok=True)
@@ -155,10 +166,16 @@ def main() -> None:
         )
         for s in schedulers:
             s.step()
-        engine.validate(
+        val_loss, val_acc = engine.validate(
             epoch, model, loss_func, val_loader, device,
             max_steps=args.max_steps_per_epoch,
         )
+        if writer is not None:
+            writer.add_scalar('val/loss', float(val_loss.avg), epoch)
+            writer.add_scalar('val/acc', float(val_acc.avg), epoch)
+            writer.add_scalar(
+                'train/lr', optimizer.param_groups[0]['lr'], epoch,
+            )
         rank = dist.get_rank() if world > 1 else 0
         if rank == 0 and (epoch + 1) % args.checkpoint_freq == 0:
             save_checkpoint(

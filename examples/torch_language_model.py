@@ -9,10 +9,8 @@ Offline image -> synthetic token streams by default.
 from __future__ import annotations
 
 import argparse
-import math
 import os
 import sys
-import time
 
 import torch
 import torch.distributed as dist
@@ -20,6 +18,8 @@ import torch.distributed as dist
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import kfac_amd  # noqa: E402
+from examples.language.dataset import synthetic_batch  # noqa: E402
+from examples.language.engine import run_training  # noqa: E402
 from kfac_amd.models import TransformerModel  # noqa: E402
 
 
@@ -57,14 +57,6 @@ def parse_args() -> argparse.Namespace:
     return p.parse_args()
 
 
-def synthetic_batch(
-    vocab: int, batch: int, seq: int, device: torch.device, seed: int,
-) -> tuple[torch.Tensor, torch.Tensor]:
-    g = torch.Generator().manual_seed(seed)
-    data = torch.randint(0, vocab, (seq + 1, batch), generator=g)
-    return data[:-1].to(device), data[1:].reshape(-1).to(device)
-
-
 def main() -> None:
     args = parse_args()
     world = int(os.environ.get('WORLD_SIZE', '1'))
@@ -100,35 +92,24 @@ def main() -> None:
             lr=lambda x: optimizer.param_groups[0]['lr'],
             skip_layers=args.kfac_skip_layers,
         )
-    criterion = torch.nn.CrossEntropyLoss()
-
     rank = dist.get_rank() if world > 1 else 0
-    step = 0
-    for epoch in range(args.epochs):
-        model.train()
-        t0 = time.time()
-        total_loss = 0.0
-        for i in range(args.steps_per_epoch):
-            data, target = synthetic_batch(
-                args.vocab, args.batch_size, args.seq_len, device,
-                seed=step * world + rank,
-            )
-            optimizer.zero_grad()
-            output = model(data)
-            loss = criterion(output.view(-1, args.vocab), target)
-            loss.backward()
-            torch.nn.utils.clip_grad_norm_(model.parameters(), 0.25)
-            if preconditioner is not None:
-                preconditioner.step()
-            optimizer.step()
-            total_loss += loss.item()
-            step += 1
-        if rank == 0:
-            ppl = math.exp(total_loss / args.steps_per_epoch)
-            print(
-                f'epoch {epoch}: ppl={ppl:.2f} '
-                f'({time.time() - t0:.1f}s)',
-            )
+
+    def batch_fn(i: int) -> tuple[torch.Tensor, torch.Tensor]:
+        return synthetic_batch(
+            args.vocab, args.batch_size, args.seq_len, device,
+            seed=i * world + rank,
+        )
+
+    run_training(
+        model,
+        optimizer,
+        preconditioner,
+        batch_fn,
+        args.epochs,
+        args.steps_per_epoch,
+        args.vocab,
+        rank=rank,
+    )
 
     if world > 1:
         dist.destroy_process_group()
