@@ -291,3 +291,49 @@ def test_reset_batch() -> None:
     assert any(l._a_count > 0 for _, l in p._layers.values())
     p.reset_batch()
     assert all(l._a_count == 0 for _, l in p._layers.values())
+
+
+def test_gradient_accumulation_factors() -> None:
+    """accumulation_steps=2: factor = EMA over the mean of 2 minibatches."""
+    torch.manual_seed(1)
+    lin_model = TinyModel()
+    p = KFACPreconditioner(
+        lin_model, factor_update_steps=1, inv_update_steps=1,
+        accumulation_steps=2, lr=0.01,
+    )
+    x1 = torch.randn(8, 10)
+    x2 = torch.randn(8, 10)
+    y = torch.randint(0, 3, (8,))
+    for x in (x1, x2):
+        torch.nn.functional.cross_entropy(lin_model(x), y).backward()
+    # after 2 mini-steps the hook-boundary fired: factors EMA'd
+    name, layer = next(iter(p._layers.values()))
+    a = layer.a_factor
+    assert a is not None
+    # manual: mean of the two minibatch covariances, identity-init EMA
+    xs = [torch.cat([x, x.new_ones(8, 1)], dim=1) for x in (x1, x2)]
+    mean_cov = sum(xb.t() @ xb / 8 for xb in xs) / 2
+    expected = 0.95 * torch.eye(11) + 0.05 * mean_cov
+    torch.testing.assert_close(a, expected, rtol=1e-5, atol=1e-6)
+    p.step()
+
+
+def test_update_factors_in_hook_equivalence() -> None:
+    """update_factors_in_hook True and False produce identical grads."""
+    results = {}
+    for in_hook in (True, False):
+        torch.manual_seed(5)
+        model = TinyModel()
+        p = KFACPreconditioner(
+            model, factor_update_steps=1, inv_update_steps=1, lr=0.01,
+            update_factors_in_hook=in_hook,
+        )
+        x = torch.randn(16, 10)
+        y = torch.randint(0, 3, (16,))
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        p.step()
+        results[in_hook] = {
+            n: prm.grad.clone() for n, prm in model.named_parameters()
+        }
+    for n in results[True]:
+        torch.testing.assert_close(results[True][n], results[False][n])
