@@ -17,7 +17,12 @@ namespace kfac {
 
 constexpr int GBT = 128;
 constexpr int GBK = 32;
-constexpr int GLDS = GBT + 4;  // +4 keeps float4 LDS rows 16B-aligned
+// Unpadded rows: required for global_load_lds (lane-linear destination).
+// Bank-conflict cost of the power-of-2 stride is minor here: the b32
+// MFMA-loop reads go 2-way on half-wave groups and the trans-path writes
+// 8-way on 4 instructions — tens of cycles against the ~4096-cycle f32
+// MFMA issue time per K-slice.
+constexpr int GLDS = GBT;
 
 enum class Epilogue : int { NONE = 0, MUL = 1, DIV_OUTER = 2 };
 
@@ -36,6 +41,29 @@ __device__ __forceinline__ void stage_gemm(
     int i0,
     float (*lds)[GLDS],
     int tid) {
+  if (!trans && (ld & 3) == 0 && i0 + GBT <= rows && k0 + GBK <= ks) {
+    // Full interior tile, 16-byte-aligned rows: async global->LDS DMA.
+    // Each wave chunk is 2 LDS rows (64 lanes x 16 B = 1 KiB); the
+    // per-lane SOURCE address mirrors the lane-linear LDS layout.
+    const int wv = tid >> 6;
+    const int l = tid & 63;
+    // 1 KiB chunks (2 rows): GBK*GBT floats / 256 per chunk, 4 waves
+    constexpr int kChunks = (GBK * GBT) / 256;        // 16
+    constexpr int kPerWave = kChunks / 4;             // 4
+#pragma unroll
+    for (int e = 0; e < kPerWave; ++e) {
+      const int ci = wv * kPerWave + e;
+      const float* gp =
+          src + (long)(k0 + 2 * ci + (l >> 5)) * ld + (i0 + (l & 31) * 4);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gp,
+          (__attribute__((address_space(3))) void*)&lds[2 * ci][0],
+          16,
+          0,
+          0);
+    }
+    return;
+  }
 #pragma unroll
   for (int e = 0; e < (GBK * GBT) / (256 * 4); ++e) {
     const int c = tid + e * 256;
