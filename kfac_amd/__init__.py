@@ -15,6 +15,7 @@ from kfac_amd import assignment
 from kfac_amd import base_preconditioner
 from kfac_amd import distributed
 from kfac_amd import enums
+from kfac_amd import gpt_neox
 from kfac_amd import hyperparams
 from kfac_amd import layers
 from kfac_amd import ops
@@ -32,6 +33,7 @@ __all__ = [
     'base_preconditioner',
     'distributed',
     'enums',
+    'gpt_neox',
     'hyperparams',
     'layers',
     'ops',

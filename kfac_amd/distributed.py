@@ -34,6 +34,26 @@ class NonSquareTensorError(Exception):
     """Raised when a symmetric op receives a non-square tensor."""
 
 
+def get_triu(tensor: torch.Tensor) -> torch.Tensor:
+    """Upper-triangle wire format (reference distributed.py:422-446)."""
+    if tensor.dim() != 2 or tensor.size(0) != tensor.size(1):
+        raise NonSquareTensorError(
+            f'get_triu requires a square tensor, got {tuple(tensor.shape)}',
+        )
+    return ops.triu_pack(tensor)
+
+
+def fill_triu(
+    shape: tuple[int, ...],
+    triu_tensor: torch.Tensor,
+) -> torch.Tensor:
+    """Rebuild the symmetric matrix from its packed upper triangle
+    (reference distributed.py:448-465)."""
+    if len(shape) != 2 or shape[0] != shape[1]:
+        raise NonSquareTensorError(f'fill_triu requires a square shape, got {shape}')
+    return ops.triu_unpack(triu_tensor, shape[0])
+
+
 def get_rank(group: dist.ProcessGroup | None = None) -> int:
     """Rank of this process (0 if torch.distributed is uninitialized)."""
     if dist.is_available() and dist.is_initialized():
@@ -78,7 +98,7 @@ class Future:
         return self._result
 
 
-class _Bucket:
+class AllreduceTensorBucket:
     """One flat allreduce bucket for a single (group, dtype, device).
 
     Tensors are packed into one flat buffer, allreduced together, and
@@ -93,6 +113,19 @@ class _Bucket:
         self._flat: torch.Tensor | None = None
         self._work: dist.Work | None = None
         self._unpacked = False
+
+    @property
+    def size(self) -> int:
+        """Bytes currently in the bucket (reference distributed.py:60)."""
+        return self._bytes
+
+    def communicated(self) -> bool:
+        """True once the bucket's allreduce has been launched."""
+        return self._flat is not None
+
+    def add_tensor(self, tensor: torch.Tensor) -> int:
+        """Reference-compatible alias of append()."""
+        return self.append(tensor)
 
     def fits(self, tensor: torch.Tensor) -> bool:
         return self._bytes + tensor.numel() * tensor.element_size() <= self._cap
@@ -141,11 +174,11 @@ class _BucketFuture:
         self._tensor = tensor
         self._comm = comm
         self._key = key
-        self._bucket: _Bucket | None = None
+        self._bucket: AllreduceTensorBucket | None = None
         self._post = post
         self._done = False
 
-    def _attach(self, bucket: _Bucket) -> None:
+    def _attach(self, bucket: AllreduceTensorBucket) -> None:
         self._bucket = bucket
 
     def wait(self) -> torch.Tensor:
@@ -177,11 +210,27 @@ class TorchDistributedCommunicator:
         """
         self._cap_bytes = int(bucket_cap_mb * 1024 * 1024)
         # key -> (open bucket, pending futures not yet attached)
-        self._open: dict[tuple, _Bucket] = {}
+        self._open: dict[tuple, AllreduceTensorBucket] = {}
         self._pending: dict[tuple, list[_BucketFuture]] = defaultdict(list)
         self._inflight: list[_Bucket] = []
         # group handle + averaging flag per bucket key
         self._groups: dict[tuple, tuple] = {}
+
+    @property
+    def bucket_cap_bytes(self) -> int:
+        """Flat-bucket size cap in bytes."""
+        return self._cap_bytes
+
+    def group_ranks(self, group: dist.ProcessGroup | None) -> frozenset[int]:
+        """Membership of a process group.
+
+        Keyed by actual rank membership — the reference's version keyed
+        by group SIZE (distributed.py:376-378), silently sharing buckets
+        between distinct equal-sized groups.
+        """
+        if group is None or not (dist.is_available() and dist.is_initialized()):
+            return frozenset()
+        return frozenset(dist.get_process_group_ranks(group))
 
     # -- plain collectives -------------------------------------------------
 
@@ -287,7 +336,7 @@ class TorchDistributedCommunicator:
             self._launch(key, group, average)
             bucket = None
         if bucket is None:
-            bucket = _Bucket(self._cap_bytes)
+            bucket = AllreduceTensorBucket(self._cap_bytes)
             self._open[key] = bucket
             self._groups[key] = (group, average)
         bucket.append(payload)

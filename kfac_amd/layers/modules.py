@@ -65,6 +65,28 @@ class ModuleHelper:
         """out = beta*out + coeff_scale * G(g)."""
         raise NotImplementedError
 
+    def get_a_factor(self, a: torch.Tensor) -> torch.Tensor:
+        """Return this minibatch's A-factor contribution (allocating form
+        of accumulate_a_factor; reference modules.py:47-50 interface)."""
+        out = torch.zeros(
+            self.a_factor_shape,
+            dtype=torch.float32,
+            device=a.device,
+        )
+        self.accumulate_a_factor(a, out, 0.0, 1.0)
+        return out
+
+    def get_g_factor(self, g: torch.Tensor) -> torch.Tensor:
+        """Return this minibatch's G-factor contribution (allocating form
+        of accumulate_g_factor)."""
+        out = torch.zeros(
+            self.g_factor_shape,
+            dtype=torch.float32,
+            device=g.device,
+        )
+        self.accumulate_g_factor(g, out, 0.0, 1.0)
+        return out
+
     def get_grad(self) -> torch.Tensor:
         """Combined (out, in[+1]) gradient matrix, bias as last column.
 

@@ -179,7 +179,7 @@ def test_allreduce_world1() -> None:
 
 
 def _bucket_invariants_body() -> None:
-    from kfac_amd.distributed import _Bucket
+    from kfac_amd.distributed import AllreduceTensorBucket as _Bucket
 
     b = _Bucket(1024)
     t = torch.ones(4)
@@ -192,7 +192,9 @@ def _bucket_invariants_body() -> None:
     b.wait_and_unpack()
 
     b2 = _Bucket(1024)
-    b2.append(torch.ones(4))
+    assert b2.size == 0
+    b2.add_tensor(torch.ones(4))
+    assert b2.size == 16 and not b2.communicated()
     with pytest.raises(RuntimeError):
         b2.wait_and_unpack()
 

@@ -166,3 +166,31 @@ def test_grad_scale_from_accum() -> None:
     # zero accum -> 1 (matches reference vg_sum==0 -> 1.0)
     s = ops.grad_scale_from_accum(torch.tensor(0.0), kl_clip=1.0, lr=1.0)
     torch.testing.assert_close(s, torch.tensor(1.0))
+
+
+def test_get_triu_fill_triu_module_api() -> None:
+    from kfac_amd.distributed import fill_triu, get_triu
+
+    n = 7
+    x = torch.randn(n, n)
+    x = (x + x.t()) / 2
+    v = get_triu(x)
+    y = fill_triu((n, n), v)
+    torch.testing.assert_close(x, y)
+
+
+def test_module_helper_get_factor_api() -> None:
+    from kfac_amd.layers.modules import LinearModuleHelper
+    from kfac_amd.layers.utils import get_cov
+    from kfac_amd.ops.reference import append_bias_ones
+
+    lin = torch.nn.Linear(5, 3)
+    h = LinearModuleHelper(lin)
+    a = torch.randn(12, 5)
+    fa = h.get_a_factor(a)
+    torch.testing.assert_close(
+        fa, get_cov(append_bias_ones(a)), rtol=1e-5, atol=1e-6,
+    )
+    g = torch.randn(12, 3)
+    fg = h.get_g_factor(g)
+    torch.testing.assert_close(fg, get_cov(g), rtol=1e-5, atol=1e-6)
