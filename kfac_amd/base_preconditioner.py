@@ -551,10 +551,21 @@ class BaseKFACPreconditioner:
             except Exception as e:  # pragma: no cover - surfaced at join
                 job['error'] = e
 
-        thread = threading.Thread(target=worker, daemon=True)
+        # Non-daemon: Python joins it before interpreter teardown, so a
+        # pending eigendecomposition can never race HIP/c10 destruction
+        # at process exit (daemon threads caused std::terminate there).
+        thread = threading.Thread(target=worker, daemon=False)
         job['thread'] = thread
         thread.start()
         self._async_job = job
+
+    def __del__(self) -> None:
+        job = getattr(self, '_async_job', None)
+        if job is not None:
+            try:
+                job['thread'].join(timeout=60)
+            except Exception:
+                pass
 
     def _finish_async_inverses(self) -> None:
         job = self._async_job
