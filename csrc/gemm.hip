@@ -26,6 +26,16 @@ constexpr int GLDS = GBT;
 
 enum class Epilogue : int { NONE = 0, MUL = 1, DIV_OUTER = 2 };
 
+// LDS column swizzle at 4-float (16 B) granularity: distributes the
+// trans-path writes (8 lanes sharing a column index across k-rows,
+// measured 2.8e9 SQ_LDS_BANK_CONFLICT cycles on the stage-4 GEMM) over
+// distinct banks, stays 16-byte aligned for float4 writes and glds, and
+// is its own inverse so the glds SOURCE pre-swizzle and the read-side
+// XOR are the same involution (guide rule 21).
+__device__ __forceinline__ int lds_swz(int k, int i) {
+  return i ^ (((k >> 2) & 7) << 2);
+}
+
 // Stage op(A)'s [k, rows i0..i0+GBT) slab into lds[k][i].
 // The thread->element mapping follows the PHYSICAL memory order so loads
 // coalesce either way: each thread moves 4 consecutive floats of the
@@ -53,8 +63,9 @@ __device__ __forceinline__ void stage_gemm(
 #pragma unroll
     for (int e = 0; e < kPerWave; ++e) {
       const int ci = wv * kPerWave + e;
+      const int krow = 2 * ci + (l >> 5);
       const float* gp =
-          src + (long)(k0 + 2 * ci + (l >> 5)) * ld + (i0 + (l & 31) * 4);
+          src + (long)(k0 + krow) * ld + (i0 + lds_swz(krow, (l & 31) * 4));
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)gp,
           (__attribute__((address_space(3))) void*)&lds[2 * ci][0],
@@ -83,7 +94,7 @@ __device__ __forceinline__ void stage_gemm(
         }
       }
 #pragma unroll
-      for (int q = 0; q < 4; ++q) lds[k + q][i] = v[q];
+      for (int q = 0; q < 4; ++q) lds[k + q][lds_swz(k + q, i)] = v[q];
     } else {
       // contiguous along i in memory: 32 threads cover one 128-i row
       const int i = (c & 31) * 4;
@@ -99,7 +110,7 @@ __device__ __forceinline__ void stage_gemm(
           v[q] = (gk < ks && i0 + i + q < rows) ? src[base + q] : 0.0f;
         }
       }
-      *(float4*)&lds[k][i] = make_float4(v[0], v[1], v[2], v[3]);
+      *(float4*)&lds[k][lds_swz(k, i)] = make_float4(v[0], v[1], v[2], v[3]);
     }
   }
 }
@@ -145,8 +156,8 @@ __device__ __forceinline__ void gemm_tile_body(
       float bv[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        av[f] = lds_a[krow][wr * 64 + f * 16 + (lane & 15)];
-        bv[f] = lds_b[krow][wc * 64 + f * 16 + (lane & 15)];
+        av[f] = lds_a[krow][lds_swz(krow, wr * 64 + f * 16 + (lane & 15))];
+        bv[f] = lds_b[krow][lds_swz(krow, wc * 64 + f * 16 + (lane & 15))];
       }
 #pragma unroll
       for (int fi = 0; fi < 4; ++fi) {
