@@ -313,8 +313,9 @@ struct PrecondDescHost {
   int64_t tile_off;
   float* wgrad;
   float* bgrad;
+  int64_t spad;
 };
-static_assert(sizeof(PrecondDescHost) == 96, "descriptor layout");
+static_assert(sizeof(PrecondDescHost) == 104, "descriptor layout");
 
 std::vector<torch::Tensor> precond_eigen_grouped(
     std::vector<torch::Tensor> grads,
@@ -372,6 +373,7 @@ std::vector<torch::Tensor> precond_eigen_grouped(
     d[l].tile_off = tile_offs[l];
     d[l].wgrad = nullptr;
     d[l].bgrad = nullptr;
+    d[l].spad = g32[l].size(1);
   }
   auto desc_dev = desc_cpu.to(grads[0].device(), /*non_blocking=*/true);
   auto stream = current_stream(grads[0]);
@@ -436,13 +438,15 @@ torch::Tensor precond_apply_grouped(
     ns[l] = n;
     offsets[l] = total;
     tile_offs[l] = tiles;
-    total += m * n;
+    total += m * ((n + 3) / 4 * 4);
     tiles += (int64_t)((m + 127) / 128) * ((n + 127) / 128);
   }
-  auto gbuf = torch::empty({total}, dev_opts);
+  // grad/out buffers are zero-initialized: the row padding (spad > n)
+  // participates in the flat kl-clip dot product and must contribute 0.
+  auto gbuf = torch::zeros({total}, dev_opts);
   auto s1 = torch::empty({total}, dev_opts);
   auto s2 = torch::empty({total}, dev_opts);
-  auto outbuf = torch::empty({total}, dev_opts);
+  auto outbuf = torch::zeros({total}, dev_opts);
   auto work = torch::empty({2}, dev_opts);  // [accum, scale]
 
   auto desc_cpu = torch::empty(
@@ -463,6 +467,7 @@ torch::Tensor precond_apply_grouped(
     d[l].wgrad = wgrads[l].data_ptr<float>();
     d[l].bgrad =
         bgrads[l].numel() > 0 ? bgrads[l].data_ptr<float>() : nullptr;
+    d[l].spad = (ns[l] + 3) / 4 * 4;
   }
   auto desc_dev = desc_cpu.to(wgrads[0].device(), /*non_blocking=*/true);
   auto stream = current_stream(wgrads[0]);

@@ -128,8 +128,11 @@ __device__ __forceinline__ void gemm_tile_body(
     int M,
     int N,
     int K,
-    bool ta,  // op(A)[m][k]; ta=false: A[m][k] (ld=K); ta=true: A[k][m] (ld=M)
-    bool tb,  // op(B)[k][n]; tb=false: B[k][n] (ld=N); tb=true: B[n][k] (ld=K)
+    bool ta,  // op(A)[m][k]; ta=false: A[m][k]; ta=true: A[k][m]
+    bool tb,  // op(B)[k][n]; tb=false: B[k][n]; tb=true: B[n][k]
+    long lda,  // physical leading dim of a (may exceed the logical dim
+    long ldb,  // for padded scratch buffers -> keeps the glds path alive)
+    long ldc,
     const float* __restrict__ e1,  // MUL: dgda [M x N]; DIV_OUTER: dg [M]
     const float* __restrict__ e2,  // DIV_OUTER: da [N]
     float damping,
@@ -146,8 +149,8 @@ __device__ __forceinline__ void gemm_tile_body(
   f32x4 acc[4][4] = {};
 
   for (int k0 = 0; k0 < K; k0 += GBK) {
-    stage_gemm(a, M, K, ta ? (long)M : (long)K, !ta, k0, i0, lds_a, tid);
-    stage_gemm(b, N, K, tb ? (long)K : (long)N, tb, k0, j0, lds_b, tid);
+    stage_gemm(a, M, K, lda, !ta, k0, i0, lds_a, tid);
+    stage_gemm(b, N, K, ldb, tb, k0, j0, lds_b, tid);
     __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < GBK; kk += 4) {
@@ -186,7 +189,7 @@ __device__ __forceinline__ void gemm_tile_body(
           } else if constexpr (EPI == Epilogue::DIV_OUTER) {
             v /= (e1[row] * e2[col] + damping);
           }
-          c[(long)row * N + col] = v;
+          c[(long)row * ldc + col] = v;
         }
       }
     }
@@ -209,8 +212,9 @@ __global__ __launch_bounds__(256) void gemm_kernel(
   __shared__ float lds_a[GBK][GLDS];
   __shared__ float lds_b[GBK][GLDS];
   gemm_tile_body<EPI>(
-      c, a, b, M, N, K, ta, tb, e1, e2, damping, (int)blockIdx.x * GBT,
-      (int)blockIdx.y * GBT, lds_a, lds_b);
+      c, a, b, M, N, K, ta, tb, ta ? (long)M : (long)K,
+      tb ? (long)K : (long)N, (long)N, e1, e2, damping,
+      (int)blockIdx.x * GBT, (int)blockIdx.y * GBT, lds_a, lds_b);
 }
 
 // ---------------------------------------------------- grouped precondition
@@ -237,6 +241,8 @@ struct PrecondDesc {
   // fully-fused path (gather/scatter directly on module gradients)
   float* wgrad;   // weight grad storage [m][n - has_bias] (fp32)
   float* bgrad;   // bias grad [m] or nullptr
+  long spad;      // physical row stride of grad/s1/s2/out buffers (>= n,
+                  // multiple of 4 so staging keeps the glds fast path)
 };
 
 template <int STAGE>
@@ -257,22 +263,23 @@ __global__ __launch_bounds__(256) void grouped_precond_kernel(
   __shared__ float lds_a[GBK][GLDS];
   __shared__ float lds_b[GBK][GLDS];
 
+  const long sp = d.spad;
   if constexpr (STAGE == 1) {
     gemm_tile_body<Epilogue::NONE>(
-        d.s1, d.qg, d.grad, m, n, m, true, false, nullptr, nullptr, 0.f, i0,
-        j0, lds_a, lds_b);
+        d.s1, d.qg, d.grad, m, n, m, true, false, (long)m, sp, sp, nullptr,
+        nullptr, 0.f, i0, j0, lds_a, lds_b);
   } else if constexpr (STAGE == 2) {
     gemm_tile_body<Epilogue::MUL>(
-        d.s2, d.s1, d.qa, m, n, n, false, false, d.dgda, nullptr, 0.f, i0,
-        j0, lds_a, lds_b);
+        d.s2, d.s1, d.qa, m, n, n, false, false, sp, (long)n, sp, d.dgda,
+        nullptr, 0.f, i0, j0, lds_a, lds_b);
   } else if constexpr (STAGE == 3) {
     gemm_tile_body<Epilogue::NONE>(
-        d.s1, d.qg, d.s2, m, n, m, false, false, nullptr, nullptr, 0.f, i0,
-        j0, lds_a, lds_b);
+        d.s1, d.qg, d.s2, m, n, m, false, false, (long)m, sp, sp, nullptr,
+        nullptr, 0.f, i0, j0, lds_a, lds_b);
   } else {
     gemm_tile_body<Epilogue::NONE>(
-        d.out, d.s1, d.qa, m, n, n, false, true, nullptr, nullptr, 0.f, i0,
-        j0, lds_a, lds_b);
+        d.out, d.s1, d.qa, m, n, n, false, true, sp, (long)n, sp, nullptr,
+        nullptr, 0.f, i0, j0, lds_a, lds_b);
   }
 }
 
@@ -300,7 +307,7 @@ __global__ __launch_bounds__(256) void grouped_grad_copy_kernel(
     const int col = j0 + e % GBT;
     if (row >= m || col >= n) continue;
     if constexpr (SCATTER) {
-      const float v = d.out[(long)row * n + col] * sc;
+      const float v = d.out[(long)row * d.spad + col] * sc;
       if (col < n_w) {
         d.wgrad[(long)row * n_w + col] = v;
       } else {
@@ -314,7 +321,7 @@ __global__ __launch_bounds__(256) void grouped_grad_copy_kernel(
         v = d.bgrad[row];
       }
       // grad buffer doubles as stage-1 B operand
-      ((float*)d.grad)[(long)row * n + col] = v;
+      ((float*)d.grad)[(long)row * d.spad + col] = v;
     }
   }
 }
