@@ -37,6 +37,15 @@ constexpr int BSTR = BKB + 8;  // row stride 144 B: 16-lane groups hit 16
                                // distinct banks ((144/4)*j mod 64, gcd 4)
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
+// k-index swizzle for the bf16 [col][m] LDS tiles: rows 8 apart share
+// write banks at the 144 B stride (measured 7.2e8 SQ_LDS_BANK_CONFLICT
+// on the LinearAcc SYRK); XOR-ing k by a per-row constant in 16-bf16
+// (32 B) units keeps every 8-element fragment contiguous and 16 B
+// aligned while spreading the colfast scalar writes over banks.
+__device__ __forceinline__ int kswz(int row, int k) {
+  return k ^ (((row >> 3) & 3) << 4);
+}
+
 // ---------------------------------------------------------------- accessors
 
 template <typename T>
@@ -371,7 +380,7 @@ __device__ __forceinline__ void stage_tile_bf16_colfast(
   acc.load16cols_bf16(m0 + k, col0 + i0, vals);
 #pragma unroll
   for (int e = 0; e < 16; ++e) {
-    lds[i0 + e][k] = vals[e];
+    lds[i0 + e][kswz(i0 + e, k)] = vals[e];
   }
 }
 
@@ -389,8 +398,8 @@ __device__ __forceinline__ void stage_tile_bf16(
   const int k0 = (tid & 3) * 16;
   __bf16 vals[16];
   acc.load16_bf16(m0 + k0, col0 + i, vals);
-  *(bf16x8*)&lds[i][k0] = *(const bf16x8*)&vals[0];
-  *(bf16x8*)&lds[i][k0 + 8] = *(const bf16x8*)&vals[8];
+  *(bf16x8*)&lds[i][kswz(i, k0)] = *(const bf16x8*)&vals[0];
+  *(bf16x8*)&lds[i][kswz(i, k0 + 8)] = *(const bf16x8*)&vals[8];
 }
 
 template <typename AccL, typename AccR>
@@ -436,10 +445,14 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16(
 #pragma unroll
     for (int kt = 0; kt < BKB; kt += 32) {
       const int kfrag = kt + (lane >> 4) * 8;
-      bf16x8 a0 = *(const bf16x8*)&lds_l[wr * 32 + (lane & 15)][kfrag];
-      bf16x8 a1 = *(const bf16x8*)&lds_l[wr * 32 + 16 + (lane & 15)][kfrag];
-      bf16x8 b0 = *(const bf16x8*)&rbuf[wc * 32 + (lane & 15)][kfrag];
-      bf16x8 b1 = *(const bf16x8*)&rbuf[wc * 32 + 16 + (lane & 15)][kfrag];
+      const int r0 = wr * 32 + (lane & 15);
+      const int r1 = r0 + 16;
+      const int c0 = wc * 32 + (lane & 15);
+      const int c1 = c0 + 16;
+      bf16x8 a0 = *(const bf16x8*)&lds_l[r0][kswz(r0, kfrag)];
+      bf16x8 a1 = *(const bf16x8*)&lds_l[r1][kswz(r1, kfrag)];
+      bf16x8 b0 = *(const bf16x8*)&rbuf[c0][kswz(c0, kfrag)];
+      bf16x8 b1 = *(const bf16x8*)&rbuf[c1][kswz(c1, kfrag)];
       acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc[1][0], 0, 0, 0);
@@ -488,11 +501,11 @@ __device__ __forceinline__ void stage_tile_bf16_big(
   const int k0 = (tid & 1) * 32;
   __bf16 vals[16];
   acc.load16_bf16(m0 + k0, col0 + i, vals);
-  *(bf16x8*)&lds[i][k0] = *(const bf16x8*)&vals[0];
-  *(bf16x8*)&lds[i][k0 + 8] = *(const bf16x8*)&vals[8];
+  *(bf16x8*)&lds[i][kswz(i, k0)] = *(const bf16x8*)&vals[0];
+  *(bf16x8*)&lds[i][kswz(i, k0 + 8)] = *(const bf16x8*)&vals[8];
   acc.load16_bf16(m0 + k0 + 16, col0 + i, vals);
-  *(bf16x8*)&lds[i][k0 + 16] = *(const bf16x8*)&vals[0];
-  *(bf16x8*)&lds[i][k0 + 24] = *(const bf16x8*)&vals[8];
+  *(bf16x8*)&lds[i][kswz(i, k0 + 16)] = *(const bf16x8*)&vals[0];
+  *(bf16x8*)&lds[i][kswz(i, k0 + 24)] = *(const bf16x8*)&vals[8];
 }
 
 template <typename Acc>
@@ -507,10 +520,12 @@ __device__ __forceinline__ void stage_tile_bf16_big_colfast(
   __bf16 vals[16];
   acc.load16cols_bf16(m0 + k, col0 + i0, vals);
 #pragma unroll
-  for (int e = 0; e < 16; ++e) lds[i0 + e][k] = vals[e];
+  for (int e = 0; e < 16; ++e) lds[i0 + e][kswz(i0 + e, k)] = vals[e];
   acc.load16cols_bf16(m0 + k, col0 + i0 + 16, vals);
 #pragma unroll
-  for (int e = 0; e < 16; ++e) lds[i0 + 16 + e][k] = vals[e];
+  for (int e = 0; e < 16; ++e) {
+    lds[i0 + 16 + e][kswz(i0 + 16 + e, k)] = vals[e];
+  }
 }
 
 template <typename AccL, typename AccR>
@@ -560,8 +575,10 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16_big(
       bf16x8 bv[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        av[f] = *(const bf16x8*)&lds_l[wr * 64 + f * 16 + (lane & 15)][kfrag];
-        bv[f] = *(const bf16x8*)&rbuf[wc * 64 + f * 16 + (lane & 15)][kfrag];
+        const int ra = wr * 64 + f * 16 + (lane & 15);
+        const int rb = wc * 64 + f * 16 + (lane & 15);
+        av[f] = *(const bf16x8*)&lds_l[ra][kswz(ra, kfrag)];
+        bv[f] = *(const bf16x8*)&rbuf[rb][kswz(rb, kfrag)];
       }
 #pragma unroll
       for (int fi = 0; fi < 4; ++fi) {
