@@ -43,7 +43,10 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 // (32 B) units keeps every 8-element fragment contiguous and 16 B
 // aligned while spreading the colfast scalar writes over banks.
 __device__ __forceinline__ int kswz(int row, int k) {
-  return k ^ (((row >> 3) & 3) << 4);
+  // rows 8 APART and 32 APART both alias banks at the 144 B stride
+  // (8*144 and 32*144 are 0 mod 128 B); mix both row bits so either
+  // spacing lands on distinct 32 B groups.
+  return k ^ ((((row >> 3) + (row >> 5)) & 3) << 4);
 }
 
 // ---------------------------------------------------------------- accessors
