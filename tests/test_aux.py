@@ -337,3 +337,28 @@ def test_update_factors_in_hook_equivalence() -> None:
         }
     for n in results[True]:
         torch.testing.assert_close(results[True][n], results[False][n])
+
+
+def test_empty_registration_step() -> None:
+    """A model with every layer skipped still steps cleanly."""
+    model = TinyModel()
+    p = KFACPreconditioner(model, skip_layers=['.*'])
+    assert len(p._layers) == 0
+    x = torch.randn(4, 10)
+    torch.nn.functional.cross_entropy(model(x), torch.randint(0, 3, (4,))).backward()
+    p.step()
+    assert p.steps == 1
+    assert p.memory_usage()['total'] == 0
+
+
+def test_kl_clip_none_disables_scaling() -> None:
+    torch.manual_seed(2)
+    model = TinyModel()
+    p = KFACPreconditioner(
+        model, factor_update_steps=1, inv_update_steps=1, kl_clip=None,
+    )
+    x = torch.randn(8, 10)
+    torch.nn.functional.cross_entropy(model(x), torch.randint(0, 3, (8,))).backward()
+    p.step()  # must not raise; grads preconditioned unscaled
+    for _, prm in model.named_parameters():
+        assert torch.isfinite(prm.grad).all()
