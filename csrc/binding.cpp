@@ -15,8 +15,6 @@
 
 namespace kfac {
 
-enum class Dtype : int { F32 = 0, BF16 = 1, F16 = 2 };
-
 template <typename T>
 hipError_t cov_linear_t(hipStream_t, const T*, long, int, int, bool, float*, float, float);
 template <typename T>

@@ -28,9 +28,6 @@ __host__ __device__ inline int ceil_div(int a, int b) {
 // fp32x4 accumulator for mfma_f32_16x16x4_f32 (4 AGPRs per lane).
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-// Supported input element types for factor accumulation.
-enum class Dtype : int { F32 = 0, BF16 = 1, F16 = 2 };
-
 template <typename T>
 __device__ __forceinline__ float to_f32(T v);
 

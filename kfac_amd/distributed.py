@@ -21,7 +21,6 @@ rank/world helpers), re-designed for RCCL over xGMI:
 
 from __future__ import annotations
 
-from collections import defaultdict
 from typing import Callable
 
 import torch
@@ -340,7 +339,6 @@ class TorchDistributedCommunicator:
             self._open[key] = bucket
             self._groups[key] = (group, average)
         bucket.append(payload)
-        self._pending[key].append(fut)
         fut._attach(bucket)
         return fut
 
@@ -351,7 +349,6 @@ class TorchDistributedCommunicator:
         scale = 1.0 / get_world_size(group) if average else 1.0
         bucket.communicate(group, scale)
         self._inflight.append(bucket)
-        self._pending[key] = []
 
     def _flush_key(self, key: tuple) -> None:
         if key in self._open and key in self._groups:
@@ -359,7 +356,9 @@ class TorchDistributedCommunicator:
             self._launch(key, group, average)
 
     def flush_allreduce_buckets(self) -> None:
-        """Launch every open bucket (trailing partial buckets)."""
+        """Launch every open bucket (trailing partial buckets) and drop
+        references to buckets whose tensors have been unpacked."""
         for key in list(self._open.keys()):
             group, average = self._groups[key]
             self._launch(key, group, average)
+        self._inflight = [b for b in self._inflight if not b._unpacked]
