@@ -208,10 +208,10 @@ class TorchDistributedCommunicator:
                 latency-bound on xGMI rings.
         """
         self._cap_bytes = int(bucket_cap_mb * 1024 * 1024)
-        # key -> (open bucket, pending futures not yet attached)
+        # key -> currently-open (not yet launched) bucket
         self._open: dict[tuple, AllreduceTensorBucket] = {}
-        self._pending: dict[tuple, list[_BucketFuture]] = defaultdict(list)
-        self._inflight: list[_Bucket] = []
+        # launched buckets kept alive until their tensors are unpacked
+        self._inflight: list[AllreduceTensorBucket] = []
         # group handle + averaging flag per bucket key
         self._groups: dict[tuple, tuple] = {}
 
