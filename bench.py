@@ -78,7 +78,11 @@ def main() -> None:
 
     if world > 1:
         os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
-        torch.distributed.init_process_group('nccl')
+        # KFAC_BENCH_BACKEND=gloo lets the full multi-rank path (DDP
+        # wrapper + KAISA collectives, cuda tensors) be validated with
+        # several ranks sharing one GPU; the driver's real run uses RCCL.
+        backend = os.environ.get('KFAC_BENCH_BACKEND', 'nccl')
+        torch.distributed.init_process_group(backend)
 
     from kfac_amd import KFACPreconditioner
     from kfac_amd import ops
@@ -177,15 +181,25 @@ def main() -> None:
                 precond_times.append(time.perf_counter() - t0)
         optimizer.step()
 
-    for _ in range(args.warmup):
+    verbose = os.environ.get('KFAC_BENCH_VERBOSE', '0') == '1'
+    if verbose:
+        import faulthandler
+
+        faulthandler.dump_traceback_later(150, repeat=True, file=sys.stderr)
+
+    for i in range(args.warmup):
         one_step(False)
+        if verbose:
+            print(f'[rank {rank}] warmup {i} done', file=sys.stderr, flush=True)
 
     if world > 1:
         torch.distributed.barrier()
     torch.cuda.synchronize()
     start = time.perf_counter()
-    for _ in range(args.steps):
+    for i in range(args.steps):
         one_step(True)
+        if verbose:
+            print(f'[rank {rank}] step {i} done', file=sys.stderr, flush=True)
     torch.cuda.synchronize()
     if world > 1:
         torch.distributed.barrier()
