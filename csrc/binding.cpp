@@ -21,6 +21,11 @@ template <typename T>
 hipError_t im2col_t(hipStream_t, const T*, T*, int, int, int, int, int, int, int, int, int, int, int);
 template <typename T>
 hipError_t cov_conv_g_t(hipStream_t, const T*, int, int, int, int, float*, float, float);
+template <typename T>
+hipError_t im2col_tr_t(hipStream_t, const T*, T*, int, int, int, int, int, int, int, int, int, int, int);
+template <typename T>
+hipError_t transpose_to_bf16_t(hipStream_t, const T*, long, int, int, __hip_bfloat16*, int);
+hipError_t cov_trans_t(hipStream_t, const __hip_bfloat16*, int, int, int, bool, float*, float, float);
 hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float);
 hipError_t precond_grouped_f32(hipStream_t, const void*, int, int);
 hipError_t precond_apply_grouped_f32(hipStream_t, const void*, int, int, float*, float*, long, float*, float*, float, float);
@@ -105,9 +110,18 @@ void cov_linear(
             (float)coeff));
       },
       [&] {
-        CHECK_OK(kfac::cov_linear_t<__hip_bfloat16>(
-            stream, (const __hip_bfloat16*)a.data_ptr(), K, M, K, bias, outp,
-            (float)beta, (float)coeff));
+        // bf16 route: transpose once to [K][Mpad] so SYRK staging is
+        // m-contiguous (16 B LDS writes) instead of column-fast scalar
+        // scatter; the extra HBM round trip is trivial vs the n^2 MFMA
+        // work (same argument as materialized im2col).
+        const int Mpad = (M + 63) / 64 * 64;
+        auto ws = torch::empty({(long)K, (long)Mpad}, a.options());
+        CHECK_OK(kfac::transpose_to_bf16_t<__hip_bfloat16>(
+            stream, (const __hip_bfloat16*)a.data_ptr(), K, M, K,
+            (__hip_bfloat16*)ws.data_ptr(), Mpad));
+        CHECK_OK(kfac::cov_trans_t(
+            stream, (const __hip_bfloat16*)ws.data_ptr(), Mpad, M, K, bias,
+            outp, (float)beta, (float)coeff));
       },
       [&] {
         CHECK_OK(kfac::cov_linear_t<__half>(
@@ -148,11 +162,11 @@ void cov_conv_a(
   // Materialize the patch matrix once (gather VALU cost paid once), then
   // run the flat-matrix SYRK with wide coalesced staging. The scratch
   // lives in the caching allocator; bytes are trivial vs 288 GB HBM3E.
-  const int K_pad = (K + 7) / 8 * 8;
-  auto scratch = torch::empty({m, (long)K_pad}, x.options());
   dispatch_dtype(
       x.scalar_type(),
       [&] {
+        const int K_pad = (K + 7) / 8 * 8;
+        auto scratch = torch::empty({m, (long)K_pad}, x.options());
         CHECK_OK(kfac::im2col_t<float>(
             stream, x.data_ptr<float>(), scratch.data_ptr<float>(), Nb, C, H,
             W, (int)kh, (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, K_pad));
@@ -161,15 +175,23 @@ void cov_conv_a(
             (float)beta, (float)coeff));
       },
       [&] {
-        CHECK_OK(kfac::im2col_t<__hip_bfloat16>(
+        // bf16: materialize the patch matrix TRANSPOSED ([K][Mpad]) —
+        // im2col touches every element exactly once either way, and the
+        // transposed image makes every SYRK staging write a contiguous
+        // 16 B LDS store (see TransAcc in syrk.hip).
+        const long Mpad = (m + 63) / 64 * 64;
+        auto scratch = torch::empty({(long)K, Mpad}, x.options());
+        CHECK_OK(kfac::im2col_tr_t<__hip_bfloat16>(
             stream, (const __hip_bfloat16*)x.data_ptr(),
             (__hip_bfloat16*)scratch.data_ptr(), Nb, C, H, W, (int)kh,
-            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, K_pad));
-        CHECK_OK(kfac::cov_linear_t<__hip_bfloat16>(
-            stream, (const __hip_bfloat16*)scratch.data_ptr(), K_pad, (int)m,
-            K, bias, outp, (float)beta, (float)coeff));
+            (int)kw, (int)sh, (int)sw, (int)ph, (int)pw, (int)Mpad));
+        CHECK_OK(kfac::cov_trans_t(
+            stream, (const __hip_bfloat16*)scratch.data_ptr(), (int)Mpad,
+            (int)m, K, bias, outp, (float)beta, (float)coeff));
       },
       [&] {
+        const int K_pad = (K + 7) / 8 * 8;
+        auto scratch = torch::empty({m, (long)K_pad}, x.options());
         CHECK_OK(kfac::im2col_t<__half>(
             stream, (const __half*)x.data_ptr(),
             (__half*)scratch.data_ptr(), Nb, C, H, W, (int)kh, (int)kw,

@@ -92,6 +92,56 @@ struct LinearAcc {
   }
 };
 
+// Transposed-image accessor: the matrix is materialized as [K][Mpad]
+// (m fastest-varying), so 16 consecutive-m values of one factor column
+// are 32 contiguous bytes of HBM and the m-fast staging path (contiguous
+// 16 B LDS writes, kLaneAlongCols=false) applies. This replaced the
+// column-fast LinearAcc staging for bf16: that path issued 16 scalar LDS
+// stores per 16 values (8192 per 128x64 tile stage, ~8x the MFMA issue
+// cycles of the slice). Rows m in [M, Mpad) are zero-filled by the
+// producer kernels so wide loads may cross M; the bias-ones column is
+// synthesized with the exact m < M bound (it accumulates the count).
+template <typename T>
+struct TransAcc {
+  const T* a;  // [K][Mpad]
+  int Mpad;
+  int M;      // real rows (loop bound; bias-ones extent)
+  int K;      // materialized columns (= rows of the transposed image)
+  int Ncols;  // K + (bias ? 1 : 0)
+  static constexpr bool kLaneAlongCols = false;
+
+  __device__ __forceinline__ float load(int m, int i) const {
+    if (m >= M || i >= Ncols) return 0.0f;
+    if (i < K) return to_f32(a[(long)i * Mpad + m]);
+    return 1.0f;
+  }
+
+  __device__ __forceinline__ void load16(int m, int i, float* out) const {
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = load(m + e, i);
+  }
+
+  __device__ __forceinline__ void load16_bf16(int m, int i, __bf16* out) const {
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      if (i < K && m + 15 < Mpad) {
+        __builtin_memcpy(out, a + (long)i * Mpad + m, 16 * sizeof(__bf16));
+        return;
+      }
+    }
+    if (i >= K && i < Ncols) {
+#pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        out[e] = (__bf16)((m + e < M) ? 1.0f : 0.0f);
+      }
+      return;
+    }
+    float tmp[16];
+    load16(m, i, tmp);
+#pragma unroll
+    for (int e = 0; e < 16; ++e) out[e] = (__bf16)tmp[e];
+  }
+};
+
 template <typename T>
 struct ConvPatchAcc {
   const T* x;
@@ -702,6 +752,133 @@ template hipError_t im2col_t<float>(hipStream_t, const float*, float*, int, int,
 template hipError_t im2col_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
 template hipError_t im2col_t<__half>(hipStream_t, const __half*, __half*, int, int, int, int, int, int, int, int, int, int, int);
 
+// Transposed im2col: dst is [K][Mpad] (m fastest-varying), the image the
+// TransAcc SYRK staging wants. Each thread gathers 16 consecutive output
+// positions of one patch column (one incremental ow/h walk) and writes
+// one contiguous 32 B run; m in [M, Mpad) is zero-filled.
+template <typename T>
+__global__ __launch_bounds__(256) void im2col_tr_kernel(
+    T* __restrict__ dst,  // [K][Mpad]
+    ConvPatchAcc<T> acc,
+    int Mpad) {
+  const int m16 = Mpad / 16;
+  const long total = (long)acc.K * m16;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(idx / m16);
+    const int m0 = (int)(idx % m16) * 16;
+    T vals[16];
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      acc.load16_bf16(m0, k, (__bf16*)vals);
+    } else {
+      float tmp[16];
+      acc.load16(m0, k, tmp);
+#pragma unroll
+      for (int e = 0; e < 16; ++e) vals[e] = (T)tmp[e];
+    }
+    __builtin_memcpy(&dst[(long)k * Mpad + m0], vals, 16 * sizeof(T));
+  }
+}
+
+template <typename T>
+hipError_t im2col_tr_t(
+    hipStream_t stream,
+    const T* x,
+    T* dst,
+    int Nb,
+    int C,
+    int H,
+    int W,
+    int kh,
+    int kw,
+    int sh,
+    int sw,
+    int ph,
+    int pw,
+    int Mpad) {
+  int OH = (H + 2 * ph - kh) / sh + 1;
+  int OW = (W + 2 * pw - kw) / sw + 1;
+  int K = C * kh * kw;
+  ConvPatchAcc<T> acc{x,  C,  H,  W,  OH, OW, kh,
+                      kw, sh, sw, ph, pw, Nb * OH * OW,
+                      K,  K};
+  long total = (long)K * (Mpad / 16);
+  int blocks = (int)min((total + 255) / 256, (long)2048);
+  im2col_tr_kernel<T><<<blocks, 256, 0, stream>>>(dst, acc, Mpad);
+  return hipGetLastError();
+}
+
+template hipError_t im2col_tr_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, __hip_bfloat16*, int, int, int, int, int, int, int, int, int, int, int);
+
+// [M][lda] row-major -> [K][Mpad] bf16, LDS-tiled 64x64 transpose (both
+// global phases fully coalesced; runs once per factor so its cost is the
+// one extra HBM round trip argued for materialized im2col above).
+template <typename T>
+__global__ __launch_bounds__(256) void transpose_to_bf16_kernel(
+    const T* __restrict__ src,
+    long lda,
+    int M,
+    int K,
+    __bf16* __restrict__ dst,
+    int Mpad) {
+  __shared__ __bf16 tile[64][72];
+  const int nmt = Mpad / 64;
+  const int nkt = ceil_div(K, 64);
+  const int tid = threadIdx.x;
+  for (int t = blockIdx.x; t < nmt * nkt; t += gridDim.x) {
+    const int m0 = (t % nmt) * 64;
+    const int k0 = (t / nmt) * 64;
+    {
+      const int m = m0 + (tid >> 2);
+      const int c0 = k0 + (tid & 3) * 16;
+      __bf16 v[16];
+      if (m < M) {
+#pragma unroll
+        for (int e = 0; e < 16; ++e) {
+          const int c = c0 + e;
+          v[e] = (__bf16)((c < K) ? to_f32(src[(long)m * lda + c]) : 0.0f);
+        }
+      } else {
+#pragma unroll
+        for (int e = 0; e < 16; ++e) v[e] = (__bf16)0.0f;
+      }
+      *(bf16x8*)&tile[tid >> 2][(tid & 3) * 16] = *(const bf16x8*)&v[0];
+      *(bf16x8*)&tile[tid >> 2][(tid & 3) * 16 + 8] = *(const bf16x8*)&v[8];
+    }
+    __syncthreads();
+    {
+      const int k = k0 + (tid >> 2);
+      if (k < K) {
+        const int r0 = (tid & 3) * 16;
+        __bf16 v[16];
+#pragma unroll
+        for (int e = 0; e < 16; ++e) v[e] = tile[r0 + e][tid >> 2];
+        __builtin_memcpy(&dst[(long)k * Mpad + m0 + r0], v,
+                         16 * sizeof(__bf16));
+      }
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+hipError_t transpose_to_bf16_t(
+    hipStream_t stream,
+    const T* src,
+    long lda,
+    int M,
+    int K,
+    __hip_bfloat16* dst,
+    int Mpad) {
+  int tiles = (Mpad / 64) * ceil_div(K, 64);
+  int blocks = min(tiles, 2048);
+  transpose_to_bf16_kernel<T><<<blocks, 256, 0, stream>>>(
+      src, lda, M, K, (__bf16*)dst, Mpad);
+  return hipGetLastError();
+}
+
+template hipError_t transpose_to_bf16_t<__hip_bfloat16>(hipStream_t, const __hip_bfloat16*, long, int, int, __hip_bfloat16*, int);
+
 // ---------------------------------------------------------------- launchers
 
 static int pick_splits(int M, int n_tiles) {
@@ -783,6 +960,23 @@ hipError_t cov_conv_g_t(
     float coeff) {
   ConvGradAcc<T> acc{g, C, OH, OW, Nb * OH * OW, C};
   return launch_syrk<use_bf16_mfma<T>()>(stream, out, acc.Ncols, acc, beta, coeff);
+}
+
+// SYRK over a pre-transposed [K][Mpad] bf16 image (im2col_tr_t /
+// transpose_to_bf16_t output): every staging write is a contiguous 16 B
+// LDS store regardless of the source layout.
+hipError_t cov_trans_t(
+    hipStream_t stream,
+    const __hip_bfloat16* ws,
+    int Mpad,
+    int M,
+    int K,
+    bool bias,
+    float* out,
+    float beta,
+    float coeff) {
+  TransAcc<__hip_bfloat16> acc{ws, Mpad, M, K, K + (bias ? 1 : 0)};
+  return launch_syrk<true>(stream, out, acc.Ncols, acc, beta, coeff);
 }
 
 // Explicit instantiations used by the binding.
