@@ -26,7 +26,7 @@ hipError_t im2col_tr_t(hipStream_t, const T*, T*, int, int, int, int, int, int, 
 template <typename T>
 hipError_t transpose_to_bf16_t(hipStream_t, const T*, long, int, int, __hip_bfloat16*, int);
 hipError_t cov_trans_t(hipStream_t, const __hip_bfloat16*, int, int, int, bool, float*, float, float);
-hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float);
+hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float, bool);
 hipError_t precond_grouped_f32(hipStream_t, const void*, int, int);
 hipError_t precond_apply_grouped_f32(hipStream_t, const void*, int, int, float*, float*, long, float*, float*, float, float);
 template <typename T>
@@ -251,12 +251,14 @@ torch::Tensor eigen_tail(
   // t2 = QG @ v2 : [m,m] x [m,n]
   CHECK_OK(kfac::gemm_f32(
       stream, t2.data_ptr<float>(), qg.data_ptr<float>(),
-      v2.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr, 0.f));
+      v2.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr, 0.f,
+      true));
   auto out = torch::empty_like(v2);
   // out = t2 @ QA^T : [m,n] x [n,n]^T
   CHECK_OK(kfac::gemm_f32(
       stream, out.data_ptr<float>(), t2.data_ptr<float>(),
-      qa.data_ptr<float>(), m, n, n, false, true, 0, nullptr, nullptr, 0.f));
+      qa.data_ptr<float>(), m, n, n, false, true, 0, nullptr, nullptr, 0.f,
+      true));
   return out;
 }
 
@@ -279,13 +281,14 @@ torch::Tensor precond_eigen_fused(
   // t1 = QG^T @ grad (reference association: (qg.t() @ grad) @ qa)
   CHECK_OK(kfac::gemm_f32(
       stream, t1.data_ptr<float>(), qg.data_ptr<float>(),
-      g32.data_ptr<float>(), m, n, m, true, false, 0, nullptr, nullptr, 0.f));
+      g32.data_ptr<float>(), m, n, m, true, false, 0, nullptr, nullptr, 0.f,
+      true));
   auto v2 = torch::empty_like(g32);
   // v2 = (t1 @ QA) * dgda   (epilogue-fused elementwise)
   CHECK_OK(kfac::gemm_f32(
       stream, v2.data_ptr<float>(), t1.data_ptr<float>(),
       qa.data_ptr<float>(), m, n, n, false, false, 1, dgda.data_ptr<float>(),
-      nullptr, 0.f));
+      nullptr, 0.f, true));
   return eigen_tail(v2, qa, qg, stream).to(dtype);
 }
 
@@ -310,13 +313,14 @@ torch::Tensor precond_eigen(
   // t1 = QG^T @ grad
   CHECK_OK(kfac::gemm_f32(
       stream, t1.data_ptr<float>(), qg.data_ptr<float>(),
-      g32.data_ptr<float>(), m, n, m, true, false, 0, nullptr, nullptr, 0.f));
+      g32.data_ptr<float>(), m, n, m, true, false, 0, nullptr, nullptr, 0.f,
+      true));
   auto v2 = torch::empty_like(g32);
   // v2 = (t1 @ QA) / (outer(dg, da) + damping)
   CHECK_OK(kfac::gemm_f32(
       stream, v2.data_ptr<float>(), t1.data_ptr<float>(),
       qa.data_ptr<float>(), m, n, n, false, false, 2, dg.data_ptr<float>(),
-      da.data_ptr<float>(), (float)damping));
+      da.data_ptr<float>(), (float)damping, true));
   return eigen_tail(v2, qa, qg, stream).to(dtype);
 }
 
@@ -527,12 +531,13 @@ torch::Tensor precond_inverse(
   CHECK_OK(kfac::gemm_f32(
       stream, t1.data_ptr<float>(), gi.data_ptr<float>(),
       g32.data_ptr<float>(), m, n, m, false, false, 0, nullptr, nullptr,
-      0.f));
+      0.f, false));
   auto out = torch::empty_like(g32);
   // out = t1 @ A^-1
   CHECK_OK(kfac::gemm_f32(
       stream, out.data_ptr<float>(), t1.data_ptr<float>(),
-      ai.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f));
+      ai.data_ptr<float>(), m, n, n, false, false, 0, nullptr, nullptr, 0.f,
+      false));
   return out.to(dtype);
 }
 

@@ -26,6 +26,27 @@ constexpr int GLDS = GBT;
 
 enum class Epilogue : int { NONE = 0, MUL = 1, DIV_OUTER = 2 };
 
+typedef __attribute__((ext_vector_type(8))) __bf16 gemm_bf16x8;
+
+// Split-precision operands: x = hi + lo with both parts bf16 keeps ~16
+// mantissa bits per element, and 3 bf16 MFMAs (hi*hi + hi*lo + lo*hi)
+// replace the 8 f32 MFMAs of a 32-deep K slice. f32 MFMA is the
+// deliberately slow path on CDNA4 (155 TF vs 2.5 PF bf16), so this cuts
+// the chain's MFMA issue cycles ~5x; the dropped lo*lo term is O(2^-32)
+// relative and the end-to-end precondition error vs the fp64 reference
+// stays ~1e-5 (tests/test_ops_gpu.py precond_* tolerances).
+__device__ __forceinline__ void split_bf16(
+    const float* v,
+    gemm_bf16x8& hi,
+    gemm_bf16x8& lo) {
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    __bf16 h = (__bf16)v[q];
+    hi[q] = h;
+    lo[q] = (__bf16)(v[q] - (float)h);  // exact f32 residual
+  }
+}
+
 // LDS column swizzle at 4-float (16 B) granularity: distributes the
 // trans-path writes (8 lanes sharing a column index across k-rows,
 // measured 2.8e9 SQ_LDS_BANK_CONFLICT cycles on the stage-4 GEMM) over
@@ -120,7 +141,7 @@ __device__ __forceinline__ void stage_gemm(
 // (64 f32 accumulators/lane); per 32-deep K slice each wave issues 128
 // mfma_f32_16x16x4_f32 against 16 LDS reads per substep — the MFMA:LDS
 // ratio that the 64x64 structure lacked (guide §5 step-2 ladder).
-template <Epilogue EPI>
+template <Epilogue EPI, bool SPLIT>
 __device__ __forceinline__ void gemm_tile_body(
     float* __restrict__ c,
     const float* __restrict__ a,
@@ -152,22 +173,55 @@ __device__ __forceinline__ void gemm_tile_body(
     stage_gemm(a, M, K, lda, !ta, k0, i0, lds_a, tid);
     stage_gemm(b, N, K, ldb, tb, k0, j0, lds_b, tid);
     __syncthreads();
-#pragma unroll
-    for (int kk = 0; kk < GBK; kk += 4) {
-      const int krow = kk + (lane >> 4);
-      float av[4];
-      float bv[4];
+    if constexpr (SPLIT) {
+      // One 16x16x32 bf16 MFMA triple covers the whole 32-deep slice:
+      // lane l supplies elements k = 8*(l>>4)..+7 of column (l&15).
+      const int kbase = (lane >> 4) * 8;
+      gemm_bf16x8 ah[4], al[4], bh[4], bl[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        av[f] = lds_a[krow][lds_swz(krow, wr * 64 + f * 16 + (lane & 15))];
-        bv[f] = lds_b[krow][lds_swz(krow, wc * 64 + f * 16 + (lane & 15))];
+        const int ca = wr * 64 + f * 16 + (lane & 15);
+        const int cb = wc * 64 + f * 16 + (lane & 15);
+        float va[8], vb[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          va[q] = lds_a[kbase + q][lds_swz(kbase + q, ca)];
+          vb[q] = lds_b[kbase + q][lds_swz(kbase + q, cb)];
+        }
+        split_bf16(va, ah[f], al[f]);
+        split_bf16(vb, bh[f], bl[f]);
       }
 #pragma unroll
       for (int fi = 0; fi < 4; ++fi) {
 #pragma unroll
         for (int fj = 0; fj < 4; ++fj) {
-          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              av[fi], bv[fj], acc[fi][fj], 0, 0, 0);
+          // cross terms first so the large hi*hi lands last in the chain
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ah[fi], bl[fj], acc[fi][fj], 0, 0, 0);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              al[fi], bh[fj], acc[fi][fj], 0, 0, 0);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ah[fi], bh[fj], acc[fi][fj], 0, 0, 0);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int kk = 0; kk < GBK; kk += 4) {
+        const int krow = kk + (lane >> 4);
+        float av[4];
+        float bv[4];
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          av[f] = lds_a[krow][lds_swz(krow, wr * 64 + f * 16 + (lane & 15))];
+          bv[f] = lds_b[krow][lds_swz(krow, wc * 64 + f * 16 + (lane & 15))];
+        }
+#pragma unroll
+        for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+          for (int fj = 0; fj < 4; ++fj) {
+            acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                av[fi], bv[fj], acc[fi][fj], 0, 0, 0);
+          }
         }
       }
     }
@@ -196,7 +250,7 @@ __device__ __forceinline__ void gemm_tile_body(
   }
 }
 
-template <Epilogue EPI>
+template <Epilogue EPI, bool SPLIT>
 __global__ __launch_bounds__(256) void gemm_kernel(
     float* __restrict__ c,
     const float* __restrict__ a,
@@ -211,7 +265,7 @@ __global__ __launch_bounds__(256) void gemm_kernel(
     float damping) {
   __shared__ float lds_a[GBK][GLDS];
   __shared__ float lds_b[GBK][GLDS];
-  gemm_tile_body<EPI>(
+  gemm_tile_body<EPI, SPLIT>(
       c, a, b, M, N, K, ta, tb, ta ? (long)M : (long)K,
       tb ? (long)K : (long)N, (long)N, e1, e2, damping,
       (int)blockIdx.x * GBT, (int)blockIdx.y * GBT, lds_a, lds_b);
@@ -265,19 +319,19 @@ __global__ __launch_bounds__(256) void grouped_precond_kernel(
 
   const long sp = d.spad;
   if constexpr (STAGE == 1) {
-    gemm_tile_body<Epilogue::NONE>(
+    gemm_tile_body<Epilogue::NONE, true>(
         d.s1, d.qg, d.grad, m, n, m, true, false, (long)m, sp, sp, nullptr,
         nullptr, 0.f, i0, j0, lds_a, lds_b);
   } else if constexpr (STAGE == 2) {
-    gemm_tile_body<Epilogue::MUL>(
+    gemm_tile_body<Epilogue::MUL, true>(
         d.s2, d.s1, d.qa, m, n, n, false, false, sp, (long)n, sp, d.dgda,
         nullptr, 0.f, i0, j0, lds_a, lds_b);
   } else if constexpr (STAGE == 3) {
-    gemm_tile_body<Epilogue::NONE>(
+    gemm_tile_body<Epilogue::NONE, true>(
         d.s1, d.qg, d.s2, m, n, m, false, false, (long)m, sp, sp, nullptr,
         nullptr, 0.f, i0, j0, lds_a, lds_b);
   } else {
-    gemm_tile_body<Epilogue::NONE>(
+    gemm_tile_body<Epilogue::NONE, true>(
         d.out, d.s1, d.qa, m, n, n, false, true, sp, (long)n, sp, nullptr,
         nullptr, 0.f, i0, j0, lds_a, lds_b);
   }
@@ -405,20 +459,36 @@ hipError_t gemm_f32(
     int epilogue,
     const float* e1,
     const float* e2,
-    float damping) {
+    float damping,
+    bool split) {
   dim3 grid(ceil_div(M, GBT), ceil_div(N, GBT));
   switch (static_cast<Epilogue>(epilogue)) {
     case Epilogue::NONE:
-      gemm_kernel<Epilogue::NONE><<<grid, 256, 0, stream>>>(
-          c, a, b, M, N, K, ta, tb, nullptr, nullptr, 0.0f);
+      if (split) {
+        gemm_kernel<Epilogue::NONE, true><<<grid, 256, 0, stream>>>(
+            c, a, b, M, N, K, ta, tb, nullptr, nullptr, 0.0f);
+      } else {
+        gemm_kernel<Epilogue::NONE, false><<<grid, 256, 0, stream>>>(
+            c, a, b, M, N, K, ta, tb, nullptr, nullptr, 0.0f);
+      }
       break;
     case Epilogue::MUL:
-      gemm_kernel<Epilogue::MUL><<<grid, 256, 0, stream>>>(
-          c, a, b, M, N, K, ta, tb, e1, nullptr, 0.0f);
+      if (split) {
+        gemm_kernel<Epilogue::MUL, true><<<grid, 256, 0, stream>>>(
+            c, a, b, M, N, K, ta, tb, e1, nullptr, 0.0f);
+      } else {
+        gemm_kernel<Epilogue::MUL, false><<<grid, 256, 0, stream>>>(
+            c, a, b, M, N, K, ta, tb, e1, nullptr, 0.0f);
+      }
       break;
     case Epilogue::DIV_OUTER:
-      gemm_kernel<Epilogue::DIV_OUTER><<<grid, 256, 0, stream>>>(
-          c, a, b, M, N, K, ta, tb, e1, e2, damping);
+      if (split) {
+        gemm_kernel<Epilogue::DIV_OUTER, true><<<grid, 256, 0, stream>>>(
+            c, a, b, M, N, K, ta, tb, e1, e2, damping);
+      } else {
+        gemm_kernel<Epilogue::DIV_OUTER, false><<<grid, 256, 0, stream>>>(
+            c, a, b, M, N, K, ta, tb, e1, e2, damping);
+      }
       break;
   }
   return hipGetLastError();

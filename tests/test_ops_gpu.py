@@ -127,13 +127,21 @@ def test_precond_eigen_fused(m, n) -> None:
 
 
 def test_precond_eigen_dgda_bf16_grad() -> None:
+    """bf16-grad round trip through the non-prediv eigen precondition.
+
+    Bases are orthogonal (eigenvectors of symmetric matrices) as in real
+    K-FAC state — random dense bases amplify rounding ~1000x (see
+    test_precond_eigen_fused) and do not represent the op's input class.
+    """
     from kfac_amd.ops import reference as ref
 
     torch.manual_seed(4)
     m, n = 128, 257
     grad = torch.randn(m, n, device='cuda', dtype=torch.bfloat16)
-    qa = torch.randn(n, n, device='cuda')
-    qg = torch.randn(m, m, device='cuda')
+    sa = torch.randn(n, n, device='cuda')
+    qa = torch.linalg.eigh(sa + sa.t())[1].contiguous()
+    sg = torch.randn(m, m, device='cuda')
+    qg = torch.linalg.eigh(sg + sg.t())[1].contiguous()
     dg = torch.rand(m, device='cuda') + 0.1
     da = torch.rand(n, device='cuda') + 0.1
     expected = ref.precond_eigen(grad, qa, qg, da=da, dg=dg, damping=1e-3)
