@@ -519,18 +519,16 @@ class BaseKFACPreconditioner:
                             if fac is not None:
                                 groups[fac.shape[0]].append((layer, fac))
                         for n, items in groups.items():
-                            if len(items) == 1:
-                                d, q = torch.linalg.eigh(
-                                    items[0][1].to(torch.float32),
-                                )
-                                d = torch.clamp(d, min=0.0).unsqueeze(0)
-                                q = q.unsqueeze(0)
+                            stack = torch.stack(
+                                [f.to(torch.float32) for _, f in items],
+                            )
+                            if stack.is_cuda:
+                                from kfac_amd import ops as _ops
+
+                                d, q = _ops.eigh_batched(stack)
                             else:
-                                stack = torch.stack(
-                                    [f.to(torch.float32) for _, f in items],
-                                )
                                 d, q = torch.linalg.eigh(stack)
-                                d = torch.clamp(d, min=0.0)
+                            d = torch.clamp(d, min=0.0)
                             for i, (layer, _) in enumerate(items):
                                 res = results.setdefault(layer, {})
                                 res[f'q{which}'] = (
@@ -707,7 +705,7 @@ class BaseKFACPreconditioner:
                 )
             groups[(factor.shape[0], factor.device, factor.dtype)].append(layer)
         for (n, dev, _dt), group in groups.items():
-            if len(group) == 1:
+            if len(group) == 1 and dev.type != 'cuda':
                 layer = group[0]
                 if which == 'a':
                     layer.compute_a_inv()
@@ -722,20 +720,13 @@ class BaseKFACPreconditioner:
                     for layer in group
                 ],
             )
-            if n <= 64 and dev.type == 'cuda':
-                # hand-written LDS-resident batched Jacobi: one launch
-                # per group vs rocSOLVER syevd's thousands
-                # (csrc/eigh.hip; eigenvalue order is irrelevant to the
-                # Kronecker preconditioner).
-                from kfac_amd import ops as _ops
+            # n <= 64: hand-written LDS-resident batched Jacobi (one
+            # launch per group); larger n: rocSOLVER syevd behind a
+            # hipGraph replay. Single-layer CUDA groups go through the
+            # same path so they share the graph cache.
+            from kfac_amd import ops as _ops
 
-                ext = _ops._load_ext()
-                if ext is not None:
-                    d, q = ext.syevj_small(stack, 20, 1e-5)
-                else:
-                    d, q = torch.linalg.eigh(stack)
-            else:
-                d, q = torch.linalg.eigh(stack)
+            d, q = _ops.eigh_batched(stack)
             d = torch.clamp(d, min=0.0)
             for i, layer in enumerate(group):
                 qv = q[i].to(layer.inv_dtype).contiguous()

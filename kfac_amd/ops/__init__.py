@@ -268,6 +268,27 @@ def eigh(x: torch.Tensor, *, clamp: bool = True) -> tuple[torch.Tensor, torch.Te
     return ref.eigh(x, clamp=clamp)
 
 
+def eigh_batched(stack: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Batched fp32 symmetric eigendecomposition of a (B, n, n) stack.
+
+    On CUDA: n <= 64 routes to the hand-written one-wave-per-matrix LDS
+    Jacobi kernel (eigenvalues unsorted — K-FAC is order-invariant);
+    larger n routes to rocSOLVER syevd whose ~50k-launch
+    tridiagonalization sequence is captured into a hipGraph on first use
+    per shape and replayed on later inverse phases (the phase is
+    launch-bound, see profiles/eigh_strategies.md). Eigenvalues are NOT
+    clamped here; callers clamp >= 0 (reference eigen.py:321,344).
+    """
+    if stack.is_cuda:
+        ext = _load_ext()
+        if ext is not None:
+            if stack.size(1) <= 64:
+                return ext.syevj_small(stack.contiguous(), 20, 1e-5)
+            w, vt = ext.syevd_batched(stack.contiguous())
+            return w, vt.transpose(1, 2)
+    return torch.linalg.eigh(stack)
+
+
 def inv_damped(x: torch.Tensor, damping: float) -> torch.Tensor:
     """(x + damping I)^-1 in fp32."""
     return ref.inv_damped(x, damping)
