@@ -136,11 +136,91 @@ __device__ __forceinline__ void stage_gemm(
   }
 }
 
+// Register staging for the pipelined split path: each thread owns 16
+// elements (4 float4) of one matrix's 32x128 slab. Global loads for
+// slice t+1 issue right after slice t's LDS writes, so their HBM
+// latency hides behind slice t's MFMAs (guide T14: one register set,
+// write after the barrier, re-issue immediately) — the glds path's
+// issue cost (~60-185 cyc per 1 KiB piece) is no longer small against
+// the bf16x3 MFMA budget (~670 cyc/slice vs f32's ~3600).
+struct StageRegs {
+  float4 v[4];
+};
+
+__device__ __forceinline__ void stage_load(
+    const float* __restrict__ src,
+    int rows,
+    int ks,
+    long ld,
+    bool trans,
+    int k0,
+    int i0,
+    int tid,
+    StageRegs& r) {
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int c = tid + e * 256;
+    float v[4];
+    if (trans) {
+      const int k = (c & 7) * 4;
+      const int i = c >> 3;
+      const int gi = i0 + i;
+      const long base = (long)gi * ld + (k0 + k);
+      if (gi < rows && k0 + k + 3 < ks) {
+        __builtin_memcpy(v, src + base, 16);
+      } else {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          v[q] = (gi < rows && k0 + k + q < ks) ? src[base + q] : 0.0f;
+        }
+      }
+    } else {
+      const int i = (c & 31) * 4;
+      const int k = c >> 5;
+      const int gk = k0 + k;
+      const long base = (long)gk * ld + (i0 + i);
+      if (gk < ks && i0 + i + 3 < rows) {
+        __builtin_memcpy(v, src + base, 16);
+      } else {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          v[q] = (gk < ks && i0 + i + q < rows) ? src[base + q] : 0.0f;
+        }
+      }
+    }
+    r.v[e] = make_float4(v[0], v[1], v[2], v[3]);
+  }
+}
+
+__device__ __forceinline__ void stage_store(
+    const StageRegs& r,
+    bool trans,
+    int tid,
+    float (*lds)[GLDS]) {
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int c = tid + e * 256;
+    if (trans) {
+      const int k = (c & 7) * 4;
+      const int i = c >> 3;
+      const float* v = (const float*)&r.v[e];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) lds[k + q][lds_swz(k + q, i)] = v[q];
+    } else {
+      const int i = (c & 31) * 4;
+      const int k = c >> 5;
+      *(float4*)&lds[k][lds_swz(k, i)] = r.v[e];
+    }
+  }
+}
+
 // One 128x128 output tile of C = op(A) @ op(B) (+ epilogue).
 // 4 waves, each owning a 64x64 quadrant as 4x4 16x16 MFMA fragments
 // (64 f32 accumulators/lane); per 32-deep K slice each wave issues 128
 // mfma_f32_16x16x4_f32 against 16 LDS reads per substep — the MFMA:LDS
 // ratio that the 64x64 structure lacked (guide §5 step-2 ladder).
+// The SPLIT (bf16x3) instantiation instead runs the register-staged
+// pipeline above with 48 bf16 MFMAs per slice.
 template <Epilogue EPI, bool SPLIT>
 __device__ __forceinline__ void gemm_tile_body(
     float* __restrict__ c,
@@ -169,11 +249,21 @@ __device__ __forceinline__ void gemm_tile_body(
 
   f32x4 acc[4][4] = {};
 
-  for (int k0 = 0; k0 < K; k0 += GBK) {
-    stage_gemm(a, M, K, lda, !ta, k0, i0, lds_a, tid);
-    stage_gemm(b, N, K, ldb, tb, k0, j0, lds_b, tid);
-    __syncthreads();
-    if constexpr (SPLIT) {
+  if constexpr (SPLIT) {
+    // Register-staged pipeline: loads for slice t+1 issue while slice
+    // t's MFMAs run; LDS writes happen between the two barriers.
+    StageRegs ra, rb;
+    stage_load(a, M, K, lda, !ta, 0, i0, tid, ra);
+    stage_load(b, N, K, ldb, tb, 0, j0, tid, rb);
+    for (int k0 = 0; k0 < K; k0 += GBK) {
+      __syncthreads();  // prior slice's LDS reads complete
+      stage_store(ra, !ta, tid, lds_a);
+      stage_store(rb, tb, tid, lds_b);
+      if (k0 + GBK < K) {
+        stage_load(a, M, K, lda, !ta, k0 + GBK, i0, tid, ra);
+        stage_load(b, N, K, ldb, tb, k0 + GBK, j0, tid, rb);
+      }
+      __syncthreads();  // this slice's LDS image ready
       // One 16x16x32 bf16 MFMA triple covers the whole 32-deep slice:
       // lane l supplies elements k = 8*(l>>4)..+7 of column (l&15).
       const int kbase = (lane >> 4) * 8;
@@ -204,7 +294,12 @@ __device__ __forceinline__ void gemm_tile_body(
               ah[fi], bh[fj], acc[fi][fj], 0, 0, 0);
         }
       }
-    } else {
+    }
+  } else {
+    for (int k0 = 0; k0 < K; k0 += GBK) {
+      stage_gemm(a, M, K, lda, !ta, k0, i0, lds_a, tid);
+      stage_gemm(b, N, K, ldb, tb, k0, j0, lds_b, tid);
+      __syncthreads();
 #pragma unroll
       for (int kk = 0; kk < GBK; kk += 4) {
         const int krow = kk + (lane >> 4);
@@ -224,8 +319,8 @@ __device__ __forceinline__ void gemm_tile_body(
           }
         }
       }
+      __syncthreads();
     }
-    __syncthreads();
   }
 
 #pragma unroll
