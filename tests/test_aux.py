@@ -362,3 +362,48 @@ def test_kl_clip_none_disables_scaling() -> None:
     p.step()  # must not raise; grads preconditioned unscaled
     for _, prm in model.named_parameters():
         assert torch.isfinite(prm.grad).all()
+
+
+def _trace_worker_synced() -> None:
+    import torch.distributed as dist
+
+    from kfac_amd import tracing
+
+    @tracing.trace(sync=True)
+    def traced_barrier_op() -> int:
+        return dist.get_rank()
+
+    traced_barrier_op()
+    traced_barrier_op()
+    t = tracing.get_trace(average=True)
+    key = next(k for k in t if k.endswith('traced_barrier_op'))
+    assert t[key] >= 0.0
+    tracing.clear_trace()
+    assert tracing.get_trace() == {}
+
+
+def test_trace_synced_world2() -> None:
+    """Reference parity: synced (dist.barrier-bracketed) tracing at world 2
+    (tracing_test.py)."""
+    from testing.distributed import run_distributed
+
+    run_distributed(2, _trace_worker_synced)
+
+
+def _harness_failing_worker() -> None:
+    import torch.distributed as dist
+
+    if dist.get_rank() == 1:
+        raise AssertionError('intentional failure on rank 1')
+
+
+def test_harness_surfaces_worker_failure() -> None:
+    """The fork harness itself is tested (reference
+    tests/testing/distributed_wrapper_test.py): a failing rank must fail
+    the test, not hang or pass silently."""
+    import pytest as _pytest
+
+    from testing.distributed import run_distributed
+
+    with _pytest.raises(AssertionError, match='ranks'):
+        run_distributed(2, _harness_failing_worker, timeout=60.0)
