@@ -359,3 +359,8 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
                 filepath = os.path.join(self.factor_checkpoint_dir, name)
                 logger.info(f'saving KFAC factors for {name} to {filepath}')
                 torch.save(layer.state_dict(), filepath)
+        # Writers and readers of the factor dir can be different ranks
+        # (inv worker saves, factor worker loads) — barrier so no rank
+        # reads a half-written file (reference gpt_neox/preconditioner.py:390).
+        if torch.distributed.is_initialized():
+            torch.distributed.barrier()
