@@ -280,7 +280,14 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
                 partition.append((name, lsd))
 
         partitions: list[Any] = [None for _ in range(get_world_size())]
-        group = torch.distributed.new_group(backend='gloo')
+        # CPU-side object gather rides a gloo group (RCCL cannot gather
+        # pickled CPU state); cached so repeated checkpointing does not
+        # create a new process group per call.
+        if getattr(self, '_gloo_ckpt_group', None) is None:
+            self._gloo_ckpt_group = torch.distributed.new_group(
+                backend='gloo',
+            )
+        group = self._gloo_ckpt_group
         torch.distributed.all_gather_object(partitions, partition, group=group)
 
         layers: dict[str, Any] = {}
