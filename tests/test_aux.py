@@ -407,3 +407,33 @@ def test_harness_surfaces_worker_failure() -> None:
 
     with _pytest.raises(AssertionError, match='ranks'):
         run_distributed(2, _harness_failing_worker, timeout=60.0)
+
+
+def test_state_dict_excludes_callable_hyperparams() -> None:
+    """Reference parity (base_preconditioner.py:215-247): callable
+    hyperparameters are not serializable and are omitted from state."""
+    model = TinyModel()
+    p = KFACPreconditioner(
+        model,
+        factor_update_steps=1,
+        inv_update_steps=1,
+        damping=lambda step: 0.001 * 0.99**step,
+        lr=0.1,
+    )
+    sd = p.state_dict(include_factors=False)
+    assert 'damping' not in sd
+    assert sd['lr'] == 0.1
+    assert sd['steps'] == 0
+
+
+def test_shared_module_registered_once() -> None:
+    """A module reused at two points in the graph gets ONE K-FAC layer
+    (named_modules dedup; reference register.py:20-28)."""
+    import torch.nn as nn
+
+    from kfac_amd.layers.register import get_flattened_modules
+
+    lin = nn.Linear(4, 4)
+    model = nn.Sequential(lin, nn.ReLU(), lin)
+    mods = get_flattened_modules(model)
+    assert sum(1 for _, m in mods if m is lin) == 1

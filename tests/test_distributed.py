@@ -205,3 +205,29 @@ def test_bucket_invariants() -> None:
     from testing.distributed import run_distributed
 
     run_distributed(1, _bucket_invariants_body)
+
+
+def _inflight_pruned_worker() -> None:
+    import torch
+
+    from kfac_amd.distributed import TorchDistributedCommunicator
+
+    comm = TorchDistributedCommunicator(bucket_cap_mb=0.001)
+    futures = [
+        comm.allreduce_bucketed(torch.full((64, 64), float(i)))
+        for i in range(8)
+    ]
+    comm.flush_allreduce_buckets()
+    for i, f in enumerate(futures):
+        t = f.wait() if hasattr(f, 'wait') else f
+        assert torch.allclose(t, torch.full((64, 64), float(i)))
+    # buckets whose tensors were unpacked must be dropped on the next
+    # flush (regression: in-flight list grew without bound)
+    comm.flush_allreduce_buckets()
+    assert comm._inflight == []
+
+
+def test_inflight_buckets_pruned() -> None:
+    from testing.distributed import run_distributed
+
+    run_distributed(2, _inflight_pruned_worker)
