@@ -122,3 +122,42 @@ def test_validation() -> None:
             work, local_rank=0, world_size=4, grad_worker_fraction=0.3,
             group_func=_mock_group,
         )
+
+
+@pytest.mark.parametrize('world', [2, 4, 6, 8, 12, 16, 32, 64])
+def test_assignment_sweep_all_fractions(world: int) -> None:
+    """Broad invariant sweep: every divisible grad-worker count, mixed
+    factor sizes, every rank's view must agree (reference
+    assignment_test.py property style)."""
+    work = {
+        f'l{i}': {'A': float(((7 * i) % 13 + 1) ** 3), 'G': float((i % 5 + 1) ** 3)}
+        for i in range(17)
+    }
+    for gw in [w for w in range(1, world + 1) if world % w == 0]:
+        frac = gw / world
+        views = [
+            KAISAAssignment(
+                work,
+                local_rank=rank,
+                world_size=world,
+                grad_worker_fraction=frac,
+                group_func=_mock_group,
+                colocate_factors=False,
+            )
+            for rank in range(world)
+        ]
+        base = views[0]
+        for layer in base.get_layers():
+            for fac in ('A', 'G'):
+                owners = {v.inv_worker(layer, fac) for v in views}
+                # every rank agrees on the single inverse worker
+                assert len(owners) == 1
+            # exactly one source grad worker per receiver group, and it
+            # is a grad worker for the layer
+            for rank, v in enumerate(views):
+                src = v.src_grad_worker(layer)
+                assert views[src].is_grad_worker(layer)
+        # greedy balance sanity: no rank has more than total work
+        # (degenerate) and every rank's view of broadcast flags agrees
+        assert len({v.broadcast_gradients() for v in views}) == 1
+        assert len({v.broadcast_inverses() for v in views}) == 1
