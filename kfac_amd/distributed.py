@@ -143,7 +143,14 @@ class AllreduceTensorBucket:
         self._flat = torch.cat(flats)
         if scale != 1.0:
             self._flat.mul_(scale)
-        self._work = dist.all_reduce(self._flat, group=group, async_op=True)
+        if dist.is_available() and dist.is_initialized():
+            self._work = dist.all_reduce(
+                self._flat, group=group, async_op=True,
+            )
+        else:
+            # world of one (same fallback as get_world_size): the scaled
+            # flat buffer IS the result
+            self._work = None
 
     def wait_and_unpack(self) -> None:
         if self._unpacked:
