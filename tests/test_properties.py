@@ -81,3 +81,54 @@ def test_bucket_capacity_property(sizes: list[int], cap_kb: int) -> None:
     # unpack writes the (scale-applied) result back into the originals
     for i, t in enumerate(tensors):
         assert torch.all(t == 2.0 * float(i))
+
+
+@SETTINGS
+@given(
+    c=st.integers(1, 6),
+    h=st.integers(3, 14),
+    kh=st.integers(1, 5),
+    sh=st.integers(1, 3),
+    ph=st.integers(0, 2),
+    nb=st.integers(1, 3),
+    bias=st.booleans(),
+    seed=st.integers(0, 2**16),
+)
+def test_cov_conv_a_reference_property(
+    c: int, h: int, kh: int, sh: int, ph: int, nb: int, bias: bool, seed: int,
+) -> None:
+    """ops.reference.cov_conv_a vs an independent unfold-based oracle
+    across random conv geometries (the CPU reference is the ground truth
+    the GPU kernels are tested against, so it gets its own oracle)."""
+    from hypothesis import assume
+
+    from kfac_amd.ops import reference as ref
+
+    oh = (h + 2 * ph - kh) // sh + 1
+    assume(oh >= 1)
+    assume(kh + 2 * ph <= h + ph)  # patch never fully in padding
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(nb, c, h, h, generator=g)
+    n = c * kh * kh + int(bias)
+    out = torch.zeros(n, n)
+    ref.cov_conv_a(
+        x,
+        kernel_size=(kh, kh),
+        stride=(sh, sh),
+        padding=(ph, ph),
+        bias=bias,
+        out=out,
+        beta=0.0,
+    )
+    # oracle: unfold -> [M, c*kh*kh] (+ ones), spatial-scaled covariance
+    patches = torch.nn.functional.unfold(
+        x, (kh, kh), padding=(ph, ph), stride=(sh, sh),
+    )  # (nb, c*kh*kh, oh*ow)
+    m2 = patches.transpose(1, 2).reshape(-1, c * kh * kh)
+    s = float(oh * oh)
+    m2 = m2 / s
+    if bias:
+        m2 = torch.cat([m2, torch.full((m2.size(0), 1), 1.0 / s)], dim=1)
+    # reference get_a_factor: [patches, ones] / s, then cov = a^T a / M
+    expected = (m2.t() @ m2) / m2.size(0)
+    torch.testing.assert_close(out, expected, rtol=1e-4, atol=1e-4)
