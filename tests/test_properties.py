@@ -132,3 +132,84 @@ def test_cov_conv_a_reference_property(
     # reference get_a_factor: [patches, ones] / s, then cov = a^T a / M
     expected = (m2.t() @ m2) / m2.size(0)
     torch.testing.assert_close(out, expected, rtol=1e-4, atol=1e-4)
+
+
+@SETTINGS
+@given(
+    m=st.integers(2, 8),
+    n=st.integers(2, 8),
+    seed=st.integers(0, 2**16),
+    damping=st.sampled_from([1e-3, 1e-2, 0.3]),
+)
+def test_precond_eigen_matches_kronecker_solve(
+    m: int, n: int, seed: int, damping: float,
+) -> None:
+    """The eigen-method preconditioned gradient equals the EXACT damped
+    Kronecker solve (G (x) A + damping I)^-1 vec(grad) — the math the
+    whole method approximates (reference eigen.py:350-385 computes it
+    via the eigenbasis; here we check against torch.linalg.solve on the
+    explicitly materialized (m*n, m*n) system)."""
+    from kfac_amd.ops import reference as ref
+
+    gen = torch.Generator().manual_seed(seed)
+    ra = torch.randn(n, n, generator=gen, dtype=torch.float64)
+    rg = torch.randn(m, m, generator=gen, dtype=torch.float64)
+    A = ra @ ra.t() / n + 0.1 * torch.eye(n, dtype=torch.float64)
+    G = rg @ rg.t() / m + 0.1 * torch.eye(m, dtype=torch.float64)
+    grad = torch.randn(m, n, generator=gen, dtype=torch.float64)
+
+    da, qa = torch.linalg.eigh(A)
+    dg, qg = torch.linalg.eigh(G)
+    out = ref.precond_eigen(
+        grad, qa.contiguous(), qg.contiguous(), da=da, dg=dg, damping=damping,
+    )
+
+    big = torch.kron(G, A) + damping * torch.eye(m * n, dtype=torch.float64)
+    exact = torch.linalg.solve(big, grad.reshape(-1)).reshape(m, n)
+    torch.testing.assert_close(out, exact, rtol=1e-8, atol=1e-10)
+
+
+@SETTINGS
+@given(
+    c=st.integers(1, 8),
+    oh=st.integers(1, 7),
+    nb=st.integers(1, 4),
+    seed=st.integers(0, 2**16),
+)
+def test_cov_conv_g_oracle(c: int, oh: int, nb: int, seed: int) -> None:
+    from kfac_amd.ops import reference as ref
+
+    gen = torch.Generator().manual_seed(seed)
+    g = torch.randn(nb, c, oh, oh, generator=gen)
+    out = torch.zeros(c, c)
+    ref.cov_conv_g(g, out=out, beta=0.0)
+    s = float(oh * oh)
+    rows = g.permute(0, 2, 3, 1).reshape(-1, c) / s
+    expected = rows.t() @ rows / rows.size(0)
+    torch.testing.assert_close(out, expected, rtol=1e-4, atol=1e-5)
+
+
+@SETTINGS
+@given(
+    n=st.integers(1, 500),
+    kl=st.sampled_from([1e-4, 1e-3, 1e-1, 10.0]),
+    lr=st.sampled_from([0.01, 0.1, 1.0]),
+    seed=st.integers(0, 2**16),
+)
+def test_grad_scale_formula(n: int, kl: float, lr: float, seed: int) -> None:
+    """scale = min(1, sqrt(kl_clip / |sum(precon*grad) * lr^2|))
+    (reference base_preconditioner.py:411-435)."""
+    import math
+
+    from kfac_amd import ops
+
+    gen = torch.Generator().manual_seed(seed)
+    p = torch.randn(n, generator=gen)
+    g = torch.randn(n, generator=gen)
+    accum = torch.zeros(())
+    ops.kl_clip_accum(accum, p, g)
+    torch.testing.assert_close(accum, (p * g).sum())
+    scale = float(ops.grad_scale_from_accum(accum, kl, lr))
+    vg = abs(float((p * g).sum()) * lr * lr)
+    expected = min(1.0, math.sqrt(kl / vg)) if vg > 0 else 1.0
+    assert abs(scale - expected) < 1e-5
