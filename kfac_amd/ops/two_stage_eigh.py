@@ -1,0 +1,122 @@
+"""Two-stage symmetric eigensolver — round-2 groundwork.
+
+Stage 1 (implemented here, torch ops): full symmetric -> symmetric BAND
+matrix of bandwidth ``b`` via blocked Householder panels — geqrf on each
+sub-panel and compact-WY two-sided application via ormqr. This is the
+GEMM-dominated 4/3 n^3 part that replaces rocSOLVER sytrd's
+memory-bound latrd chain (the measured inverse-phase bottleneck, see
+docs/eigh_two_stage_plan.md); on GPU, geqrf/ormqr dispatch to rocSOLVER's
+blocked implementations and the bulk of the flops are plain GEMMs that
+round 2 moves onto the split-precision MFMA path.
+
+Stage 2+3 (band -> tridiagonal -> eigenpairs): validated here through
+LAPACK's banded solver (scipy eig_banded) on CPU — numerically exact but
+CPU-bound (measured 15 s at n=4608, so NOT a production path); round 2
+replaces it with a batched bulge-chasing kernel + rocsolver_sstedc.
+
+The module exists so the stage-1 math and the pipeline plumbing
+(band layout, Q1 accumulation order, eigenvector back-transform) are
+already validated end-to-end (tests/test_two_stage.py) before the
+kernels are written.
+"""
+
+from __future__ import annotations
+
+import torch
+
+__all__ = ['reduce_to_band', 'apply_q1', 'eigh_two_stage_cpu']
+
+
+def reduce_to_band(
+    a: torch.Tensor,
+    band: int,
+) -> tuple[torch.Tensor, list[tuple[int, torch.Tensor, torch.Tensor]]]:
+    """Reduce a symmetric matrix to symmetric banded form.
+
+    Returns ``(B, panels)`` where ``B`` is symmetric with bandwidth
+    ``band`` (``B[i, j] == 0`` for ``|i - j| > band``), ``panels`` is the
+    list of ``(row0, geqrf_a, geqrf_tau)`` Householder panels, and
+    ``A == Q1 @ B @ Q1.T`` with ``Q1`` the product of the panel
+    reflectors (apply with :func:`apply_q1`).
+
+    The per-panel work is one tall-skinny QR (geqrf) plus two ormqr
+    applications to the trailing submatrix — i.e. (I - V T V^T)^T S
+    (I - V T V^T), which LAPACK/rocSOLVER evaluate as blocked GEMMs.
+    """
+    if a.dim() != 2 or a.size(0) != a.size(1):
+        raise ValueError(f'expected square matrix, got {tuple(a.shape)}')
+    if band < 1:
+        raise ValueError('band must be >= 1')
+    n = a.size(0)
+    b = a.clone()
+    panels: list[tuple[int, torch.Tensor, torch.Tensor]] = []
+    j = 0
+    while j + band < n:
+        r0 = j + band
+        ncols = min(band, n - r0)  # never wider than the rows below
+        panel = b[r0:, j : j + ncols].contiguous()
+        qr_a, tau = torch.geqrf(panel)
+        k = min(panel.size(0), panel.size(1))
+        block = torch.zeros_like(panel)
+        block[:k, :] = torch.triu(qr_a[:k, :])
+        b[r0:, j : j + ncols] = block
+        b[j : j + ncols, r0:] = block.t()
+        # Q^T from the left over ALL columns right of the panel and Q
+        # from the right over all those rows — with a ragged panel
+        # (ncols < band) the strip j+ncols..r0-1 is not yet banded and
+        # MUST be transformed too, or the result is not a similarity.
+        rest = slice(j + ncols, n)
+        b[r0:, rest] = torch.ormqr(
+            qr_a, tau, b[r0:, rest].contiguous(), left=True, transpose=True,
+        )
+        b[rest, r0:] = torch.ormqr(
+            qr_a, tau, b[rest, r0:].contiguous(), left=False, transpose=False,
+        )
+        s = b[r0:, r0:]
+        b[r0:, r0:] = 0.5 * (s + s.t())  # exact-symmetry hygiene
+        b[r0:, j + ncols : r0] = b[j + ncols : r0, r0:].t()
+        panels.append((r0, qr_a, tau))
+        j += ncols
+    return b, panels
+
+
+def apply_q1(
+    panels: list[tuple[int, torch.Tensor, torch.Tensor]],
+    x: torch.Tensor,
+) -> torch.Tensor:
+    """Compute ``Q1 @ x`` for the band reduction's accumulated basis.
+
+    ``Q1 = Q_p1 @ Q_p2 @ ... @ Q_pk`` (panel order), so the panels apply
+    right-to-left; each panel only touches rows ``row0:``.
+    """
+    out = x.clone()
+    for row0, qr_a, tau in reversed(panels):
+        out[row0:] = torch.ormqr(
+            qr_a, tau, out[row0:].contiguous(), left=True, transpose=False,
+        )
+    return out
+
+
+def eigh_two_stage_cpu(
+    a: torch.Tensor,
+    band: int = 32,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Full two-stage eigendecomposition (CPU validation pipeline).
+
+    Stage 1: :func:`reduce_to_band`; stages 2+3: LAPACK banded solver.
+    Returns ``(w, v)`` with the torch.linalg.eigh convention (ascending
+    eigenvalues, eigenvectors in columns).
+    """
+    import numpy as np
+    from scipy.linalg import eig_banded
+
+    dt = a.dtype
+    b_mat, panels = reduce_to_band(a.to(torch.float64), band)
+    n = a.size(0)
+    ab = np.zeros((band + 1, n))
+    bm = b_mat.numpy()
+    for i in range(band + 1):
+        ab[i, : n - i] = np.diagonal(bm, -i)
+    w, v = eig_banded(ab, lower=True)
+    vec = apply_q1(panels, torch.from_numpy(v))
+    return torch.from_numpy(w).to(dt), vec.to(dt)
