@@ -45,6 +45,20 @@ def get_module_helper(module: torch.nn.Module) -> ModuleHelper | None:
     if isinstance(module, torch.nn.Linear):
         return LinearModuleHelper(module)
     if isinstance(module, torch.nn.Conv2d):
+        if module.groups != 1:
+            # Grouped/depthwise convs have block-diagonal Kronecker
+            # factors the K-FAC formulation here (and the reference,
+            # which crashes on them mid-precondition) does not model:
+            # skip cleanly so the layer trains unpreconditioned.
+            import warnings
+
+            warnings.warn(
+                f'K-FAC does not support grouped Conv2d '
+                f'(groups={module.groups}); layer will not be '
+                'preconditioned.',
+                stacklevel=2,
+            )
+            return None
         return Conv2dModuleHelper(module)
     return None
 

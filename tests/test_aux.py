@@ -437,3 +437,26 @@ def test_shared_module_registered_once() -> None:
     model = nn.Sequential(lin, nn.ReLU(), lin)
     mods = get_flattened_modules(model)
     assert sum(1 for _, m in mods if m is lin) == 1
+
+
+def test_grouped_conv_skipped_gracefully() -> None:
+    """Grouped/depthwise Conv2d is skipped with a warning instead of
+    crashing mid-precondition (the reference has no groups handling and
+    fails with an opaque shape error)."""
+    import warnings as _warnings
+
+    m = torch.nn.Sequential(
+        torch.nn.Conv2d(8, 16, 3, padding=1, groups=4),
+        torch.nn.Flatten(),
+        torch.nn.Linear(16 * 8 * 8, 4),
+    )
+    with _warnings.catch_warnings(record=True) as rec:
+        _warnings.simplefilter('always')
+        p = KFACPreconditioner(m, factor_update_steps=1, inv_update_steps=1)
+    assert any('grouped' in str(w.message) for w in rec)
+    assert len(p._layers) == 1  # only the Linear
+    x = torch.randn(2, 8, 8, 8)
+    torch.nn.functional.cross_entropy(
+        m(x), torch.randint(0, 4, (2,)),
+    ).backward()
+    p.step()  # completes; conv trains unpreconditioned
