@@ -216,3 +216,57 @@ class Conv2dModuleHelper(ModuleHelper):
         coeff_scale: float,
     ) -> None:
         ops.cov_conv_g(g, out=out, beta=beta, coeff_scale=coeff_scale)
+
+
+class Conv1dModuleHelper(Conv2dModuleHelper):
+    """Adapter for torch.nn.Conv1d — treated as a (length x 1) Conv2d.
+
+    Capability extension beyond the reference (which supports Linear and
+    Conv2d only, modules.py:36-43): the (N, C, L) input is viewed as
+    (N, C, L, 1) so the existing fused conv covariance kernels apply
+    unchanged with kw = sw = 1, pw = 0.
+    """
+
+    def __init__(self, module: torch.nn.Conv1d):
+        self.module = module
+
+    @property
+    def a_factor_shape(self) -> tuple[int, int]:
+        (k,) = cast(tuple, self.module.kernel_size)
+        n = self.module.in_channels * k + int(self.has_bias())
+        return (n, n)
+
+    def accumulate_a_factor(
+        self,
+        a: torch.Tensor,
+        out: torch.Tensor,
+        beta: float,
+        coeff_scale: float,
+    ) -> None:
+        (k,) = cast(tuple, self.module.kernel_size)
+        (s,) = cast(tuple, self.module.stride)
+        (p,) = cast(tuple, self.module.padding)
+        ops.cov_conv_a(
+            a.unsqueeze(-1).contiguous(),
+            kernel_size=(k, 1),
+            stride=(s, 1),
+            padding=(p, 0),
+            bias=self.has_bias(),
+            out=out,
+            beta=beta,
+            coeff_scale=coeff_scale,
+        )
+
+    def accumulate_g_factor(
+        self,
+        g: torch.Tensor,
+        out: torch.Tensor,
+        beta: float,
+        coeff_scale: float,
+    ) -> None:
+        ops.cov_conv_g(
+            g.unsqueeze(-1).contiguous(),
+            out=out,
+            beta=beta,
+            coeff_scale=coeff_scale,
+        )

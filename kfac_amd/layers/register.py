@@ -60,6 +60,14 @@ def get_module_helper(module: torch.nn.Module) -> ModuleHelper | None:
             )
             return None
         return Conv2dModuleHelper(module)
+    if isinstance(module, torch.nn.Conv1d):
+        if module.groups != 1 or not isinstance(
+            module.padding, (tuple, list),
+        ):
+            return None
+        from kfac_amd.layers.modules import Conv1dModuleHelper
+
+        return Conv1dModuleHelper(module)
     return None
 
 

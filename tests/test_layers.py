@@ -181,3 +181,39 @@ def test_inverse_world4_broadcast_symmetric() -> None:
     run_distributed(
         4, _drive_layer, 'inverse', False, True, broadcast=True,
     )
+
+
+def test_conv1d_end_to_end() -> None:
+    """Conv1d support (extension beyond the reference): factors, eigen
+    precondition and a converging training loop on CPU."""
+    import torch
+
+    from kfac_amd import KFACPreconditioner
+
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(
+        torch.nn.Conv1d(4, 8, 3, padding=1),
+        torch.nn.ReLU(),
+        torch.nn.Conv1d(8, 8, 5, stride=2, padding=2),
+        torch.nn.Flatten(),
+        torch.nn.Linear(8 * 8, 3),
+    )
+    p = KFACPreconditioner(model, factor_update_steps=1, inv_update_steps=2)
+    assert len(p._layers) == 3
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    x = torch.randn(16, 4, 16)
+    y = torch.randint(0, 3, (16,))
+    losses = []
+    for _ in range(15):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        p.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0]
+    # A factor shape: in*k + bias column
+    layer = next(
+        lay for _, (n, lay) in p._layers.items() if n == '0'
+    )
+    assert layer.a_factor.shape == (4 * 3 + 1, 4 * 3 + 1)
