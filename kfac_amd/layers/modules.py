@@ -270,3 +270,75 @@ class Conv1dModuleHelper(Conv2dModuleHelper):
             beta=beta,
             coeff_scale=coeff_scale,
         )
+
+
+class Conv3dModuleHelper(ModuleHelper):
+    """Adapter for torch.nn.Conv3d (extension beyond the reference).
+
+    Patch extraction (3D im2col) runs as torch unfold views; the heavy
+    covariance SYRK still routes through ops.cov_linear (the fused HIP
+    kernel on GPU) with the conv coefficient 1/(M*s^2) — same semantics
+    as Conv2d: [patches, ones] / spatial_size, cov scaled by rows.
+    """
+
+    def __init__(self, module: torch.nn.Conv3d):
+        self.module = module
+
+    @property
+    def a_factor_shape(self) -> tuple[int, int]:
+        kd, kh, kw = cast(tuple, self.module.kernel_size)
+        n = self.module.in_channels * kd * kh * kw + int(self.has_bias())
+        return (n, n)
+
+    @property
+    def g_factor_shape(self) -> tuple[int, int]:
+        n = self.module.out_channels
+        return (n, n)
+
+    def _patches(self, x: torch.Tensor) -> tuple[torch.Tensor, int]:
+        kd, kh, kw = cast(tuple, self.module.kernel_size)
+        sd, sh, sw = cast(tuple, self.module.stride)
+        pd, ph, pw = cast(tuple, self.module.padding)
+        if pd or ph or pw:
+            x = torch.nn.functional.pad(x, (pw, pw, ph, ph, pd, pd))
+        p = x.unfold(2, kd, sd).unfold(3, kh, sh).unfold(4, kw, sw)
+        # (N, C, OD, OH, OW, kd, kh, kw) -> (N, OD, OH, OW, C, kd, kh, kw)
+        p = p.permute(0, 2, 3, 4, 1, 5, 6, 7)
+        spatial = p.size(1) * p.size(2) * p.size(3)
+        flat = p.reshape(-1, p.size(4) * p.size(5) * p.size(6) * p.size(7))
+        return flat.contiguous(), spatial
+
+    def accumulate_a_factor(
+        self,
+        a: torch.Tensor,
+        out: torch.Tensor,
+        beta: float,
+        coeff_scale: float,
+    ) -> None:
+        flat, s = self._patches(a)
+        m = flat.size(0)
+        ops.cov_linear(
+            flat,
+            bias=self.has_bias(),
+            out=out,
+            beta=beta,
+            coeff=coeff_scale / (m * s * s),
+        )
+
+    def accumulate_g_factor(
+        self,
+        g: torch.Tensor,
+        out: torch.Tensor,
+        beta: float,
+        coeff_scale: float,
+    ) -> None:
+        s = g.size(2) * g.size(3) * g.size(4)
+        rows = g.permute(0, 2, 3, 4, 1).reshape(-1, g.size(1)).contiguous()
+        m = rows.size(0)
+        ops.cov_linear(
+            rows,
+            bias=False,
+            out=out,
+            beta=beta,
+            coeff=coeff_scale / (m * s * s),
+        )

@@ -68,6 +68,14 @@ def get_module_helper(module: torch.nn.Module) -> ModuleHelper | None:
         from kfac_amd.layers.modules import Conv1dModuleHelper
 
         return Conv1dModuleHelper(module)
+    if isinstance(module, torch.nn.Conv3d):
+        if module.groups != 1 or not isinstance(
+            module.padding, (tuple, list),
+        ):
+            return None
+        from kfac_amd.layers.modules import Conv3dModuleHelper
+
+        return Conv3dModuleHelper(module)
     return None
 
 

@@ -217,3 +217,63 @@ def test_conv1d_end_to_end() -> None:
         lay for _, (n, lay) in p._layers.items() if n == '0'
     )
     assert layer.a_factor.shape == (4 * 3 + 1, 4 * 3 + 1)
+
+
+def test_conv3d_matches_conv2d_when_depth1() -> None:
+    """Conv3d with kd=1 on depth-1 input is exactly a Conv2d: the K-FAC
+    factors must agree with the Conv2d helper's."""
+    import torch
+
+    from kfac_amd.layers.modules import (
+        Conv2dModuleHelper,
+        Conv3dModuleHelper,
+    )
+
+    torch.manual_seed(3)
+    c2 = torch.nn.Conv2d(3, 6, (3, 3), stride=(2, 2), padding=(1, 1))
+    c3 = torch.nn.Conv3d(3, 6, (1, 3, 3), stride=(1, 2, 2), padding=(0, 1, 1))
+    h2 = Conv2dModuleHelper(c2)
+    h3 = Conv3dModuleHelper(c3)
+    x2 = torch.randn(4, 3, 10, 10)
+    x3 = x2.unsqueeze(2)  # depth 1
+    n = 3 * 9 + 1
+    a2 = torch.zeros(n, n)
+    a3 = torch.zeros(n, n)
+    h2.accumulate_a_factor(x2, a2, 0.0, 1.0)
+    h3.accumulate_a_factor(x3, a3, 0.0, 1.0)
+    torch.testing.assert_close(a2, a3, rtol=1e-5, atol=1e-6)
+    g2 = torch.randn(4, 6, 5, 5)
+    g3 = g2.unsqueeze(2)
+    o2 = torch.zeros(6, 6)
+    o3 = torch.zeros(6, 6)
+    h2.accumulate_g_factor(g2, o2, 0.0, 1.0)
+    h3.accumulate_g_factor(g3, o3, 0.0, 1.0)
+    torch.testing.assert_close(o2, o3, rtol=1e-5, atol=1e-6)
+
+
+def test_conv3d_end_to_end() -> None:
+    import torch
+
+    from kfac_amd import KFACPreconditioner
+
+    torch.manual_seed(1)
+    model = torch.nn.Sequential(
+        torch.nn.Conv3d(2, 4, 3, padding=1),
+        torch.nn.ReLU(),
+        torch.nn.Flatten(),
+        torch.nn.Linear(4 * 4 * 6 * 6, 3),
+    )
+    p = KFACPreconditioner(model, factor_update_steps=1, inv_update_steps=2)
+    assert len(p._layers) == 2
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    x = torch.randn(8, 2, 4, 6, 6)
+    y = torch.randint(0, 3, (8,))
+    losses = []
+    for _ in range(15):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        p.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0]
