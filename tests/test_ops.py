@@ -15,7 +15,12 @@ from kfac_amd.layers.utils import append_bias_ones
 from kfac_amd.layers.utils import get_cov
 from kfac_amd.ops import reference as ref
 
-torch.manual_seed(0)
+
+@pytest.fixture(autouse=True)
+def _seed_rng():
+    # per-test seeding: module-level seeding runs at import (collection)
+    # time and earlier tests shift the global RNG stream
+    torch.manual_seed(0)
 
 
 def literal_cov(a: torch.Tensor) -> torch.Tensor:
