@@ -33,6 +33,13 @@ from testing.models import TinyModel  # noqa: E402
 
 # ---------------------------------------------------------------- hyperparams
 
+
+@pytest.fixture(autouse=True)
+def _seed_rng():
+    # deterministic per-test RNG regardless of suite ordering
+    torch.manual_seed(0)
+
+
 def test_exp_decay_factor_averaging() -> None:
     fn = exp_decay_factor_averaging()
     assert fn(0) == fn(1) == 0.0

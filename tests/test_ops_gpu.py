@@ -29,6 +29,12 @@ def _require_ext():
     )
 
 
+@pytest.fixture(autouse=True)
+def _seed_rng():
+    # deterministic per-test RNG regardless of suite ordering
+    torch.manual_seed(0)
+
+
 def _ext():
     from kfac_amd import _kfaccore
 
