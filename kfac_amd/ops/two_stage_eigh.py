@@ -26,7 +26,9 @@ import torch
 
 __all__ = [
     'reduce_to_band',
+    'reduce_to_band_batched',
     'apply_q1',
+    'apply_q1_batched',
     'band_to_tridiag',
     'eigh_two_stage_cpu',
     'eigh_two_stage_self',
@@ -214,3 +216,64 @@ def eigh_two_stage_self(
     w, z = torch.linalg.eigh(tri)
     vec = apply_q1(panels, q2 @ z)
     return w.to(dt), vec.to(dt)
+
+
+def reduce_to_band_batched(
+    a: torch.Tensor,
+    band: int,
+) -> tuple[torch.Tensor, list[tuple[int, torch.Tensor, torch.Tensor]]]:
+    """Batched band reduction of a (B, n, n) stack of symmetric matrices.
+
+    Same panel schedule as :func:`reduce_to_band` but every geqrf/ormqr
+    is BATCHED over the group — the form the inverse phase needs (K-FAC
+    eigendecomposes same-size factor groups: 3x4608, 24x3072, ...) and
+    the fix for the measured per-panel launch latency
+    (docs/eigh_two_stage_plan.md: 72 sequential panels at b=64 were
+    10x slower than 36 at b=128; batching multiplies the work per
+    launch by the group size instead).
+    """
+    if a.dim() != 3 or a.size(1) != a.size(2):
+        raise ValueError(f'expected (B, n, n), got {tuple(a.shape)}')
+    n = a.size(1)
+    b = a.clone()
+    panels: list[tuple[int, torch.Tensor, torch.Tensor]] = []
+    j = 0
+    while j + band < n:
+        r0 = j + band
+        ncols = min(band, n - r0)
+        panel = b[:, r0:, j : j + ncols].contiguous()
+        qr_a, tau = torch.geqrf(panel)
+        k = min(panel.size(1), panel.size(2))
+        block = torch.zeros_like(panel)
+        block[:, :k, :] = torch.triu(qr_a[:, :k, :])
+        b[:, r0:, j : j + ncols] = block
+        b[:, j : j + ncols, r0:] = block.transpose(1, 2)
+        rest = slice(j + ncols, n)
+        b[:, r0:, rest] = torch.ormqr(
+            qr_a, tau, b[:, r0:, rest].contiguous(),
+            left=True, transpose=True,
+        )
+        b[:, rest, r0:] = torch.ormqr(
+            qr_a, tau, b[:, rest, r0:].contiguous(),
+            left=False, transpose=False,
+        )
+        s = b[:, r0:, r0:]
+        b[:, r0:, r0:] = 0.5 * (s + s.transpose(1, 2))
+        b[:, r0:, j + ncols : r0] = b[:, j + ncols : r0, r0:].transpose(1, 2)
+        panels.append((r0, qr_a, tau))
+        j += ncols
+    return b, panels
+
+
+def apply_q1_batched(
+    panels: list[tuple[int, torch.Tensor, torch.Tensor]],
+    x: torch.Tensor,
+) -> torch.Tensor:
+    """Batched ``Q1 @ x`` for :func:`reduce_to_band_batched` output."""
+    out = x.clone()
+    for row0, qr_a, tau in reversed(panels):
+        out[:, row0:] = torch.ormqr(
+            qr_a, tau, out[:, row0:].contiguous(),
+            left=True, transpose=False,
+        )
+    return out

@@ -102,3 +102,26 @@ def test_self_pipeline_matches_eigh(n: int, band: int) -> None:
     torch.testing.assert_close(
         a @ v, v @ torch.diag(w), rtol=1e-9, atol=1e-9,
     )
+
+
+@pytest.mark.parametrize('bsz,n,band', [(3, 100, 32), (5, 65, 16), (2, 150, 64)])
+def test_batched_band_reduction_matches_loop(bsz: int, n: int, band: int) -> None:
+    from kfac_amd.ops.two_stage_eigh import (
+        apply_q1_batched,
+        reduce_to_band_batched,
+    )
+
+    stack = torch.stack([_spd(n, 100 * i + n) for i in range(bsz)])
+    bb, panels = reduce_to_band_batched(stack, band)
+    eye = torch.eye(n, dtype=torch.float64).expand(bsz, n, n).contiguous()
+    q1 = apply_q1_batched(panels, eye)
+    torch.testing.assert_close(
+        q1 @ bb @ q1.transpose(1, 2), stack, rtol=1e-9, atol=1e-9,
+    )
+    for i in range(bsz):
+        b1, _ = reduce_to_band(stack[i], band)
+        torch.testing.assert_close(bb[i], b1, rtol=1e-9, atol=1e-9)
+        torch.testing.assert_close(
+            torch.linalg.eigvalsh(bb[i]), torch.linalg.eigvalsh(stack[i]),
+            rtol=1e-9, atol=1e-9,
+        )
