@@ -62,6 +62,12 @@ def parse_args() -> argparse.Namespace:
         default=1,
         help='pipeline eigendecompositions behind training steps (0=off)',
     )
+    p.add_argument(
+        '--lm-full',
+        action='store_true',
+        help='LM only: precondition attention projections too (not just '
+        'MLP linears) — exercises the n=3072/768 factor pipeline',
+    )
     return p.parse_args()
 
 
@@ -127,8 +133,11 @@ def main() -> None:
             'hybrid-opt': DistributedStrategy.HYBRID_OPT,
             'mem-opt': DistributedStrategy.MEM_OPT,
         }[args.strategy]
-    # HYBRID at world 2 means grad_worker_fraction 0.5 = 1/world -> MEM_OPT
-    # semantics; KAISAAssignment handles every divisible fraction.
+    # HYBRID at world 2 means grad_worker_fraction 0.5 = 1/world, which IS
+    # MEM-OPT placement (1 grad worker per layer); label the output with
+    # the effective semantics so a 2-GPU SCALE row is not mislabeled.
+    if strategy is DistributedStrategy.HYBRID_OPT and world == 2:
+        strategy_name = 'mem-opt'
 
     lr = 0.1
     optimizer = torch.optim.SGD(
@@ -149,7 +158,15 @@ def main() -> None:
             allreduce_bucket_cap_mb=25.0,
             compute_method=args.compute_method,
             compute_eigenvalue_outer_product=args.compute_method == 'eigen',
-            skip_layers=KFAC_SKIP_LAYERS if is_lm else [],
+            skip_layers=(
+                (
+                    ['embed.*', '.*embed_out.*']
+                    if args.lm_full
+                    else KFAC_SKIP_LAYERS
+                )
+                if is_lm
+                else []
+            ),
             inv_update_async=bool(args.async_inverse),
         )
 
@@ -195,6 +212,7 @@ def main() -> None:
     if world > 1:
         torch.distributed.barrier()
     torch.cuda.synchronize()
+    steps_at_start = precon.steps if precon is not None else 0
     start = time.perf_counter()
     for i in range(args.steps):
         one_step(True)
@@ -210,6 +228,42 @@ def main() -> None:
         t = torch.tensor([elapsed], device=device)
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
+
+    # All-in accounting: a short driver window may contain fewer inverse
+    # phases than the schedule implies (expected = steps/inv_update_steps),
+    # flattering ms_per_step. Measure one synchronous inverse phase now and
+    # add the missing pro-rata share so the amortized number is reported no
+    # matter what --steps the driver picks (conservative for async configs,
+    # whose phases partially overlap training).
+    all_in_ms_per_step = None
+    inv_phase_ms = None
+    if precon is not None:
+        inv = args.inv_update_steps
+        observed = sum(
+            1
+            for s in range(steps_at_start, steps_at_start + args.steps)
+            if s % inv == 0
+        )
+        expected = args.steps / inv
+        if world > 1:
+            torch.distributed.barrier()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        precon._compute_local_inverses()
+        precon._broadcast_inverses()
+        torch.cuda.synchronize()
+        if world > 1:
+            torch.distributed.barrier()
+        phase_s = time.perf_counter() - t0
+        if world > 1:
+            t = torch.tensor([phase_s], device=device)
+            torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+            phase_s = float(t.item())
+        inv_phase_ms = phase_s * 1000.0
+        missing = max(0.0, expected - observed)
+        all_in_ms_per_step = (
+            (elapsed + missing * phase_s) / args.steps * 1000.0
+        )
 
     global_batch = bs * world
     if is_lm:
@@ -248,6 +302,9 @@ def main() -> None:
                 'inv_update_steps': args.inv_update_steps,
                 'async_inverse': bool(args.async_inverse),
                 'compute_method': args.compute_method,
+                'lm_full': args.lm_full if is_lm else None,
+                'all_in_ms_per_step': all_in_ms_per_step,
+                'inv_phase_ms': inv_phase_ms,
                 'precond_step_ms_mean': (
                     sum(precond_times) / len(precond_times) * 1000.0
                     if precond_times

@@ -321,7 +321,11 @@ class TorchDistributedCommunicator:
         world = get_world_size(group)
         if world <= 1:
             return tensor
-        key = self._group_key(group) + (tensor.dtype, tensor.device)
+        # average is part of the key: the scale is applied per BUCKET at
+        # launch, so tensors with different averaging must not share one
+        # (otherwise the flag recorded by the bucket's first tensor would
+        # silently mis-scale later appends).
+        key = self._group_key(group) + (tensor.dtype, tensor.device, average)
         if symmetric:
             if tensor.dim() != 2 or tensor.size(0) != tensor.size(1):
                 raise NonSquareTensorError(

@@ -17,11 +17,11 @@ def exp_decay_factor_averaging(
         ValueError: if ``min_value <= 0``.
     """
     if min_value <= 0:
-        raise ValueError('min_value must be greater than 0')
+        raise ValueError(f'min_value must be positive, got {min_value}')
 
     def _factor_weight(step: int) -> float:
         if step < 0:
-            raise ValueError(f'step value cannot be negative. Got step={step}.')
+            raise ValueError(f'negative step ({step}) passed to schedule')
         if step == 0:
             step = 1
         return min(1 - (1 / step), min_value)

@@ -59,6 +59,24 @@ def get_module_helper(module: torch.nn.Module) -> ModuleHelper | None:
                 stacklevel=2,
             )
             return None
+        if module.dilation != (1, 1) or not isinstance(
+            module.padding, (tuple, list),
+        ):
+            # String padding ('same'/'valid') has no (ph, pw) for the
+            # fused im2col, and dilated kernels would need dilated patch
+            # extraction — neither is modeled, so skip with a warning
+            # (same guard as the Conv1d/Conv3d helpers) rather than
+            # computing a silently wrong A factor.
+            import warnings
+
+            warnings.warn(
+                f'K-FAC Conv2d supports dilation=(1, 1) and numeric '
+                f'padding only (got dilation={module.dilation}, '
+                f'padding={module.padding!r}); layer will not be '
+                'preconditioned.',
+                stacklevel=2,
+            )
+            return None
         return Conv2dModuleHelper(module)
     if isinstance(module, torch.nn.Conv1d):
         if module.groups != 1 or not isinstance(

@@ -79,4 +79,14 @@ class LambdaParamScheduler:
             new = current * factor
             if name in _INT_PARAMS:
                 new = int(new)
+                if new < 1:
+                    # A decayed step-count hitting 0 would crash the
+                    # ``steps % update_steps`` modulo in the hooks; the
+                    # reference silently truncates (kfac/scheduler.py) —
+                    # fail loudly instead.
+                    raise ValueError(
+                        f'{name} schedule produced {current * factor:.4g} '
+                        f'at step {at}, which truncates to {new}; '
+                        'step-count hyperparameters must stay >= 1.',
+                    )
             setattr(self._preconditioner, attr, new)

@@ -68,6 +68,20 @@ def test_lambda_param_scheduler() -> None:
     assert p.factor_update_steps == 3
 
 
+def test_scheduler_step_count_truncation_guard() -> None:
+    # A decay schedule that would truncate factor_update_steps to 0 must
+    # raise instead of silently producing a modulo-by-zero time bomb
+    # (the reference's scheduler truncates silently).
+    model = TinyModel()
+    p = KFACPreconditioner(model, factor_update_steps=1)
+    sched = LambdaParamScheduler(
+        p, factor_update_steps_lambda=lambda step: 0.5,
+    )
+    with pytest.raises(ValueError, match='stay >= 1'):
+        sched.step()
+    assert p.factor_update_steps == 1  # unchanged on failure
+
+
 def test_scheduler_rejects_callable_params() -> None:
     model = TinyModel()
     p = KFACPreconditioner(model, damping=lambda s: 0.01)
@@ -467,3 +481,37 @@ def test_grouped_conv_skipped_gracefully() -> None:
         m(x), torch.randint(0, 4, (2,)),
     ).backward()
     p.step()  # completes; conv trains unpreconditioned
+
+
+def test_dilated_and_string_padded_conv2d_skipped() -> None:
+    """Conv2d with dilation != 1 or 'same'/'valid' padding is skipped with
+    a warning: the fused im2col models neither, and computing a wrong A
+    factor silently would be worse than not preconditioning."""
+    import warnings as _warnings
+
+    for conv in (
+        torch.nn.Conv2d(4, 8, 3, dilation=2),
+        torch.nn.Conv2d(4, 8, 3, padding='same'),
+    ):
+        m = torch.nn.Sequential(
+            conv, torch.nn.Flatten(), torch.nn.Linear(8 * 4 * 4, 2),
+        )
+        with _warnings.catch_warnings(record=True) as rec:
+            _warnings.simplefilter('always')
+            p = KFACPreconditioner(
+                m, factor_update_steps=1, inv_update_steps=1,
+            )
+        assert any('Conv2d supports' in str(w.message) for w in rec)
+        assert len(p._layers) == 1  # only the Linear
+
+
+def test_bucket_key_includes_average_flag() -> None:
+    """Tensors reduced with different average flags must not share a
+    bucket (the per-bucket scale is recorded at first open)."""
+    from kfac_amd.distributed import TorchDistributedCommunicator
+
+    tdc = TorchDistributedCommunicator()
+    a = torch.ones(4)
+    k1 = (tdc._group_key(None)) + (a.dtype, a.device, True)
+    k2 = (tdc._group_key(None)) + (a.dtype, a.device, False)
+    assert k1 != k2
