@@ -514,6 +514,38 @@ torch::Tensor precond_apply_grouped(
   return work.narrow(0, 1, 1);
 }
 
+// Raw MFMA GEMM: C = op(A) @ op(B), fp32 in/out.  split=true runs the
+// bf16x3 split-precision path (hi/lo bf16 decomposition, fp32-class
+// accuracy at bf16 MFMA rates).  Building block for the QDWH polar
+// iterations and the batched Cholesky host loops.
+torch::Tensor gemm(
+    torch::Tensor a,
+    torch::Tensor b,
+    bool ta,
+    bool tb,
+    bool split) {
+  check_gpu_contig(a, "a");
+  check_gpu_contig(b, "b");
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm expects 2D");
+  TORCH_CHECK(
+      a.scalar_type() == torch::kFloat32 &&
+          b.scalar_type() == torch::kFloat32,
+      "gemm: fp32 only");
+  const int M = (int)(ta ? a.size(1) : a.size(0));
+  const int K = (int)(ta ? a.size(0) : a.size(1));
+  const int Kb = (int)(tb ? b.size(1) : b.size(0));
+  const int N = (int)(tb ? b.size(0) : b.size(1));
+  TORCH_CHECK(K == Kb, "gemm: inner dims mismatch");
+  auto out = torch::empty(
+      {(long)M, (long)N},
+      torch::TensorOptions().device(a.device()).dtype(torch::kFloat32));
+  CHECK_OK(kfac::gemm_f32(
+      current_stream(a), out.data_ptr<float>(), a.data_ptr<float>(),
+      b.data_ptr<float>(), M, N, K, ta, tb, 0, nullptr, nullptr, 0.f,
+      split));
+  return out;
+}
+
 torch::Tensor precond_inverse(
     torch::Tensor grad,
     torch::Tensor a_inv,
@@ -822,6 +854,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("precond_eigen_fused", &precond_eigen_fused, "Kronecker precondition (prediv)");
   m.def("precond_eigen", &precond_eigen, "Kronecker precondition (dg/da)");
   m.def("precond_inverse", &precond_inverse, "G^-1 grad A^-1");
+  m.def(
+      "gemm",
+      &gemm,
+      "C = op(A) op(B), fp32; split=true -> bf16x3 MFMA path",
+      pybind11::arg("a"),
+      pybind11::arg("b"),
+      pybind11::arg("ta") = false,
+      pybind11::arg("tb") = false,
+      pybind11::arg("split") = true);
   m.def(
       "precond_apply_grouped",
       &precond_apply_grouped,

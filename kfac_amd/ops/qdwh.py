@@ -186,12 +186,18 @@ def _range_basis(
     spectrum) keeps such directions away from the cut; the refinements
     are defense in depth for stale hints.
     """
+    # Pass-1 ridge sized for a SQUARE Gaussian sketch: kappa(Y) ~ 2k with
+    # a heavy tail, so the fp32 Gram can be numerically indefinite at
+    # production sizes (observed at k ~ 1500).  A 1e-3-relative ridge
+    # keeps the Cholesky alive and bounds kappa of the resulting basis
+    # to ~sqrt(ridge)/sigma_min; the projector refinements + later
+    # passes scrub the ridge-induced error completely.
     y = 0.5 * (omega[:, :k] - u @ omega[:, :k])
-    q = _chol_qr(y.unsqueeze(0), ridge=1e-6).squeeze(0)
+    q = _chol_qr(y.unsqueeze(0), ridge=1e-3).squeeze(0)
     for _ in range(refinements):
         y = 0.5 * (q - u @ q)
-        q = _chol_qr(y.unsqueeze(0)).squeeze(0)
-    return q
+        q = _chol_qr(y.unsqueeze(0), ridge=1e-6).squeeze(0)
+    return _chol_qr(q.unsqueeze(0)).squeeze(0)
 
 
 def _complement_basis(
@@ -215,9 +221,9 @@ def _complement_basis(
         return y - q1 @ (q1.transpose(-1, -2) @ y)
 
     y = 0.5 * (omega[:, :m] + u @ omega[:, :m])
-    q = _chol_qr(proj_out(y).unsqueeze(0), ridge=1e-6).squeeze(0)
-    q = _chol_qr(proj_out(q).unsqueeze(0)).squeeze(0)
-    return q
+    q = _chol_qr(proj_out(y).unsqueeze(0), ridge=1e-3).squeeze(0)
+    q = _chol_qr(proj_out(q).unsqueeze(0), ridge=1e-6).squeeze(0)
+    return _chol_qr(proj_out(q).unsqueeze(0)).squeeze(0)
 
 
 def split_spectrum(
