@@ -38,6 +38,7 @@ hipError_t kl_clip_accum_t(hipStream_t, float*, const T*, const T*, long);
 hipError_t triu_pack_f32(hipStream_t, float*, const float*, int);
 hipError_t syevj_small_f32(hipStream_t, const float*, float*, float*, int, int, int, float);
 hipError_t triu_unpack_f32(hipStream_t, float*, const float*, int);
+hipError_t chol_diag_inv_f32(hipStream_t, float*, float*, int, int, int, int, float);
 
 }  // namespace kfac
 
@@ -514,6 +515,34 @@ torch::Tensor precond_apply_grouped(
   return work.narrow(0, 1, 1);
 }
 
+// One blocked-Cholesky diagonal step: factor the m x m block of each
+// matrix at (j, j) in LDS (lower, diag floored at eps) and write its
+// triangular inverse into dinv (B, 128, 128).  The O(n^3) panel /
+// trailing GEMMs around it are driven from Python (ops/blocked.py).
+void chol_diag_inv(
+    torch::Tensor a,
+    torch::Tensor dinv,
+    int64_t j,
+    int64_t m,
+    double eps) {
+  check_gpu_contig(a, "a");
+  check_gpu_contig(dinv, "dinv");
+  TORCH_CHECK(
+      a.dim() == 3 && a.size(1) == a.size(2), "a must be (B, n, n)");
+  TORCH_CHECK(a.scalar_type() == torch::kFloat32, "fp32 only");
+  const int B = (int)a.size(0);
+  const int n = (int)a.size(1);
+  TORCH_CHECK(m >= 1 && m <= 128, "block size must be in [1, 128]");
+  TORCH_CHECK(j >= 0 && j + m <= n, "block out of range");
+  TORCH_CHECK(
+      dinv.dim() == 3 && dinv.size(0) == B && dinv.size(1) == 128 &&
+          dinv.size(2) == 128,
+      "dinv must be (B, 128, 128)");
+  CHECK_OK(kfac::chol_diag_inv_f32(
+      current_stream(a), a.data_ptr<float>(), dinv.data_ptr<float>(), B, n,
+      (int)j, (int)m, (float)eps));
+}
+
 // Raw MFMA GEMM: C = op(A) @ op(B), fp32 in/out.  split=true runs the
 // bf16x3 split-precision path (hi/lo bf16 decomposition, fp32-class
 // accuracy at bf16 MFMA rates).  Building block for the QDWH polar
@@ -854,6 +883,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("precond_eigen_fused", &precond_eigen_fused, "Kronecker precondition (prediv)");
   m.def("precond_eigen", &precond_eigen, "Kronecker precondition (dg/da)");
   m.def("precond_inverse", &precond_inverse, "G^-1 grad A^-1");
+  m.def(
+      "chol_diag_inv",
+      &chol_diag_inv,
+      "in-LDS Cholesky + triangular inverse of one diagonal block",
+      pybind11::arg("a"),
+      pybind11::arg("dinv"),
+      pybind11::arg("j"),
+      pybind11::arg("m"),
+      pybind11::arg("eps") = 1e-30);
   m.def(
       "gemm",
       &gemm,

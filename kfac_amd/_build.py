@@ -20,6 +20,7 @@ SOURCES = [
     os.path.join(CSRC, 'syrk.hip'),
     os.path.join(CSRC, 'gemm.hip'),
     os.path.join(CSRC, 'eigh.hip'),
+    os.path.join(CSRC, 'chol.hip'),
     os.path.join(CSRC, 'binding.cpp'),
 ]
 HEADERS = [os.path.join(CSRC, 'common.h')]

@@ -136,15 +136,33 @@ def polar_sign(
 
         X <- (b/c) X + (a - b/c) * X (I + c X^2)^{-1}
 
-    with the capped dynamic (a, b, c) schedule.  All batched: one
-    symmetric GEMM, one Cholesky, two triangular solves per step.
+    with the capped dynamic (a, b, c) schedule.  All batched.
+
+    On GPU the inner solve runs the blocked-Cholesky engine
+    (ops/blocked.py: LDS diagonal-block kernel + batched GEMMs) instead
+    of rocSOLVER potrf/potrs (measured 2-4 TF).  Iterations with large
+    "c" (kappa(Z) ~ c) stay on exact-fp32 GEMMs; once c <= 100 the
+    well-conditioned tail runs the hipBLASLt xf32 split-precision path
+    (~2.5x fp32 throughput at ~4.5e-6 relative accuracy).
     """
+    from kfac_amd.ops import blocked
+
+    use_blocked = x.is_cuda and blocked._ext() is not None
     eye = torch.eye(x.size(-1), dtype=x.dtype, device=x.device)
     for a, b, c in _coeffs(l0, c_max):
-        z = torch.baddbmm(eye, x, x, beta=1.0, alpha=c)
-        w = torch.linalg.cholesky(z)
-        # y = X Z^{-1}  ==  solve Z y^T = X  (Z, X symmetric)
-        y = torch.cholesky_solve(x, w)
+        if use_blocked:
+            wide = c <= 100.0
+            with blocked.gemm_engine(wide):
+                z = torch.baddbmm(eye, x, x, beta=1.0, alpha=c)
+            y = blocked.spd_solve_right(
+                x, z, tf32_chol=wide, tf32_apply=wide,
+            )
+        else:
+            z = torch.baddbmm(eye, x, x, beta=1.0, alpha=c)
+            w = torch.linalg.cholesky(z)
+            # y = Z^{-1} X == (X Z^{-1})^T; Z, X symmetric and the
+            # iterate is symmetrized below, so the transpose is free.
+            y = torch.cholesky_solve(x, w)
         x = (b / c) * x + (a - b / c) * y
         x = 0.5 * (x + x.transpose(-1, -2))
     return x
@@ -157,7 +175,11 @@ def _chol_qr(y: torch.Tensor, ridge: float = 0.0) -> torch.Tensor:
     relative ``ridge`` keeps the Gram positive definite when the sketch
     may be ill-conditioned (first pass of a complement basis); the
     second pass scrubs the ridge-induced error."""
-    g = y.transpose(-1, -2) @ y
+    from kfac_amd.ops import blocked
+
+    use_blocked = y.is_cuda and blocked._ext() is not None
+    with blocked.gemm_engine(use_blocked):
+        g = y.transpose(-1, -2) @ y
     if ridge > 0.0:
         scale = torch.diagonal(g, dim1=-2, dim2=-1).mean(
             dim=-1, keepdim=True,
@@ -165,6 +187,12 @@ def _chol_qr(y: torch.Tensor, ridge: float = 0.0) -> torch.Tensor:
         g = g + (ridge * scale).unsqueeze(-1) * torch.eye(
             g.size(-1), dtype=g.dtype, device=g.device,
         )
+    if use_blocked:
+        l, dinvs = blocked.potrf_batched(g, tf32=True, keep_dinv=True)
+        t = blocked.trinv_batched(l, dinvs or None, tf32=True)
+        with blocked.gemm_engine(True):
+            # Y R^{-1} with R = L^T:  Y L^{-T} = Y T^T
+            return y @ t.transpose(-1, -2)
     r = torch.linalg.cholesky(g, upper=True)
     return torch.linalg.solve_triangular(r, y, upper=True, left=False)
 
@@ -255,23 +283,27 @@ def split_spectrum(
     ks = torch.round((n - tr_u) / 2.0).long().clamp(1, n - 1)
     ks_host = ks.tolist()  # one host sync per split level
 
+    from kfac_amd.ops import blocked
+
     omega = torch.randn(n, n, dtype=a.dtype, device=a.device, generator=generator)
     a_lo: list[torch.Tensor] = []
     q_lo: list[torch.Tensor] = []
     a_hi: list[torch.Tensor] = []
     q_hi: list[torch.Tensor] = []
-    for i in range(bsz):
-        k = int(ks_host[i])
-        q1 = _range_basis(u[i], omega, k)
-        q2 = _complement_basis(u[i], omega, n - k, q1)
-        t1 = a[i] @ q1
-        m1 = q1.transpose(-1, -2) @ t1
-        t2 = a[i] @ q2
-        m2 = q2.transpose(-1, -2) @ t2
-        a_lo.append(0.5 * (m1 + m1.transpose(-1, -2)))
-        q_lo.append(q1)
-        a_hi.append(0.5 * (m2 + m2.transpose(-1, -2)))
-        q_hi.append(q2)
+    use_wide = a.is_cuda and blocked._ext() is not None
+    with blocked.gemm_engine(use_wide):
+        for i in range(bsz):
+            k = int(ks_host[i])
+            q1 = _range_basis(u[i], omega, k)
+            q2 = _complement_basis(u[i], omega, n - k, q1)
+            t1 = a[i] @ q1
+            m1 = q1.transpose(-1, -2) @ t1
+            t2 = a[i] @ q2
+            m2 = q2.transpose(-1, -2) @ t2
+            a_lo.append(0.5 * (m1 + m1.transpose(-1, -2)))
+            q_lo.append(q1)
+            a_hi.append(0.5 * (m2 + m2.transpose(-1, -2)))
+            q_hi.append(q2)
     return a_lo, q_lo, a_hi, q_hi
 
 
@@ -363,13 +395,20 @@ def eigh_qdwh(
                 hints = torch.stack([jobs[i][3] for i in idxs])
             sigma = _median_shift(sub, hints)
             a_lo, q_lo, a_hi, q_hi = split_spectrum(sub, sigma, generator)
+            from kfac_amd.ops import blocked as _blocked
+
+            _wide = sub.is_cuda and _blocked._ext() is not None
             for j, i in enumerate(idxs):
                 mat_i, q_parent, _, hint_i = jobs[i]
                 for a_c, q_c, lo in (
                     (a_lo[j], q_lo[j], True),
                     (a_hi[j], q_hi[j], False),
                 ):
-                    q_full = q_c if q_parent is None else q_parent @ q_c
+                    if q_parent is None:
+                        q_full = q_c
+                    else:
+                        with _blocked.gemm_engine(_wide):
+                            q_full = q_parent @ q_c
                     h_c = None
                     if hint_i is not None:
                         s = float(sigma[j])
@@ -395,12 +434,19 @@ def eigh_qdwh(
     by_size = {}
     for idx, job in enumerate(jobs):
         by_size.setdefault(job[2].size(-1), []).append(idx)
+    from kfac_amd.ops import blocked as _blk
+
     for sz, idxs in by_size.items():
         sub = torch.stack([jobs[i][2] for i in idxs])
         w_leaf, v_leaf = leaf_fn(sub)
+        _wide = sub.is_cuda and _blk._ext() is not None
         for j, i in enumerate(idxs):
             mat_i, q_parent, _, _ = jobs[i]
-            vec = v_leaf[j] if q_parent is None else q_parent @ v_leaf[j]
+            if q_parent is None:
+                vec = v_leaf[j]
+            else:
+                with _blk.gemm_engine(_wide):
+                    vec = q_parent @ v_leaf[j]
             k = vec.size(-1)
             off = fill[mat_i]
             w_out[mat_i, off : off + k] = w_leaf[j]

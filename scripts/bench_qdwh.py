@@ -76,22 +76,34 @@ def main() -> None:
     torch.cuda.set_device(device)
     print(f'extension available: {ops.extension_available()}')
 
+    from kfac_amd.ops import blocked
+
     print('\n== primitive timings (fp32, batched) ==')
     for b, n in [(3, 4608), (24, 3072), (6, 2304), (14, 1024)]:
         x = torch.randn(b, n, n, device=device)
         s = x @ x.transpose(-1, -2) + n * torch.eye(n, device=device)
         t_bmm = timed(lambda: x @ x)
+        with blocked.gemm_engine(True):
+            t_bmm_w = timed(lambda: x @ x)
         t_chol = timed(lambda: torch.linalg.cholesky(s))
         l = torch.linalg.cholesky(s)
         t_solve = timed(lambda: torch.cholesky_solve(x, l))
-        t_trsm = timed(
-            lambda: torch.linalg.solve_triangular(l, x, upper=False),
+        t_bchol = timed(lambda: blocked.potrf_batched(s, tf32=True))
+        lb, dinvs = blocked.potrf_batched(s, tf32=True, keep_dinv=True)
+        t_btri = timed(
+            lambda: blocked.trinv_batched(lb, dinvs, tf32=True),
+        )
+        t_bsolve = timed(
+            lambda: blocked.spd_solve_right(x, s),
         )
         tf = 2 * b * n ** 3 / (t_bmm / 1e3) / 1e12
+        tfw = 2 * b * n ** 3 / (t_bmm_w / 1e3) / 1e12
         print(
-            f'  {b}x{n}: bmm {t_bmm:8.2f} ms ({tf:6.1f} TF) | '
-            f'potrf {t_chol:8.2f} ms | potrs {t_solve:8.2f} ms | '
-            f'trsm {t_trsm:8.2f} ms',
+            f'  {b}x{n}: bmm {t_bmm:7.2f} ms ({tf:5.1f} TF) '
+            f'xf32 {t_bmm_w:7.2f} ms ({tfw:5.1f} TF) | '
+            f'potrf {t_chol:7.2f} -> {t_bchol:7.2f} ms | '
+            f'trinv {t_btri:7.2f} ms | '
+            f'potrs {t_solve:7.2f} -> solve {t_bsolve:7.2f} ms',
         )
 
     print('\n== leaf solver calibration (syevd batched) ==')
