@@ -523,9 +523,11 @@ class BaseKFACPreconditioner:
                                 [f.to(torch.float32) for _, f in items],
                             )
                             if stack.is_cuda:
-                                from kfac_amd import ops as _ops
-
-                                d, q = _ops.eigh_batched(stack)
+                                d, q = BaseKFACPreconditioner._group_eigh(
+                                    stack,
+                                    [lyr for lyr, _ in items],
+                                    which,
+                                )
                             else:
                                 d, q = torch.linalg.eigh(stack)
                             d = torch.clamp(d, min=0.0)
@@ -690,6 +692,63 @@ class BaseKFACPreconditioner:
                 else:
                     layer.g_inv = result
 
+    # Force a dense (syevd) re-anchor after this many consecutive warm
+    # phases, bounding any slow random walk of Q's orthogonality.
+    _WARM_REFRESH = 32
+
+    @staticmethod
+    def _group_eigh(
+        stack: torch.Tensor,
+        group: list[Any],
+        which: str,
+    ) -> tuple[torch.Tensor, torch.Tensor]:
+        """Eigendecompose one same-size factor group.
+
+        Warm path (GPU, n >= 512): K-FAC recomputes slowly-drifting EMA
+        factors, so the previous phase's eigenbasis (layer.qa / .qg)
+        makes T = Q^T F' Q near-diagonal; the adaptive block-Jacobi
+        (ops/warm_eigh.py) finishes in 0-2 sweeps of a few block pairs
+        — measured 1e-6..3e-2 off-diagonal mass on real ResNet-50
+        trajectories (profiles/jacobi_warm.md).  Cold starts, bad warm
+        starts and the periodic re-anchor run rocSOLVER syevd (n > 64)
+        or the LDS Jacobi kernel (n <= 64) via ops.eigh_batched.
+        """
+        import os
+
+        from kfac_amd import ops as _ops
+
+        n = stack.size(-1)
+        attr_q = 'qa' if which == 'a' else 'qg'
+        attr_cnt = f'_warm_phases_{which}'
+        prev = [getattr(layer, attr_q, None) for layer in group]
+        warm_enabled = (
+            stack.is_cuda
+            and n >= 512
+            and os.environ.get('KFAC_AMD_WARM_EIGH', '1') == '1'
+            and all(
+                isinstance(q, torch.Tensor)
+                and q.shape == (n, n)
+                and getattr(layer, attr_cnt, 0)
+                < BaseKFACPreconditioner._WARM_REFRESH
+                for q, layer in zip(prev, group)
+            )
+        )
+        if warm_enabled:
+            from kfac_amd.ops.warm_eigh import warm_eigh_batched
+
+            q_prev = torch.stack([q.to(torch.float32) for q in prev])
+            d, q, ok = warm_eigh_batched(stack, q_prev)
+            if ok:
+                for layer in group:
+                    setattr(
+                        layer, attr_cnt, getattr(layer, attr_cnt, 0) + 1,
+                    )
+                return d, q
+        d, q = _ops.eigh_batched(stack)
+        for layer in group:
+            setattr(layer, attr_cnt, 0)
+        return d, q
+
     @staticmethod
     def _batched_eigh(layers: list[Any], which: str) -> None:
         """Group same-size factors, eigendecompose each group in one call."""
@@ -720,13 +779,7 @@ class BaseKFACPreconditioner:
                     for layer in group
                 ],
             )
-            # n <= 64: hand-written LDS-resident batched Jacobi (one
-            # launch per group); larger n: rocSOLVER syevd behind a
-            # hipGraph replay. Single-layer CUDA groups go through the
-            # same path so they share the graph cache.
-            from kfac_amd import ops as _ops
-
-            d, q = _ops.eigh_batched(stack)
+            d, q = BaseKFACPreconditioner._group_eigh(stack, group, which)
             d = torch.clamp(d, min=0.0)
             for i, layer in enumerate(group):
                 qv = q[i].to(layer.inv_dtype).contiguous()

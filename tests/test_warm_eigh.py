@@ -1,0 +1,122 @@
+"""Numerics of the warm-started adaptive block-Jacobi eigensolver
+(CPU path; the LDS-subproblem route is covered by GPU tests)."""
+
+from __future__ import annotations
+
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from kfac_amd.ops.warm_eigh import warm_eigh_batched  # noqa: E402
+
+
+def drifted_pair(n, seed, rot=0.15, val=0.2):
+    """(F_prev, F_next) with a small basis rotation + eigenvalue drift."""
+    g = torch.Generator().manual_seed(seed)
+    q, _ = torch.linalg.qr(torch.randn(n, n, generator=g, dtype=torch.float64))
+    vals = torch.logspace(-4, 0, n, dtype=torch.float64)
+    f0 = (q * vals) @ q.T
+    s = torch.randn(n, n, generator=g, dtype=torch.float64) * rot / n ** 0.5
+    s = 0.5 * (s - s.T)
+    qd = q @ torch.matrix_exp(s)
+    vals2 = vals * (
+        1.0 + val * (torch.rand(n, generator=g, dtype=torch.float64) - 0.5)
+    )
+    f1 = (qd * vals2) @ qd.T
+    return f0.to(torch.float32), f1.to(torch.float32)
+
+
+def gates(f_next, d, q, rec_tol=2e-4, orth_tol=5e-5):
+    a64 = f_next.to(torch.float64)
+    q64 = q.to(torch.float64)
+    rec = (q64 * d.to(torch.float64)) @ q64.transpose(-1, -2)
+    rec_err = float(torch.linalg.norm(rec - a64) / torch.linalg.norm(a64))
+    n = f_next.size(-1)
+    eye = torch.eye(n, dtype=torch.float64)
+    orth = float(
+        torch.linalg.norm(q64.transpose(-1, -2) @ q64 - eye) / n ** 0.5,
+    )
+    assert rec_err < rec_tol, rec_err
+    assert orth < orth_tol, orth
+
+
+@pytest.mark.parametrize('n', [96, 200, 257])
+def test_warm_converges_small_drift(n: int) -> None:
+    f0, f1 = drifted_pair(n, seed=n)
+    _, q0 = torch.linalg.eigh(f0)
+    d, q, ok = warm_eigh_batched(
+        f1.unsqueeze(0), q0.unsqueeze(0), b=32,
+    )
+    assert ok
+    gates(f1, d.squeeze(0), q.squeeze(0))
+
+
+def test_warm_batch_mixed_drift() -> None:
+    n = 160
+    pairs = [drifted_pair(n, seed=7, rot=0.02), drifted_pair(n, seed=8, rot=0.2)]
+    f1 = torch.stack([p[1] for p in pairs])
+    q0 = torch.stack([torch.linalg.eigh(p[0])[1] for p in pairs])
+    d, q, ok = warm_eigh_batched(f1, q0, b=32)
+    assert ok
+    for i in range(2):
+        gates(f1[i], d[i], q[i])
+
+
+def test_warm_zero_drift_is_noop_fast() -> None:
+    n = 128
+    f0, _ = drifted_pair(n, seed=3)
+    _, q0 = torch.linalg.eigh(f0)
+    d, q, ok = warm_eigh_batched(f0.unsqueeze(0), q0.unsqueeze(0), b=32)
+    assert ok
+    gates(f0, d.squeeze(0), q.squeeze(0))
+
+
+def test_warm_bails_on_cold_start() -> None:
+    n = 128
+    f0, _ = drifted_pair(n, seed=11)
+    _, f1 = drifted_pair(n, seed=12, rot=1.0)  # unrelated basis
+    _, q0 = torch.linalg.eigh(f0)
+    _, _, ok = warm_eigh_batched(f1.unsqueeze(0), q0.unsqueeze(0), b=32)
+    assert not ok
+
+
+def test_warm_identity_q_on_diagonal_matrix() -> None:
+    # T already diagonal: zero rounds, exact result
+    n = 96
+    d_true = torch.linspace(0.1, 2.0, n)
+    f = torch.diag(d_true)
+    q0 = torch.eye(n)
+    d, q, ok = warm_eigh_batched(f.unsqueeze(0), q0.unsqueeze(0), b=32)
+    assert ok
+    torch.testing.assert_close(
+        d.squeeze(0).sort().values, d_true, atol=1e-5, rtol=1e-5,
+    )
+    gates(f, d.squeeze(0), q.squeeze(0))
+
+
+def test_warm_multiple_pairs_same_matrix_cross_blocks() -> None:
+    """Regression: two pairs of the SAME matrix in one round must apply
+    both rotations to their cross blocks (row pass + column pass read
+    order).  Construct coupling that forces >= 2 simultaneous pairs."""
+    torch.manual_seed(0)
+    n, b = 128, 32
+    t = torch.diag(torch.linspace(0.5, 2.0, n).to(torch.float32))
+    # strong coupling between blocks (0,1) and (2,3)
+    t[0:32, 32:64] = 0.03 * torch.randn(32, 32)
+    t[64:96, 96:128] = 0.03 * torch.randn(32, 32)
+    t = 0.5 * (t + t.T)
+    q0 = torch.eye(n)
+    d, q, ok = warm_eigh_batched(t.unsqueeze(0), q0.unsqueeze(0), b=b)
+    assert ok
+    gates(t, d.squeeze(0), q.squeeze(0))
+    w_ref = torch.linalg.eigvalsh(t.to(torch.float64))
+    torch.testing.assert_close(
+        d.squeeze(0).sort().values.to(torch.float64),
+        w_ref,
+        atol=1e-4,
+        rtol=1e-4,
+    )

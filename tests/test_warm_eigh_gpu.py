@@ -1,0 +1,126 @@
+"""GPU gates + integration for the warm-started block-Jacobi path."""
+
+from __future__ import annotations
+
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():  # pragma: no cover
+    pytest.skip('requires a GPU', allow_module_level=True)
+
+
+def drifted_batch(bsz, n, seed, rot=0.1):
+    g = torch.Generator(device='cuda').manual_seed(seed)
+    w = torch.randn(bsz, n, 2 * n, device='cuda', generator=g)
+    f0 = (w @ w.transpose(-1, -2)) / (2 * n)
+    f0 = f0 + torch.diag(torch.logspace(-4, 0, n, device='cuda')).unsqueeze(0)
+    f0 = 0.5 * (f0 + f0.transpose(-1, -2))
+    # EMA-like drift toward a fresh covariance
+    w2 = torch.randn(bsz, n, 2 * n, device='cuda', generator=g)
+    c = (w2 @ w2.transpose(-1, -2)) / (2 * n)
+    f1 = (1 - rot) * f0 + rot * c
+    f1 = 0.5 * (f1 + f1.transpose(-1, -2))
+    return f0, f1
+
+
+def gates(f, d, q, rec_tol=1.5e-4, orth_tol=5e-5):
+    a64 = f.to(torch.float64)
+    q64 = q.to(torch.float64)
+    rec = (q64 * d.to(torch.float64).unsqueeze(1)) @ q64.transpose(-1, -2)
+    rec_err = (
+        torch.linalg.norm(rec - a64, dim=(-2, -1))
+        / torch.linalg.norm(a64, dim=(-2, -1))
+    ).max()
+    n = f.size(-1)
+    eye = torch.eye(n, dtype=torch.float64, device=f.device)
+    orth = (
+        torch.linalg.norm(
+            q64.transpose(-1, -2) @ q64 - eye, dim=(-2, -1),
+        )
+        / n ** 0.5
+    ).max()
+    assert float(rec_err) < rec_tol, float(rec_err)
+    assert float(orth) < orth_tol, float(orth)
+
+
+@pytest.mark.parametrize('n', [512, 1024, 1025])
+def test_warm_eigh_gpu_gates(n: int) -> None:
+    from kfac_amd import ops
+    from kfac_amd.ops.warm_eigh import warm_eigh_batched
+
+    f0, f1 = drifted_batch(3, n, seed=n)
+    _, q0 = ops.eigh_batched(f0)
+    d, q, ok = warm_eigh_batched(f1, q0, b=32)
+    assert ok
+    gates(f1, d, q)
+    # eigenvalues agree with a dense solve after sorting
+    w_ref = torch.linalg.eigvalsh(f1.to(torch.float64))
+    err = (
+        (d.sort(dim=-1).values.to(torch.float64) - w_ref).abs().max()
+        / w_ref.abs().max()
+    )
+    assert float(err) < 2e-4, float(err)
+
+
+def test_warm_path_used_in_preconditioner() -> None:
+    """Second inverse phase goes through the warm solver (counter
+    advances) and training stays sane."""
+    from kfac_amd import KFACPreconditioner
+
+    torch.manual_seed(3)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(768, 768),
+        torch.nn.ReLU(),
+        torch.nn.Linear(768, 10),
+    ).cuda()
+    precon = KFACPreconditioner(
+        model,
+        factor_update_steps=1,
+        inv_update_steps=2,
+        lr=0.1,
+        inv_update_async=False,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    for _ in range(5):
+        opt.zero_grad(set_to_none=True)
+        x = torch.randn(32, 768, device='cuda')
+        y = torch.randint(0, 10, (32,), device='cuda')
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        assert torch.isfinite(loss)
+    big = [
+        layer
+        for _, (name, layer) in precon._layers.items()
+        if layer.module.a_factor_shape[0] >= 512
+    ]
+    assert big, 'expected a factor >= 512'
+    assert any(getattr(layer, '_warm_phases_a', 0) > 0 for layer in big)
+
+
+def test_warm_bail_falls_back_cleanly() -> None:
+    """A garbage warm basis must not poison the result: the group falls
+    back to syevd inside _group_eigh."""
+    from kfac_amd.base_preconditioner import BaseKFACPreconditioner
+
+    class _Dummy:
+        pass
+
+    f0, f1 = drifted_batch(2, 512, seed=9, rot=0.05)
+    layers = []
+    for _ in range(2):
+        d = _Dummy()
+        d.qa = torch.linalg.qr(torch.randn(512, 512, device='cuda'))[0]
+        layers.append(d)
+    # unrelated random orthogonal bases -> off0 is huge -> bail -> syevd
+    dvals, q = BaseKFACPreconditioner._group_eigh(f1, layers, 'a')
+    gates(f1, dvals, q)
+    assert all(layer._warm_phases_a == 0 for layer in layers)
