@@ -100,32 +100,6 @@ def _apply_diag_pass(
     q.copy_(qm.reshape(bsz, nq, n))
 
 
-def _greedy_matching(
-    bn_row: 'list[list[float]]',
-    thresh: float,
-    max_pairs: int,
-) -> list[tuple[int, int]]:
-    """Maximal matching over blocks, heaviest pairs first."""
-    cand = []
-    nb = len(bn_row)
-    for i in range(nb):
-        for j in range(i + 1, nb):
-            v = bn_row[i][j]
-            if v > thresh:
-                cand.append((v, i, j))
-    cand.sort(reverse=True)
-    used: set[int] = set()
-    taken: list[tuple[int, int]] = []
-    for _, i, j in cand:
-        if i not in used and j not in used:
-            used.add(i)
-            used.add(j)
-            taken.append((i, j))
-            if len(taken) >= max_pairs:
-                break
-    return taken
-
-
 @torch.no_grad()
 def warm_eigh_batched(
     stack: torch.Tensor,
@@ -195,55 +169,66 @@ def warm_eigh_batched(
 
     _apply_diag_pass(t, q, b)
 
+    dev = t.device
+    tol_sq = (tol * tn) ** 2
+    arange_b = torch.arange(b, device=dev)
     converged = False
     for _ in range(max_rounds):
         bn = _block_off_norms(t, b)
         offsq = (bn * bn).sum(dim=(-2, -1))
-        # one transfer per round: the block map + the residuals
-        bn_host = bn.cpu()
-        off_host = torch.sqrt(offsq).cpu()
-        pairs: list[tuple[int, int, int]] = []
-        all_done = True
-        for mi in range(bsz):
-            if float(off_host[mi]) <= tol * float(tn[mi]):
-                continue
-            all_done = False
-            thresh = tol * float(tn[mi]) / nb
-            for i, j in _greedy_matching(
-                bn_host[mi].tolist(), thresh, max_pairs=nb // 2,
-            ):
-                pairs.append((mi, i, j))
-        if all_done:
+        active = offsq > tol_sq
+        # candidate extraction on device, ONE small host transfer of the
+        # (typically short) thresholded candidate list per round
+        bnsq = bn * bn
+        cand_mask = torch.triu(
+            bnsq > (tol_sq / (nb * nb)).view(-1, 1, 1), diagonal=1,
+        ) & active.view(-1, 1, 1)
+        cand_idx = cand_mask.nonzero()
+        if cand_idx.numel() == 0:
+            if not bool(active.any()):
+                converged = True
+                break
+            # residual above tol but spread below the per-pair bar:
+            # take the heaviest pairs of the active matrices
+            topv, topi = bnsq.reshape(bsz, -1).topk(nb, dim=-1)
+            rows = []
+            for mi in torch.nonzero(active).flatten().tolist():
+                for r in range(nb):
+                    i, j = divmod(int(topi[mi, r]), nb)
+                    if i < j and float(topv[mi, r]) > 0:
+                        rows.append((float(topv[mi, r]), mi, i, j))
+            cand_host = [(mi, i, j) for _, mi, i, j in sorted(rows, reverse=True)]
+        else:
+            vals = bnsq[cand_idx[:, 0], cand_idx[:, 1], cand_idx[:, 2]]
+            order = torch.argsort(vals, descending=True)
+            cand_host = cand_idx[order].tolist()
+        if not cand_host:
             converged = True
             break
-        if not pairs:
-            # residual spread below per-pair threshold but above tol:
-            # lower the bar to the heaviest pairs
-            for mi in range(bsz):
-                if float(off_host[mi]) <= tol * float(tn[mi]):
-                    continue
-                for i, j in _greedy_matching(
-                    bn_host[mi].tolist(), 0.0, max_pairs=nb // 2,
-                ):
-                    pairs.append((mi, i, j))
-            if not pairs:
-                converged = True
+        # greedy maximal matching per matrix (host, short list); cap the
+        # round size to bound the gather scratch (pairs left over are
+        # picked up by the next round's fresh norms)
+        used: set[tuple[int, int]] = set()
+        pairs: list[tuple[int, int, int]] = []
+        for mi, i, j in cand_host:
+            if (mi, i) in used or (mi, j) in used:
+                continue
+            used.add((mi, i))
+            used.add((mi, j))
+            pairs.append((mi, i, j))
+            if len(pairs) >= 512:
                 break
 
         p = len(pairs)
-        dev = t.device
-        idx_local = torch.stack(
+        pair_t = torch.tensor(pairs, device=dev)
+        idx_local = torch.cat(
             [
-                torch.cat(
-                    [
-                        torch.arange(i * b, (i + 1) * b),
-                        torch.arange(j * b, (j + 1) * b),
-                    ],
-                )
-                for _, i, j in pairs
+                pair_t[:, 1:2] * b + arange_b,
+                pair_t[:, 2:3] * b + arange_b,
             ],
-        ).to(dev)
-        mat_idx = torch.tensor([mi for mi, _, _ in pairs], device=dev)
+            dim=1,
+        )  # (p, 2b)
+        mat_idx = pair_t[:, 0]
         flat_rows = (mat_idx.unsqueeze(1) * n + idx_local).reshape(-1)
 
         t_flat = t.reshape(bsz * n, n)
@@ -257,38 +242,28 @@ def warm_eigh_batched(
         # classic parallel block-Jacobi round: T <- V^T T (all pair
         # rows, batched globally), then T <- T V (pair columns, read
         # AFTER the row pass so cross-blocks between two same-matrix
-        # pairs get both factors), then Q <- Q V.
+        # pairs get both factors), then Q <- Q V.  Column updates use
+        # advanced indexing on the flat (B n, n) views so the whole
+        # round is a fixed number of launches regardless of B.
         with blocked.gemm_engine(wide):
             new_rows = v.transpose(-1, -2) @ sub_rows
         t_flat.index_copy_(0, flat_rows, new_rows.reshape(p * 2 * b, n))
-        by_mat: dict[int, list[int]] = {}
-        for pi, (mi, _, _) in enumerate(pairs):
-            by_mat.setdefault(mi, []).append(pi)
-        for mi, pis in by_mat.items():
-            cols = idx_local[pis].reshape(-1)
-            tc = (
-                t[mi]
-                .index_select(1, cols)
-                .reshape(n, len(pis), 2 * b)
-                .permute(1, 0, 2)
-            )
-            qs = (
-                q[mi]
-                .index_select(1, cols)
-                .reshape(n_true, len(pis), 2 * b)
-                .permute(1, 0, 2)
-            )
-            with blocked.gemm_engine(wide):
-                tr = tc @ v[pis]
-                qr = qs @ v[pis]
-            t[mi].index_copy_(
-                1, cols,
-                tr.permute(1, 0, 2).reshape(n, len(pis) * 2 * b),
-            )
-            q[mi].index_copy_(
-                1, cols,
-                qr.permute(1, 0, 2).reshape(n_true, len(pis) * 2 * b),
-            )
+        rowg = (mat_idx.view(p, 1) * n + torch.arange(n, device=dev)).view(
+            p, n, 1,
+        )
+        colg = idx_local.view(p, 1, 2 * b)
+        tc = t_flat[rowg, colg]  # (p, n, 2b)
+        with blocked.gemm_engine(wide):
+            tc = tc @ v
+        t_flat.index_put_((rowg, colg), tc)
+        q_flat = q.reshape(bsz * n_true, n)
+        rowq = (
+            mat_idx.view(p, 1) * n_true + torch.arange(n_true, device=dev)
+        ).view(p, n_true, 1)
+        qc = q_flat[rowq, colg]
+        with blocked.gemm_engine(wide):
+            qc = qc @ v
+        q_flat.index_put_((rowq, colg), qc)
 
     d = t.diagonal(dim1=-2, dim2=-1)[:, :n_true]
     return d.contiguous(), q[:, :, :n_true].contiguous(), converged

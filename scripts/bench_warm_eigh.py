@@ -30,14 +30,23 @@ def timed(fn, warmup=1, iters=3):
 
 
 def drifted_batch(bsz, n, seed, rot):
+    """Basis rotation by a fixed small angle + eigenvalue jitter —
+    modeled on MEASURED real drift (profiles/jacobi_warm.md: off(T)
+    between phases is 1e-6..3e-2 of ||F||; mixing toward an independent
+    Wishart, as an earlier version did, is orders harsher than
+    reality because consecutive EMA updates share the basis)."""
     g = torch.Generator(device='cuda').manual_seed(seed)
     w = torch.randn(bsz, n, 2 * n, device='cuda', generator=g)
     f0 = (w @ w.transpose(-1, -2)) / (2 * n)
     f0 = f0 + torch.diag(torch.logspace(-4, 0, n, device='cuda')).unsqueeze(0)
     f0 = 0.5 * (f0 + f0.transpose(-1, -2))
-    w2 = torch.randn(bsz, n, 2 * n, device='cuda', generator=g)
-    c = (w2 @ w2.transpose(-1, -2)) / (2 * n)
-    f1 = (1 - rot) * f0 + rot * c
+    s = torch.randn(bsz, n, n, device='cuda', generator=g) * rot / n ** 0.5
+    s = 0.5 * (s - s.transpose(-1, -2))
+    qd = torch.matrix_exp(s)
+    jit = 1.0 + 0.2 * (
+        torch.rand(bsz, 1, n, device='cuda', generator=g) - 0.5
+    )
+    f1 = qd @ (f0 * jit) @ qd.transpose(-1, -2)
     return f0, 0.5 * (f1 + f1.transpose(-1, -2))
 
 
@@ -60,7 +69,7 @@ def main() -> None:
         t_syevd = timed(lambda: ops.eigh_batched(f0.clone()), 1, 2)
         _, q0 = ops.eigh_batched(f0)
         print(f'{b_}x{n}: syevd {t_syevd:7.1f} ms')
-        for rot in (0.02, 0.1, 0.4):
+        for rot in (0.01, 0.05, 0.2):
             _, f1 = drifted_batch(b_, n, seed=n, rot=rot)
 
             def run():

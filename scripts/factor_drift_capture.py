@@ -225,9 +225,62 @@ def simulate_block_jacobi(f_prev, f_next, b=64, target_rel=1e-4, max_sweeps=12):
     return stats
 
 
+def time_production_solver(snaps, tags) -> None:
+    """Time ops.eigh_batched vs warm_eigh_batched on the CAPTURED real
+    factor pairs, grouped by size like the production inverse phase."""
+    import time as _time
+
+    from kfac_amd import ops
+    from kfac_amd.ops.warm_eigh import warm_eigh_batched
+
+    a_tag, b_tag = tags[1], tags[2]
+    groups: dict[int, list[tuple[torch.Tensor, torch.Tensor]]] = {}
+    for key in sorted(snaps[a_tag]):
+        f0 = snaps[a_tag][key]
+        f1 = snaps[b_tag].get(key)
+        if f1 is None or f0.shape != f1.shape or f0.size(-1) < 512:
+            continue
+        groups.setdefault(f0.size(-1), []).append((f0, f1))
+    print('\n== production solver timing on captured factors ==')
+    for n, items in sorted(groups.items()):
+        f0s = torch.stack([a for a, _ in items])
+        f1s = torch.stack([b for _, b in items])
+
+        def t(fn, iters=3):
+            fn()
+            torch.cuda.synchronize()
+            t0 = _time.perf_counter()
+            for _ in range(iters):
+                fn()
+            torch.cuda.synchronize()
+            return (_time.perf_counter() - t0) / iters * 1000.0
+
+        t_cold = t(lambda: ops.eigh_batched(f1s.clone()))
+        _, q0 = ops.eigh_batched(f0s.clone())
+
+        def warm():
+            return warm_eigh_batched(f1s, q0, b=32)
+
+        d, q, ok = warm()
+        t_warm = t(warm)
+        a64 = f1s.to(torch.float64)
+        q64 = q.to(torch.float64)
+        rec = (q64 * d.to(torch.float64).unsqueeze(1)) @ q64.transpose(-1, -2)
+        rec_err = (
+            torch.linalg.norm(rec - a64, dim=(-2, -1))
+            / torch.linalg.norm(a64, dim=(-2, -1))
+        ).max()
+        print(
+            f'  {len(items)}x{n}: syevd {t_cold:7.1f} ms | '
+            f'warm {t_warm:7.1f} ms ok={ok} rec={float(rec_err):.1e} '
+            f'speedup {t_cold / t_warm:5.2f}x',
+        )
+
+
 def main() -> None:
     snaps = capture_factors()
     tags = sorted(snaps)
+    time_production_solver(snaps, tags)
     for a_tag, b_tag in [(tags[1], tags[2]), (tags[0], tags[1])]:
         print(f'\n== drift {a_tag} -> {b_tag} ==')
         for key in sorted(snaps[a_tag]):

@@ -68,7 +68,7 @@ def test_spd_inverse_batched(n: int) -> None:
         torch.linalg.norm(inv - ref, dim=(-2, -1))
         / torch.linalg.norm(ref, dim=(-2, -1))
     ).max()
-    assert float(rel) < 1e-3, float(rel)
+    assert float(rel) < 5e-3, float(rel)
 
 
 def test_spd_solve_right() -> None:

@@ -88,10 +88,14 @@ def test_warm_path_used_in_preconditioner() -> None:
         inv_update_async=False,
     )
     opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    # fixed batch: keeps the factor eigenbasis stable between phases so
+    # the warm path is deterministically taken (tiny fresh batches give
+    # rank-32 covariances whose bases churn — a legitimate bail).
+    # Numerics under drift are covered by test_warm_eigh_gpu_gates.
+    x = torch.randn(512, 768, device='cuda')
+    y = torch.randint(0, 10, (512,), device='cuda')
     for _ in range(5):
         opt.zero_grad(set_to_none=True)
-        x = torch.randn(32, 768, device='cuda')
-        y = torch.randint(0, 10, (32,), device='cuda')
         loss = torch.nn.functional.cross_entropy(model(x), y)
         loss.backward()
         precon.step()
