@@ -108,7 +108,7 @@ def warm_eigh_batched(
     b: int = 32,
     tol: float = 1e-4,
     bail_rel: float = 0.25,
-    max_rounds: int = 150,
+    max_rounds: int = 24,
 ) -> tuple[torch.Tensor, torch.Tensor, bool]:
     """Batched warm-started eigendecomposition.
 
@@ -154,7 +154,12 @@ def warm_eigh_batched(
         qp[:, :, :n_true] = q_prev
         q = qp
     else:
-        q = q_prev.clone()
+        # contiguous_format is load-bearing: eigh_batched returns the
+        # eigenvector matrix as a TRANSPOSED view (rocSOLVER writes
+        # V^T row-major), and a strides-preserving clone would make
+        # q.reshape(...) below a silent COPY — every index_put_ update
+        # of Q would then be dropped.
+        q = q_prev.clone(memory_format=torch.contiguous_format)
 
     tn = torch.linalg.norm(stack.reshape(bsz, -1), dim=-1).clamp_min(1e-30)
     nb = n // b
