@@ -720,9 +720,17 @@ class BaseKFACPreconditioner:
         n = stack.size(-1)
         attr_q = 'qa' if which == 'a' else 'qg'
         attr_cnt = f'_warm_phases_{which}'
+        attr_cd = f'_warm_cooldown_{which}'
         prev = [getattr(layer, attr_q, None) for layer in group]
+        cooldown = max(
+            (getattr(layer, attr_cd, 0) for layer in group), default=0,
+        )
+        if cooldown > 0:
+            for layer in group:
+                setattr(layer, attr_cd, cooldown - 1)
         warm_enabled = (
-            stack.is_cuda
+            cooldown == 0
+            and stack.is_cuda
             and n >= 512
             and os.environ.get('KFAC_AMD_WARM_EIGH', '1') == '1'
             and all(
@@ -744,6 +752,12 @@ class BaseKFACPreconditioner:
                         layer, attr_cnt, getattr(layer, attr_cnt, 0) + 1,
                     )
                 return d, q
+            # bad warm start or no convergence within budget: skip the
+            # warm attempt for this group for a few phases (the heavy
+            # groups — e.g. degenerate identity-decay clusters — tend
+            # to stay heavy between adjacent phases).
+            for layer in group:
+                setattr(layer, attr_cd, 3)
         d, q = _ops.eigh_batched(stack)
         for layer in group:
             setattr(layer, attr_cnt, 0)
