@@ -638,18 +638,19 @@ void kl_clip_accum(
       });
 }
 
-// Batched symmetric eigensolver via rocSOLVER syevd, with the whole
-// ~50k-launch tridiagonalization sequence captured into a hipGraph the
-// first time each (batch, n) shape appears and REPLAYED on later inverse
-// phases. rocprofv3 shows the syevd phase is launch-bound (latrd/larfg
-// kernels of 3-12 us each; the host cannot feed them fast enough), which
-// is exactly the case hipGraphs exist for — K-FAC recomputes the same
-// factor shapes every inv_update_steps, so one capture amortizes over
-// the whole run. Falls back to direct rocSOLVER calls if capture fails
-// (or KFAC_AMD_NO_EIGH_GRAPH=1).
+// Batched symmetric eigensolver via rocSOLVER syevd.
+//
+// NOTE (round 2): the round-1 version captured the ~50k-launch
+// tridiagonalization sequence into a hipGraph and replayed it on later
+// phases.  That is UNSOUND: syevd's tridiagonal eigeniteration issues a
+// data-dependent number of kernels, so a captured sequence replayed on
+// different data computes garbage — observed as rec errors ~1e2 on
+// small G-factor groups whose spectra needed more iterations than the
+// capture-time data (profiles/jacobi_warm.md).  The capture machinery
+// is deleted; the warm-started block-Jacobi path (ops/warm_eigh.py)
+// supersedes its performance role, and this direct call serves cold
+// starts and fallbacks.
 struct SyevdEntry {
-  hipGraphExec_t exec = nullptr;
-  bool tried_capture = false;
   torch::Tensor a;     // persistent input/output (B, n, n)
   torch::Tensor w;     // eigenvalues (B, n)
   torch::Tensor e;     // tridiagonal workspace (B, n)
@@ -699,11 +700,6 @@ std::tuple<torch::Tensor, torch::Tensor> syevd_batched(torch::Tensor stack) {
 
   const auto key = std::make_tuple((int)stack.device().index(), B, n);
   auto it = cache.find(key);
-  const bool no_graph = [] {
-    const char* env = getenv("KFAC_AMD_NO_EIGH_GRAPH");
-    return env != nullptr && env[0] == '1';
-  }();
-
   if (it == cache.end()) {
     SyevdEntry ent;
     ent.a = torch::empty_like(stack);
@@ -717,53 +713,8 @@ std::tuple<torch::Tensor, torch::Tensor> syevd_batched(torch::Tensor stack) {
   SyevdEntry& ent = it->second;
   ent.a.copy_(stack);
   rocblas_set_stream(handle, stream);
-
-  if (ent.exec == nullptr && !ent.tried_capture && !no_graph) {
-    ent.tried_capture = true;
-    // Warmup call sizes the rocblas device workspace so the capture pass
-    // performs no allocations (allocation during capture aborts it).
-    auto st = run_syevd(handle, ent, n, B);
-    if (st == rocblas_status_success) {
-      ent.a.copy_(stack);
-      hipStream_t cap;
-      if (hipStreamCreateWithFlags(&cap, hipStreamNonBlocking) ==
-          hipSuccess) {
-        rocblas_set_stream(handle, cap);
-        hipGraph_t graph = nullptr;
-        bool ok =
-            hipStreamBeginCapture(cap, hipStreamCaptureModeThreadLocal) ==
-            hipSuccess;
-        if (ok) {
-          ok = run_syevd(handle, ent, n, B) == rocblas_status_success;
-          if (hipStreamEndCapture(cap, &graph) != hipSuccess) ok = false;
-        }
-        if (ok && graph != nullptr) {
-          hipGraphExec_t exec = nullptr;
-          if (hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) ==
-              hipSuccess) {
-            ent.exec = exec;
-          }
-        }
-        if (graph != nullptr) (void)hipGraphDestroy(graph);
-        rocblas_set_stream(handle, stream);
-        (void)hipStreamDestroy(cap);
-        (void)hipGetLastError();  // clear any capture-abort residue
-      }
-      if (ent.exec == nullptr) {
-        // capture failed; the warmup result in ent is stale — recompute
-        // below via the direct path.
-        ent.a.copy_(stack);
-      }
-    }
-  }
-
-  if (ent.exec != nullptr) {
-    CHECK_OK(hipGraphLaunch(ent.exec, stream));
-  } else {
-    auto st = run_syevd(handle, ent, n, B);
-    TORCH_CHECK(
-        st == rocblas_status_success, "rocsolver syevd failed: ", st);
-  }
+  auto st = run_syevd(handle, ent, n, B);
+  TORCH_CHECK(st == rocblas_status_success, "rocsolver syevd failed: ", st);
   // Column-major eigenvectors: the row-major clone holds V^T per matrix.
   return {ent.w.clone(), ent.a.clone()};
 }

@@ -746,18 +746,38 @@ class BaseKFACPreconditioner:
 
             q_prev = torch.stack([q.to(torch.float32) for q in prev])
             d, q, ok = warm_eigh_batched(stack, q_prev)
-            if ok:
+            ok_host = ok.tolist()
+            if all(ok_host):
                 for layer in group:
                     setattr(
                         layer, attr_cnt, getattr(layer, attr_cnt, 0) + 1,
                     )
                 return d, q
-            # bad warm start or no convergence within budget: skip the
-            # warm attempt for this group for a few phases (the heavy
-            # groups — e.g. degenerate identity-decay clusters — tend
-            # to stay heavy between adjacent phases).
-            for layer in group:
-                setattr(layer, attr_cd, 3)
+            # per-matrix fallback: dense-solve only the unconverged
+            # matrices (bad warm start / budget exceeded), keep the
+            # converged ones; failed layers skip warm attempts for a
+            # few phases (heavy groups — e.g. degenerate identity-decay
+            # clusters re-mixing — tend to stay heavy).
+            bad = [i for i, okv in enumerate(ok_host) if not okv]
+            d2, q2 = _ops.eigh_batched(
+                stack[bad].contiguous()
+                if len(bad) > 1
+                else stack[bad[0]].unsqueeze(0).contiguous(),
+            )
+            d = d.clone()
+            q = q.contiguous().clone()
+            for k, i in enumerate(bad):
+                d[i] = d2[k]
+                q[i] = q2[k]
+            for i, layer in enumerate(group):
+                if ok_host[i]:
+                    setattr(
+                        layer, attr_cnt, getattr(layer, attr_cnt, 0) + 1,
+                    )
+                else:
+                    setattr(layer, attr_cd, 3)
+                    setattr(layer, attr_cnt, 0)
+            return d, q
         d, q = _ops.eigh_batched(stack)
         for layer in group:
             setattr(layer, attr_cnt, 0)

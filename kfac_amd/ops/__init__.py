@@ -273,11 +273,14 @@ def eigh_batched(stack: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
 
     On CUDA: n <= 64 routes to the hand-written one-wave-per-matrix LDS
     Jacobi kernel (eigenvalues unsorted — K-FAC is order-invariant);
-    larger n routes to rocSOLVER syevd whose ~50k-launch
-    tridiagonalization sequence is captured into a hipGraph on first use
-    per shape and replayed on later inverse phases (the phase is
-    launch-bound, see profiles/eigh_strategies.md). Eigenvalues are NOT
-    clamped here; callers clamp >= 0 (reference eigen.py:321,344).
+    larger n routes to rocSOLVER syevd.  This is the COLD path: the
+    preconditioner's inverse phase uses the warm-started block-Jacobi
+    solver (ops/warm_eigh.py) whenever the previous phase's eigenbasis
+    is available.  (Round 1 wrapped syevd in a hipGraph replay; that
+    was removed in round 2 after it was shown unsound — syevd's
+    tridiagonal iteration launches a data-dependent kernel sequence,
+    see profiles/jacobi_warm.md.)  Eigenvalues are NOT clamped here;
+    callers clamp >= 0 (reference eigen.py:321,344).
     """
     if stack.is_cuda:
         ext = _load_ext()

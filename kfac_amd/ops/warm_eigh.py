@@ -124,9 +124,10 @@ def warm_eigh_batched(
     Returns:
         (d, q, converged): eigenvalue estimates ``d = diag(T)``
         (UNSORTED — aligned with q's columns, which stay maximally
-        close to ``q_prev``'s order), eigenvectors ``q``, and whether
-        every matrix met ``tol``.  On ``converged=False`` the caller
-        should fall back to a dense solve.
+        close to ``q_prev``'s order), eigenvectors ``q``, and a
+        per-matrix boolean mask of which matrices met ``tol``.  The
+        caller solves the unconverged ones densely; converged entries
+        are final.
     """
     bsz, n_true, _ = stack.shape
     from kfac_amd.ops import blocked
@@ -164,21 +165,27 @@ def warm_eigh_batched(
     tn = torch.linalg.norm(stack.reshape(bsz, -1), dim=-1).clamp_min(1e-30)
     nb = n // b
 
-    # quick bail on a bad warm start (one host sync)
+    # per-matrix bail on a bad warm start (one host sync); the rest of
+    # the batch proceeds — one cold matrix must not sink its group.
     off0 = torch.linalg.norm(
         (t - torch.diag_embed(t.diagonal(dim1=-2, dim2=-1))).reshape(bsz, -1),
         dim=-1,
     )
-    if bool((off0 > bail_rel * tn).any()):
-        return t.diagonal(dim1=-2, dim2=-1)[:, :n_true], q[:, :, :n_true], False
+    failed = off0 > bail_rel * tn
+    if bool(failed.all()):
+        return (
+            t.diagonal(dim1=-2, dim2=-1)[:, :n_true],
+            q[:, :, :n_true],
+            ~failed,
+        )
 
     _apply_diag_pass(t, q, b)
 
     dev = t.device
     tol_sq = (tol * tn) ** 2
     arange_b = torch.arange(b, device=dev)
-    converged = False
     entry_offsq: torch.Tensor | None = None
+    offsq = tol_sq  # placeholder; overwritten in the loop
     for rnd in range(max_rounds):
         bn = _block_off_norms(t, b)
         offsq = (bn * bn).sum(dim=(-2, -1))
@@ -187,11 +194,13 @@ def warm_eigh_batched(
         elif rnd in (8, 16):
             # progress check: heavy-tail cases (re-mixing degenerate
             # clusters produce star-shaped couplings that fragment the
-            # matching) burn rounds without converging — hand them to
-            # the dense solver early instead of grinding the budget.
-            if bool(((offsq / entry_offsq) > 0.16).any()):
-                break
-        active = offsq > tol_sq
+            # matching) burn rounds without converging — hand those
+            # matrices to the dense solver instead of grinding the
+            # budget for the whole batch.
+            failed = failed | (
+                (offsq > tol_sq) & ((offsq / entry_offsq) > 0.16)
+            )
+        active = (offsq > tol_sq) & ~failed
         # candidate extraction on device, ONE small host transfer of the
         # (typically short) thresholded candidate list per round
         bnsq = bn * bn
@@ -201,7 +210,6 @@ def warm_eigh_batched(
         cand_idx = cand_mask.nonzero()
         if cand_idx.numel() == 0:
             if not bool(active.any()):
-                converged = True
                 break
             # residual above tol but spread below the per-pair bar:
             # take the heaviest pairs of the active matrices
@@ -218,7 +226,6 @@ def warm_eigh_batched(
             order = torch.argsort(vals, descending=True)
             cand_host = cand_idx[order].tolist()
         if not cand_host:
-            converged = True
             break
         # greedy maximal matching per matrix (host, short list); cap the
         # round size to bound the gather scratch (pairs left over are
@@ -280,5 +287,8 @@ def warm_eigh_batched(
             qc = qc @ v
         q_flat.index_put_((rowq, colg), qc)
 
+    bnf = _block_off_norms(t, b)
+    final_off = (bnf * bnf).sum(dim=(-2, -1))
+    mask = ~failed & (final_off <= tol_sq * 1.0001)
     d = t.diagonal(dim1=-2, dim2=-1)[:, :n_true]
-    return d.contiguous(), q[:, :, :n_true].contiguous(), converged
+    return d.contiguous(), q[:, :, :n_true].contiguous(), mask

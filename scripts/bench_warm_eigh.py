@@ -78,7 +78,7 @@ def main() -> None:
             d, q, ok = run()
             t_warm = timed(run, 1, 2)
             print(
-                f'   warm rot={rot:0.2f}: {t_warm:7.1f} ms ok={ok} '
+                f'   warm rot={rot:0.2f}: {t_warm:7.1f} ms ok={bool(ok.all())} '
                 f'rec={quality(f1, d, q):.1e} '
                 f'speedup {t_syevd / t_warm:5.2f}x',
             )

@@ -272,7 +272,7 @@ def time_production_solver(snaps, tags) -> None:
         ).max()
         print(
             f'  {len(items)}x{n}: syevd {t_cold:7.1f} ms | '
-            f'warm {t_warm:7.1f} ms ok={ok} rec={float(rec_err):.1e} '
+            f'warm {t_warm:7.1f} ms ok={bool(ok.all())} rec={float(rec_err):.1e} '
             f'speedup {t_cold / t_warm:5.2f}x',
         )
 
