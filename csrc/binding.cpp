@@ -39,6 +39,8 @@ hipError_t triu_pack_f32(hipStream_t, float*, const float*, int);
 hipError_t syevj_small_f32(hipStream_t, const float*, float*, float*, int, int, int, float);
 hipError_t triu_unpack_f32(hipStream_t, float*, const float*, int);
 hipError_t chol_diag_inv_f32(hipStream_t, float*, float*, int, int, int, int, float);
+hipError_t gather_cols_f32(hipStream_t, const float*, const long*, const long*, float*, int, int, int, int);
+hipError_t scatter_cols_f32(hipStream_t, float*, const long*, const long*, const float*, int, int, int, int);
 
 }  // namespace kfac
 
@@ -515,6 +517,55 @@ torch::Tensor precond_apply_grouped(
   return work.narrow(0, 1, 1);
 }
 
+// Batched column gather/scatter for the warm block-Jacobi rounds:
+// out[p] = t[mat[p]][:, idx[p]] without materializing index grids.
+torch::Tensor gather_cols(
+    torch::Tensor t,     // (B, rows, n) fp32 contiguous
+    torch::Tensor mat,   // (p,) int64
+    torch::Tensor idx) {  // (p, m) int64
+  check_gpu_contig(t, "t");
+  check_gpu_contig(mat, "mat");
+  check_gpu_contig(idx, "idx");
+  TORCH_CHECK(t.dim() == 3 && t.scalar_type() == torch::kFloat32, "t");
+  TORCH_CHECK(
+      mat.scalar_type() == torch::kInt64 &&
+          idx.scalar_type() == torch::kInt64,
+      "int64 indices required");
+  const int p = (int)mat.size(0);
+  const int rows = (int)t.size(1);
+  const int n = (int)t.size(2);
+  const int m = (int)idx.size(1);
+  auto out = torch::empty(
+      {(long)p, (long)rows, (long)m},
+      torch::TensorOptions().device(t.device()).dtype(torch::kFloat32));
+  CHECK_OK(kfac::gather_cols_f32(
+      current_stream(t), t.data_ptr<float>(), mat.data_ptr<long>(),
+      idx.data_ptr<long>(), out.data_ptr<float>(), p, rows, n, m));
+  return out;
+}
+
+void scatter_cols(
+    torch::Tensor t,
+    torch::Tensor mat,
+    torch::Tensor idx,
+    torch::Tensor src) {
+  check_gpu_contig(t, "t");
+  check_gpu_contig(mat, "mat");
+  check_gpu_contig(idx, "idx");
+  check_gpu_contig(src, "src");
+  const int p = (int)mat.size(0);
+  const int rows = (int)t.size(1);
+  const int n = (int)t.size(2);
+  const int m = (int)idx.size(1);
+  TORCH_CHECK(
+      src.dim() == 3 && src.size(0) == p && src.size(1) == rows &&
+          src.size(2) == m,
+      "src shape");
+  CHECK_OK(kfac::scatter_cols_f32(
+      current_stream(t), t.data_ptr<float>(), mat.data_ptr<long>(),
+      idx.data_ptr<long>(), src.data_ptr<float>(), p, rows, n, m));
+}
+
 // One blocked-Cholesky diagonal step: factor the m x m block of each
 // matrix at (j, j) in LDS (lower, diag floored at eps) and write its
 // triangular inverse into dinv (B, 128, 128).  The O(n^3) panel /
@@ -715,6 +766,47 @@ std::tuple<torch::Tensor, torch::Tensor> syevd_batched(torch::Tensor stack) {
   rocblas_set_stream(handle, stream);
   auto st = run_syevd(handle, ent, n, B);
   TORCH_CHECK(st == rocblas_status_success, "rocsolver syevd failed: ", st);
+  // syevd's tridiagonal QL iteration can FAIL to converge on hard
+  // spectra (observed on fully-decayed scalar-identity G factors with
+  // denormal off-diagonals): info != 0 and the outputs are garbage.
+  // Re-solve failed matrices with the (slower, robust) Jacobi solver.
+  auto info_host = ent.info.cpu();
+  bool any_failed = false;
+  for (int i = 0; i < B; ++i) {
+    if (info_host[i].item<int>() != 0) {
+      any_failed = true;
+      break;
+    }
+  }
+  if (any_failed) {
+    auto a2 = stack.clone();
+    auto residual = torch::empty({B}, stack.options());
+    auto opts_i =
+        torch::TensorOptions().device(stack.device()).dtype(torch::kInt32);
+    auto n_sweeps = torch::empty({B}, opts_i);
+    auto status = rocsolver_ssyevj_strided_batched(
+        handle,
+        rocblas_esort_ascending,
+        rocblas_evect_original,
+        rocblas_fill_upper,
+        n,
+        a2.data_ptr<float>(),
+        n,
+        (rocblas_stride)n * n,
+        0.0f,
+        residual.data_ptr<float>(),
+        (rocblas_int)100,
+        n_sweeps.data_ptr<int>(),
+        ent.w.data_ptr<float>(),
+        (rocblas_stride)n,
+        ent.info.data_ptr<int>(),
+        B);
+    TORCH_CHECK(
+        status == rocblas_status_success,
+        "rocsolver syevj fallback failed: ",
+        status);
+    return {ent.w.clone(), a2};
+  }
   // Column-major eigenvectors: the row-major clone holds V^T per matrix.
   return {ent.w.clone(), ent.a.clone()};
 }
@@ -834,6 +926,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("precond_eigen_fused", &precond_eigen_fused, "Kronecker precondition (prediv)");
   m.def("precond_eigen", &precond_eigen, "Kronecker precondition (dg/da)");
   m.def("precond_inverse", &precond_inverse, "G^-1 grad A^-1");
+  m.def(
+      "gather_cols",
+      &gather_cols,
+      "batched per-pair column gather (warm Jacobi rounds)");
+  m.def(
+      "scatter_cols",
+      &scatter_cols,
+      "batched per-pair column scatter (warm Jacobi rounds)");
   m.def(
       "chol_diag_inv",
       &chol_diag_inv,

@@ -117,4 +117,82 @@ hipError_t chol_diag_inv_f32(
   return hipGetLastError();
 }
 
+// ---- column gather/scatter for the warm block-Jacobi rounds ----
+//
+// The rotation rounds update pair COLUMNS of T (and Q) across the whole
+// batch.  torch advanced indexing materializes p x rows x 2b int64
+// index grids (~160 MB per round at n=4608) — these kernels move only
+// the payload.  idx holds per-pair column indices (two contiguous
+// b-runs, so the global accesses coalesce).
+
+__global__ void gather_cols_kernel(
+    const float* __restrict__ t,    // (B, rows, n) flat
+    const long* __restrict__ mat,   // (p,)
+    const long* __restrict__ idx,   // (p, m)
+    float* __restrict__ out,        // (p, rows, m)
+    int rows,
+    int n,
+    int m) {
+  const long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)gridDim.x * blockDim.x;
+  const long count = (long)rows * m;
+  const long pi = blockIdx.y;
+  for (long i = e; i < count; i += total) {
+    const int r = (int)(i / m);
+    const int c = (int)(i % m);
+    out[(pi * rows + r) * (long)m + c] =
+        t[(mat[pi] * rows + r) * (long)n + idx[pi * m + c]];
+  }
+}
+
+__global__ void scatter_cols_kernel(
+    float* __restrict__ t,
+    const long* __restrict__ mat,
+    const long* __restrict__ idx,
+    const float* __restrict__ src,  // (p, rows, m)
+    int rows,
+    int n,
+    int m) {
+  const long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long total = (long)gridDim.x * blockDim.x;
+  const long count = (long)rows * m;
+  const long pi = blockIdx.y;
+  for (long i = e; i < count; i += total) {
+    const int r = (int)(i / m);
+    const int c = (int)(i % m);
+    t[(mat[pi] * rows + r) * (long)n + idx[pi * m + c]] =
+        src[(pi * rows + r) * (long)m + c];
+  }
+}
+
+hipError_t gather_cols_f32(
+    hipStream_t stream,
+    const float* t,
+    const long* mat,
+    const long* idx,
+    float* out,
+    int p,
+    int rows,
+    int n,
+    int m) {
+  dim3 grid(64, p);
+  gather_cols_kernel<<<grid, 256, 0, stream>>>(t, mat, idx, out, rows, n, m);
+  return hipGetLastError();
+}
+
+hipError_t scatter_cols_f32(
+    hipStream_t stream,
+    float* t,
+    const long* mat,
+    const long* idx,
+    const float* src,
+    int p,
+    int rows,
+    int n,
+    int m) {
+  dim3 grid(64, p);
+  scatter_cols_kernel<<<grid, 256, 0, stream>>>(t, mat, idx, src, rows, n, m);
+  return hipGetLastError();
+}
+
 }  // namespace kfac

@@ -270,22 +270,41 @@ def warm_eigh_batched(
         with blocked.gemm_engine(wide):
             new_rows = v.transpose(-1, -2) @ sub_rows
         t_flat.index_copy_(0, flat_rows, new_rows.reshape(p * 2 * b, n))
-        rowg = (mat_idx.view(p, 1) * n + torch.arange(n, device=dev)).view(
-            p, n, 1,
-        )
-        colg = idx_local.view(p, 1, 2 * b)
-        tc = t_flat[rowg, colg]  # (p, n, 2b)
-        with blocked.gemm_engine(wide):
-            tc = tc @ v
-        t_flat.index_put_((rowg, colg), tc)
-        q_flat = q.reshape(bsz * n_true, n)
-        rowq = (
-            mat_idx.view(p, 1) * n_true + torch.arange(n_true, device=dev)
-        ).view(p, n_true, 1)
-        qc = q_flat[rowq, colg]
-        with blocked.gemm_engine(wide):
-            qc = qc @ v
-        q_flat.index_put_((rowq, colg), qc)
+        ext = None
+        if wide:
+            from kfac_amd import ops as _ops
+
+            ext = _ops._load_ext()
+        if ext is not None:
+            # payload-only column moves (csrc/chol.hip): torch advanced
+            # indexing would materialize p x n x 2b int64 index grids
+            # (~160 MB/round at n=4608).
+            tc = ext.gather_cols(t, mat_idx, idx_local)
+            with blocked.gemm_engine(wide):
+                tc = (tc @ v).contiguous()
+            ext.scatter_cols(t, mat_idx, idx_local, tc)
+            qc = ext.gather_cols(q, mat_idx, idx_local)
+            with blocked.gemm_engine(wide):
+                qc = (qc @ v).contiguous()
+            ext.scatter_cols(q, mat_idx, idx_local, qc)
+        else:
+            rowg = (
+                mat_idx.view(p, 1) * n + torch.arange(n, device=dev)
+            ).view(p, n, 1)
+            colg = idx_local.view(p, 1, 2 * b)
+            tc = t_flat[rowg, colg]  # (p, n, 2b)
+            with blocked.gemm_engine(wide):
+                tc = tc @ v
+            t_flat.index_put_((rowg, colg), tc)
+            q_flat = q.reshape(bsz * n_true, n)
+            rowq = (
+                mat_idx.view(p, 1) * n_true
+                + torch.arange(n_true, device=dev)
+            ).view(p, n_true, 1)
+            qc = q_flat[rowq, colg]
+            with blocked.gemm_engine(wide):
+                qc = qc @ v
+            q_flat.index_put_((rowq, colg), qc)
 
     bnf = _block_off_norms(t, b)
     final_off = (bnf * bnf).sum(dim=(-2, -1))
