@@ -250,7 +250,7 @@ def warm_eigh_batched(
             ],
             dim=1,
         )  # (p, 2b)
-        mat_idx = pair_t[:, 0]
+        mat_idx = pair_t[:, 0].contiguous()
         flat_rows = (mat_idx.unsqueeze(1) * n + idx_local).reshape(-1)
 
         t_flat = t.reshape(bsz * n, n)
