@@ -602,6 +602,20 @@ class BaseKFACPreconditioner:
         from kfac_amd.layers.eigen import KFACEigenLayer
         from kfac_amd.layers.inverse import KFACInverseLayer
 
+        import os
+        import time as _time
+
+        trace = os.environ.get('KFAC_AMD_PHASE_TRACE', '0') == '1'
+
+        def _mark(label: str, t0: float) -> float:
+            if not trace:
+                return 0.0
+            torch.cuda.synchronize()
+            now = _time.perf_counter()
+            print(f'[kfac phase] {label}: {(now - t0) * 1e3:8.2f} ms')
+            return now
+
+        t0 = _time.perf_counter() if trace else 0.0
         rank = get_rank()
         eigen_a: list[KFACEigenLayer] = []
         eigen_g: list[KFACEigenLayer] = []
@@ -629,8 +643,11 @@ class BaseKFACPreconditioner:
                 other.append((name, layer))
 
         damping = self.damping
+        t0 = _mark('gather', t0)
         self._batched_eigh(eigen_a, 'a')
+        t0 = _mark('eigh A', t0)
         self._batched_eigh(eigen_g, 'g')
+        t0 = _mark('eigh G', t0)
         # prediv fusion: dgda on the G worker (requires colocated factors)
         for layer in eigen_g:
             if layer.prediv_eigenvalues:
@@ -640,6 +657,7 @@ class BaseKFACPreconditioner:
                 layer.dgda = 1 / (torch.outer(dg, da) + damping)
                 layer.dg = None
                 layer.da = None
+        t0 = _mark('dgda', t0)
         self._batched_cholesky_inverse(inverse_a, 'a', damping)
         self._batched_cholesky_inverse(inverse_g, 'g', damping)
         for name, layer in other:
@@ -647,6 +665,7 @@ class BaseKFACPreconditioner:
                 layer.compute_a_inv(damping=damping)
             if rank == self._assignment.inv_worker(name, 'G'):
                 layer.compute_g_inv(damping=damping)
+        _mark('inverse/other', t0)
 
     @staticmethod
     def _batched_cholesky_inverse(
@@ -797,6 +816,10 @@ class BaseKFACPreconditioner:
                     'been computed',
                 )
             groups[(factor.shape[0], factor.device, factor.dtype)].append(layer)
+        import os
+        import time as _time
+
+        trace = os.environ.get('KFAC_AMD_PHASE_TRACE', '0') == '1'
         for (n, dev, _dt), group in groups.items():
             if len(group) == 1 and dev.type != 'cuda':
                 layer = group[0]
@@ -805,6 +828,9 @@ class BaseKFACPreconditioner:
                 else:
                     layer.compute_g_inv_no_prediv()
                 continue
+            if trace:
+                torch.cuda.synchronize()
+                tg = _time.perf_counter()
             stack = torch.stack(
                 [
                     (layer.a_factor if which == 'a' else layer.g_factor).to(
@@ -814,6 +840,14 @@ class BaseKFACPreconditioner:
                 ],
             )
             d, q = BaseKFACPreconditioner._group_eigh(stack, group, which)
+            if trace:
+                torch.cuda.synchronize()
+                warm = getattr(group[0], f'_warm_phases_{which}', 0) > 0
+                print(
+                    f'[kfac phase]   {which.upper()} group '
+                    f'{len(group):3d}x{n:<5d} warm={warm} '
+                    f'{(_time.perf_counter() - tg) * 1e3:8.2f} ms',
+                )
             d = torch.clamp(d, min=0.0)
             for i, layer in enumerate(group):
                 qv = q[i].to(layer.inv_dtype).contiguous()
