@@ -108,7 +108,7 @@ def warm_eigh_batched(
     b: int = 32,
     tol: float = 1e-4,
     bail_rel: float = 0.25,
-    max_rounds: int = 24,
+    max_rounds: int = 40,
 ) -> tuple[torch.Tensor, torch.Tensor, bool]:
     """Batched warm-started eigendecomposition.
 
@@ -191,14 +191,17 @@ def warm_eigh_batched(
         offsq = (bn * bn).sum(dim=(-2, -1))
         if entry_offsq is None:
             entry_offsq = offsq.clamp_min(1e-30)
-        elif rnd in (8, 16):
-            # progress check: heavy-tail cases (re-mixing degenerate
-            # clusters produce star-shaped couplings that fragment the
-            # matching) burn rounds without converging — hand those
-            # matrices to the dense solver instead of grinding the
-            # budget for the whole batch.
+        elif rnd in (8, 16, 24, 32):
+            # progress check: matrices whose off mass is not shrinking
+            # go to the dense solver instead of grinding the budget.
+            # Thresholds tuned on real ResNet-50 factor groups
+            # (profiles/jacobi_warm.md): the legitimate heavy cases
+            # (identity-decay clusters re-mixing) converge in ~20-30
+            # rounds with steady 10-20%/round reduction, while truly
+            # cold starts stall near 1.0.
+            limit = {8: 0.25, 16: 0.04, 24: 6.4e-3, 32: 1e-3}[rnd]
             failed = failed | (
-                (offsq > tol_sq) & ((offsq / entry_offsq) > 0.16)
+                (offsq > tol_sq) & ((offsq / entry_offsq) > limit)
             )
         active = (offsq > tol_sq) & ~failed
         # candidate extraction on device, ONE small host transfer of the
