@@ -100,6 +100,76 @@ def _apply_diag_pass(
     q.copy_(qm.reshape(bsz, nq, n))
 
 
+def _dense_band_pass(
+    t: torch.Tensor,
+    q: torch.Tensor,
+    stack: torch.Tensor,
+    b: int,
+    tol: float,
+    tn: torch.Tensor,
+    failed: torch.Tensor,
+) -> None:
+    """Dense correction for clustered off-mass.
+
+    Real K-FAC factors concentrate inter-phase mixing in a contiguous
+    eigenvalue range (the decayed-identity cluster re-mixing with fresh
+    covariance directions, profiles/jacobi_warm.md).  Pairwise block
+    rotations need >= band-width/b rounds for such a clique; instead,
+    eigendecompose the active band ONCE (batched dense solve on the
+    w x w submatrix) and apply the rotation with GEMMs.  Scattered
+    leftovers are handled by the adaptive rounds afterwards.
+    """
+    bsz, n, _ = t.shape
+    nb = n // b
+    n_true = q.size(1)
+    bn = _block_off_norms(t, b)
+    thresh = (tol * tn).view(-1, 1, 1) / nb
+    act = bn > thresh
+    pair_counts = act.sum(dim=(-2, -1)) // 2
+    blk_active = act.any(dim=-1)
+    idx = torch.arange(nb, device=t.device)
+    lo = torch.where(blk_active, idx, nb).min(dim=-1).values
+    hi = torch.where(blk_active, idx, -1).max(dim=-1).values
+    width = (hi - lo + 1).clamp_min(0)
+    # dense when the active pairs are clique-like within a band that is
+    # worth solving densely (<= 60% of the matrix)
+    use = (
+        (pair_counts > 2 * width)
+        & (width * b <= int(0.6 * n))
+        & (width >= 4)
+        & ~failed
+    )
+    sel = torch.nonzero(use).flatten().tolist()
+    if not sel:
+        return
+    lo_h = lo.tolist()
+    hi_h = hi.tolist()
+    w = max((hi_h[i] - lo_h[i] + 1) * b for i in sel)
+    from kfac_amd import ops as _ops
+    from kfac_amd.ops import blocked
+
+    wide = t.is_cuda
+    subs = []
+    starts = []
+    for i in sel:
+        c0 = min(lo_h[i] * b, n - w)
+        starts.append(c0)
+        subs.append(t[i, c0 : c0 + w, c0 : c0 + w])
+    sub = torch.stack(subs).contiguous()
+    if t.is_cuda:
+        _, v = _ops.eigh_batched(sub)
+        v = v.contiguous()
+    else:
+        _, v = torch.linalg.eigh(sub)
+    for k, i in enumerate(sel):
+        c0 = starts[k]
+        with blocked.gemm_engine(wide):
+            rows = v[k].transpose(-1, -2) @ t[i, c0 : c0 + w, :]
+            t[i, c0 : c0 + w, :] = rows
+            t[i, :, c0 : c0 + w] = t[i, :, c0 : c0 + w] @ v[k]
+            q[i, :, c0 : c0 + w] = q[i, :, c0 : c0 + w] @ v[k]
+
+
 @torch.no_grad()
 def warm_eigh_batched(
     stack: torch.Tensor,
@@ -180,6 +250,7 @@ def warm_eigh_batched(
         )
 
     _apply_diag_pass(t, q, b)
+    _dense_band_pass(t, q, stack, b, tol, tn, failed)
 
     dev = t.device
     tol_sq = (tol * tn) ** 2
