@@ -843,9 +843,10 @@ class BaseKFACPreconditioner:
             if trace:
                 torch.cuda.synchronize()
                 warm = getattr(group[0], f'_warm_phases_{which}', 0) > 0
+                cd = getattr(group[0], f'_warm_cooldown_{which}', 0)
                 print(
                     f'[kfac phase]   {which.upper()} group '
-                    f'{len(group):3d}x{n:<5d} warm={warm} '
+                    f'{len(group):3d}x{n:<5d} warm={warm} cd={cd} '
                     f'{(_time.perf_counter() - tg) * 1e3:8.2f} ms',
                 )
             d = torch.clamp(d, min=0.0)
