@@ -740,6 +740,23 @@ class BaseKFACPreconditioner:
         attr_q = 'qa' if which == 'a' else 'qg'
         attr_cnt = f'_warm_phases_{which}'
         attr_cd = f'_warm_cooldown_{which}'
+        if stack.is_cuda:
+            # near-diagonal screen FIRST: vanished-gradient factors are
+            # scalar-identity EMAs; (diag, I) is the exact answer and
+            # cheaper than even the warm path's T-build GEMMs.
+            diag = stack.diagonal(dim1=-2, dim2=-1)
+            off = torch.linalg.norm(
+                (stack - torch.diag_embed(diag)).reshape(stack.size(0), -1),
+                dim=-1,
+            )
+            tnorm = torch.linalg.norm(
+                stack.reshape(stack.size(0), -1), dim=-1,
+            ).clamp_min(1e-30)
+            if bool((off <= 1e-7 * tnorm).all()):
+                eye = torch.eye(
+                    n, dtype=stack.dtype, device=stack.device,
+                ).expand_as(stack).contiguous()
+                return diag.clone(), eye
         prev = [getattr(layer, attr_q, None) for layer in group]
         cooldown = max(
             (getattr(layer, attr_cd, 0) for layer in group), default=0,

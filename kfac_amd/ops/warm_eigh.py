@@ -126,16 +126,19 @@ def _dense_band_pass(
     thresh = (tol * tn).view(-1, 1, 1) / nb
     act = bn > thresh
     pair_counts = act.sum(dim=(-2, -1)) // 2
-    blk_active = act.any(dim=-1)
+    # band = the contiguous range of HIGH-DEGREE blocks (clique
+    # members); a bounding box over all active pairs is inflated past
+    # the width cap by scattered outliers, which the adaptive rounds
+    # handle anyway.
+    degree = act.sum(dim=-1)
+    core = degree >= 3
     idx = torch.arange(nb, device=t.device)
-    lo = torch.where(blk_active, idx, nb).min(dim=-1).values
-    hi = torch.where(blk_active, idx, -1).max(dim=-1).values
+    lo = torch.where(core, idx, nb).min(dim=-1).values
+    hi = torch.where(core, idx, -1).max(dim=-1).values
     width = (hi - lo + 1).clamp_min(0)
-    # dense when the active pairs are clique-like within a band that is
-    # worth solving densely (<= 60% of the matrix)
     use = (
         (pair_counts > 2 * width)
-        & (width * b <= int(0.6 * n))
+        & (width * b <= int(0.7 * n))
         & (width >= 4)
         & ~failed
     )
