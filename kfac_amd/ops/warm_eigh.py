@@ -28,6 +28,8 @@ which also serves the cold first phase.  Replaces the reference's
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 __all__ = ['warm_eigh_batched']
@@ -142,6 +144,11 @@ def _dense_band_pass(
         & (width >= 4)
         & ~failed
     )
+    if os.environ.get('KFAC_AMD_WARM_TRACE', '0') == '1':
+        print(
+            f'[warm] n={n} b={b} pairs={pair_counts.tolist()} '
+            f'width={width.tolist()} use={use.tolist()}',
+        )
     sel = torch.nonzero(use).flatten().tolist()
     if not sel:
         return
@@ -319,6 +326,8 @@ def warm_eigh_batched(
                 break
 
         p = len(pairs)
+        if rnd < 3 and os.environ.get('KFAC_AMD_WARM_TRACE', '0') == '1':
+            print(f'[warm]   round {rnd}: {p} pairs')
         pair_t = torch.tensor(pairs, device=dev)
         idx_local = torch.cat(
             [
