@@ -149,6 +149,11 @@ def _dense_band_pass(
             f'[warm] n={n} b={b} pairs={pair_counts.tolist()} '
             f'width={width.tolist()} use={use.tolist()}',
         )
+    # fail fast on broad scattered drift at sizes where the dense
+    # solver is cheap: grinding ~40 rotation rounds only pays off for
+    # large n (measured: 3x4608 full-spread converges at 1.4x syevd,
+    # but 12x769 full-spread costs 3x MORE than syevd).
+    failed |= (~use) & (pair_counts > 3 * nb) & (n < 1536)
     sel = torch.nonzero(use).flatten().tolist()
     if not sel:
         return
