@@ -153,7 +153,8 @@ def _dense_band_pass(
     # solver is cheap: grinding ~40 rotation rounds only pays off for
     # large n (measured: 3x4608 full-spread converges at 1.4x syevd,
     # but 12x769 full-spread costs 3x MORE than syevd).
-    failed |= (~use) & (pair_counts > 3 * nb) & (n < 1536)
+    if t.is_cuda and n < 1536:
+        failed |= (~use) & (pair_counts > 3 * nb)
     sel = torch.nonzero(use).flatten().tolist()
     if not sel:
         return
