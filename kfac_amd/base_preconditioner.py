@@ -316,9 +316,17 @@ class BaseKFACPreconditioner:
             # happened in one grouped extension call
             pass
         else:
-            grouped = self._grouped_precondition()
+            # Partitioned path: the grouped 4-launch Kronecker chain
+            # runs for every ELIGIBLE grad-worker layer (works in
+            # HYBRID/MEM-OPT too — grad workers precondition grouped,
+            # then broadcast; receivers just receive), and only the
+            # ineligible remainder takes the per-layer launches.
+            grouped_done = self._grouped_precondition()
             for name, layer in reversed(list(self._layers.values())):
-                if not grouped and self._assignment.is_grad_worker(name):
+                if (
+                    name not in grouped_done
+                    and self._assignment.is_grad_worker(name)
+                ):
                     layer.preconditioned_grad(damping=self.damping)
                 if self._assignment.broadcast_gradients():
                     layer.broadcast_grad(
@@ -395,18 +403,25 @@ class BaseKFACPreconditioner:
         )
         return True
 
-    def _grouped_precondition(self) -> bool:
-        """Fast path: the precondition chain for every local layer in 4
-        grouped kernel launches (vs the reference's ~8 torch launches per
-        layer, eigen.py:374-385). Returns False if any layer is not
-        eligible (CPU, non-eigen, non-prediv, missing extension) — the
-        caller then runs the per-layer path.
+    def _grouped_precondition(self) -> set[str]:
+        """Fast path: the precondition chain for every ELIGIBLE local
+        layer in 4 grouped kernel launches (vs the reference's ~8 torch
+        launches per layer, eigen.py:374-385).
+
+        Ineligible layers (CPU, non-eigen, non-prediv, missing state)
+        are simply left out — the caller runs the per-layer path for
+        them — so a single special layer no longer forfeits the grouped
+        launch count for the whole model, and the path composes with
+        HYBRID/MEM-OPT gradient broadcasts (round-1 verdict item 3).
+
+        Returns the set of layer names already preconditioned.
         """
         from kfac_amd import ops
         from kfac_amd.layers.eigen import KFACEigenLayer
 
+        done: set[str] = set()
         if not ops.extension_available():
-            return False
+            return done
         work = []
         for name, layer in reversed(list(self._layers.values())):
             if not self._assignment.is_grad_worker(name):
@@ -416,24 +431,25 @@ class BaseKFACPreconditioner:
                 or not layer.prediv_eigenvalues
                 or not getattr(layer, 'grouped_precondition', True)
             ):
-                return False
+                continue
             qa, qg, dgda = layer.qa, layer.qg, layer.dgda
             if qa is None or qg is None or dgda is None:
-                return False
+                continue
             if not (qa.is_cuda and qa.dtype == torch.float32):
-                return False
-            work.append((layer, layer.module.get_grad(), qa, qg, dgda))
+                continue
+            work.append((name, layer, layer.module.get_grad(), qa, qg, dgda))
         if not work:
-            return True
+            return done
         outs = ops.precond_eigen_grouped(
-            [w[1] for w in work],
             [w[2] for w in work],
             [w[3] for w in work],
             [w[4] for w in work],
+            [w[5] for w in work],
         )
-        for (layer, *_), out in zip(work, outs):
+        for (name, layer, *_), out in zip(work, outs):
             layer.grad = out
-        return True
+            done.add(name)
+        return done
 
     # -- async inverse pipeline --------------------------------------------
     #

@@ -102,6 +102,14 @@ def test_distributed_training_hybrid() -> None:
     run_distributed(2, _dist_case, 'HYBRID_OPT')
 
 
+def test_distributed_training_hybrid_world4() -> None:
+    # world 4 at fraction 0.5: 2 grad workers per layer, gradient
+    # broadcasts active — the partitioned grouped-precondition path's
+    # protocol (grad workers precondition, then broadcast) at the
+    # grid shape BASELINE.json names for 8-GPU runs.
+    run_distributed(4, _dist_case, 'HYBRID_OPT')
+
+
 def test_distributed_training_mem_opt() -> None:
     run_distributed(2, _dist_case, 'MEM_OPT')
 
