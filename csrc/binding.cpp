@@ -33,6 +33,7 @@ hipError_t cov_trans_t(hipStream_t, const __hip_bfloat16*, int, int, int, bool, 
 hipError_t gemm_f32(hipStream_t, float*, const float*, const float*, int, int, int, bool, bool, int, const float*, const float*, float, bool);
 hipError_t precond_grouped_f32(hipStream_t, const void*, int, int);
 hipError_t precond_apply_grouped_f32(hipStream_t, const void*, int, int, float*, float*, long, float*, float*, float, float);
+hipError_t precond_apply_only_f32(hipStream_t, const void*, int, int, float*, float*, float, float);
 template <typename T>
 hipError_t kl_clip_accum_t(hipStream_t, float*, const T*, const T*, long);
 hipError_t triu_pack_f32(hipStream_t, float*, const float*, int);
@@ -626,6 +627,82 @@ torch::Tensor gemm(
   return out;
 }
 
+// Fused kl-clip + scaled in-place grad write for ALREADY-preconditioned
+// per-layer gradients (the HYBRID/MEM-OPT path: grad workers ran the
+// grouped chain, receivers got broadcasts; clip+apply remain).  accum
+// must be pre-filled with any per-layer contributions from layers not
+// in this call.  3 launches replace ~3 per layer.
+torch::Tensor apply_scaled_grouped(
+    std::vector<torch::Tensor> precons,
+    std::vector<torch::Tensor> wgrads,
+    std::vector<torch::Tensor> bgrads,
+    torch::Tensor accum,
+    double kl_clip,
+    double lr) {
+  const int L = (int)precons.size();
+  TORCH_CHECK(L > 0, "empty layer list");
+  TORCH_CHECK(
+      (int)wgrads.size() == L && (int)bgrads.size() == L, "length mismatch");
+  check_gpu_contig(accum, "accum");
+  TORCH_CHECK(accum.scalar_type() == torch::kFloat32, "accum fp32");
+  int64_t tiles = 0;
+  std::vector<int64_t> tile_offs(L);
+  for (int l = 0; l < L; ++l) {
+    check_gpu_contig(precons[l], "precon");
+    check_gpu_contig(wgrads[l], "wgrad");
+    TORCH_CHECK(
+        precons[l].scalar_type() == torch::kFloat32 &&
+            wgrads[l].scalar_type() == torch::kFloat32,
+        "fp32 required");
+    const int64_t m = precons[l].size(0);
+    const int64_t n = precons[l].size(1);
+    const bool has_bias = bgrads[l].numel() > 0;
+    TORCH_CHECK(
+        wgrads[l].numel() == m * (n - (has_bias ? 1 : 0)),
+        "wgrad size mismatch");
+    if (has_bias) {
+      check_gpu_contig(bgrads[l], "bgrad");
+      TORCH_CHECK(bgrads[l].numel() == m, "bias size mismatch");
+    }
+    tile_offs[l] = tiles;
+    tiles += (int64_t)((m + 127) / 128) * ((n + 127) / 128);
+  }
+  auto desc_cpu = torch::empty(
+      {L * (int64_t)(sizeof(PrecondDescHost) / 8)},
+      torch::TensorOptions().dtype(torch::kInt64).pinned_memory(true));
+  auto* d = (PrecondDescHost*)desc_cpu.data_ptr<int64_t>();
+  for (int l = 0; l < L; ++l) {
+    d[l].m = precons[l].size(0);
+    d[l].n = precons[l].size(1);
+    d[l].grad = nullptr;
+    d[l].qa = nullptr;
+    d[l].qg = nullptr;
+    d[l].dgda = nullptr;
+    d[l].s1 = nullptr;
+    d[l].s2 = nullptr;
+    d[l].out = precons[l].data_ptr<float>();
+    d[l].tile_off = tile_offs[l];
+    d[l].wgrad = wgrads[l].data_ptr<float>();
+    d[l].bgrad =
+        bgrads[l].numel() > 0 ? bgrads[l].data_ptr<float>() : nullptr;
+    d[l].spad = precons[l].size(1);
+  }
+  auto desc_dev = desc_cpu.to(precons[0].device(), /*non_blocking=*/true);
+  auto scale = torch::empty(
+      {1},
+      torch::TensorOptions().device(accum.device()).dtype(torch::kFloat32));
+  CHECK_OK(kfac::precond_apply_only_f32(
+      current_stream(accum),
+      desc_dev.data_ptr<int64_t>(),
+      L,
+      (int)tiles,
+      accum.data_ptr<float>(),
+      scale.data_ptr<float>(),
+      (float)kl_clip,
+      (float)lr));
+  return scale;
+}
+
 torch::Tensor precond_inverse(
     torch::Tensor grad,
     torch::Tensor a_inv,
@@ -952,6 +1029,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       pybind11::arg("ta") = false,
       pybind11::arg("tb") = false,
       pybind11::arg("split") = true);
+  m.def(
+      "apply_scaled_grouped",
+      &apply_scaled_grouped,
+      "fused kl-clip + scaled in-place apply for preconditioned grads");
   m.def(
       "precond_apply_grouped",
       &precond_apply_grouped,

@@ -475,6 +475,60 @@ __global__ __launch_bounds__(256) void grouped_grad_copy_kernel(
   }
 }
 
+// Grouped kl-clip dot product against the MODULE gradients: each tile
+// accumulates sum(precon .* [wgrad|bgrad]) into *accum.  Used by the
+// broadcast-composing apply path (HYBRID/MEM-OPT), where the precon
+// grads already exist per layer (grouped chain or received broadcast)
+// and only clip+scale+write-back remain to be fused.
+__global__ __launch_bounds__(256) void grouped_dot_kernel(
+    const PrecondDesc* __restrict__ desc,
+    int n_layers,
+    float* __restrict__ accum) {
+  const int tile = blockIdx.x;
+  int l = 0;
+  while (l + 1 < n_layers && tile >= (int)desc[l + 1].tile_off) ++l;
+  const PrecondDesc d = desc[l];
+  const int m = (int)d.m;
+  const int n = (int)d.n;
+  const int local = tile - (int)d.tile_off;
+  const int ntj = ceil_div(n, GBT);
+  const int i0 = (local / ntj) * GBT;
+  const int j0 = (local % ntj) * GBT;
+  const int n_w = (d.bgrad != nullptr) ? n - 1 : n;
+  float s = 0.0f;
+  for (int e = threadIdx.x; e < GBT * GBT; e += 256) {
+    const int row = i0 + e / GBT;
+    const int col = j0 + e % GBT;
+    if (row >= m || col >= n) continue;
+    const float p = d.out[(long)row * d.spad + col];
+    const float g =
+        (col < n_w) ? d.wgrad[(long)row * n_w + col] : d.bgrad[row];
+    s += p * g;
+  }
+  __shared__ float wsum[4];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+  }
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) wsum[wave] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    atomicAdd(accum, wsum[0] + wsum[1] + wsum[2] + wsum[3]);
+  }
+}
+
+hipError_t precond_apply_only_f32(
+    hipStream_t stream,
+    const void* desc_dev,
+    int n_layers,
+    int total_tiles,
+    float* accum,   // PRE-ACCUMULATED with any per-layer contributions
+    float* scale,
+    float kl_clip,
+    float lr);
+
 // scale = min(1, sqrt(kl_clip / |accum * lr^2|)); kl_clip <= 0 -> 1.
 __global__ void grad_scale_kernel(
     float* __restrict__ scale,
@@ -522,6 +576,24 @@ hipError_t precond_apply_grouped_f32(
   zero_kernel<<<1, 1, 0, stream>>>(accum);
   KFAC_HIP_CHECK(
       kl_clip_accum_t<float>(stream, accum, outbuf, gbuf, total_elems));
+  grad_scale_kernel<<<1, 1, 0, stream>>>(scale, accum, kl_clip, lr);
+  grouped_grad_copy_kernel<true>
+      <<<total_tiles, 256, 0, stream>>>(desc, n_layers, scale);
+  return hipGetLastError();
+}
+
+hipError_t precond_apply_only_f32(
+    hipStream_t stream,
+    const void* desc_dev,
+    int n_layers,
+    int total_tiles,
+    float* accum,
+    float* scale,
+    float kl_clip,
+    float lr) {
+  auto desc = (const PrecondDesc*)desc_dev;
+  grouped_dot_kernel<<<total_tiles, 256, 0, stream>>>(
+      desc, n_layers, accum);
   grad_scale_kernel<<<1, 1, 0, stream>>>(scale, accum, kl_clip, lr);
   grouped_grad_copy_kernel<true>
       <<<total_tiles, 256, 0, stream>>>(desc, n_layers, scale);

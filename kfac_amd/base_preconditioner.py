@@ -335,10 +335,13 @@ class BaseKFACPreconditioner:
                     )
             self._tdc.flush_allreduce_buckets()
 
-            scale = None if self.kl_clip is None else self._compute_grad_scale()
-
-            for _, layer in reversed(list(self._layers.values())):
-                layer.update_grad(scale=scale)
+            if self.kl_clip is None or not self._grouped_apply():
+                scale = (
+                    None if self.kl_clip is None
+                    else self._compute_grad_scale()
+                )
+                for _, layer in reversed(list(self._layers.values())):
+                    layer.update_grad(scale=scale)
 
         self._steps += 1
         self._mini_steps = defaultdict(int)
@@ -450,6 +453,81 @@ class BaseKFACPreconditioner:
             layer.grad = out
             done.add(name)
         return done
+
+    def _grouped_apply(self) -> bool:
+        """Fused kl-clip + scaled in-place grad write for the broadcast
+        path: after every layer's preconditioned gradient exists (from
+        the grouped chain or a received broadcast), 3 kernel launches
+        compute the global clip scale and write back all eligible
+        layers; ineligible layers contribute to the same device scale
+        and are written per-layer.  Returns False (caller runs the
+        per-layer path) only when no layer is eligible.
+        """
+        import os
+
+        from kfac_amd import ops
+
+        if not ops.extension_available():
+            return False
+        if os.environ.get('KFAC_AMD_NO_GROUPED_APPLY', '0') == '1':
+            return False
+        kl_clip = self.kl_clip
+        lr = self.lr
+        elig: list[tuple[Any, torch.Tensor, torch.Tensor, torch.Tensor]] = []
+        inelig: list[Any] = []
+        for name, layer in reversed(list(self._layers.values())):
+            g = layer.grad  # waits any in-flight broadcast future
+            if g is None:
+                return False
+            ok = (
+                g.is_cuda
+                and g.dtype == torch.float32
+                and g.is_contiguous()
+            )
+            wg = bg = None
+            if ok:
+                wg = layer.module.get_weight_grad()
+                ok = (
+                    wg is not None
+                    and wg.dtype == torch.float32
+                    and wg.is_contiguous()
+                )
+            if ok and layer.module.has_bias():
+                bg = layer.module.get_bias_grad()
+                ok = bg.dtype == torch.float32 and bg.is_contiguous()
+            if ok:
+                elig.append(
+                    (
+                        layer,
+                        g,
+                        wg.view(wg.size(0), -1),
+                        bg if bg is not None else g.new_empty(0),
+                    ),
+                )
+            else:
+                inelig.append(layer)
+        if not elig:
+            return False
+        device = elig[0][1].device
+        accum = torch.zeros((1,), dtype=torch.float32, device=device)
+        for layer in inelig:
+            ops.kl_clip_accum(
+                accum[0], layer.grad, layer.module.get_grad(),
+            )
+        ext = ops._load_ext()
+        scale = ext.apply_scaled_grouped(
+            [e[1] for e in elig],
+            [e[2] for e in elig],
+            [e[3] for e in elig],
+            accum,
+            float(kl_clip),
+            float(lr),
+        )
+        for layer, *_ in elig:
+            layer.grad = None  # consumed by the fused write
+        for layer in inelig:
+            layer.update_grad(scale=scale[0])
+        return True
 
     # -- async inverse pipeline --------------------------------------------
     #

@@ -222,3 +222,47 @@ def test_partitioned_grouped_precondition_mixed_set() -> None:
             atol=1e-6,
             msg=lambda m: f'{name}: {m}',
         )
+
+
+def test_grouped_apply_matches_per_layer() -> None:
+    """Fused kl-clip + scaled write (the broadcast-path tail) must match
+    the per-layer _compute_grad_scale + update_grad numerics."""
+    import os
+
+    from kfac_amd import KFACPreconditioner
+    from testing.models import LeNet
+
+    torch.manual_seed(99)
+    x0 = torch.randn(32, 1, 28, 28)
+    y0 = torch.randint(0, 10, (32,))
+    results = {}
+    for fused_apply in (True, False):
+        torch.manual_seed(99)
+        model = LeNet().cuda()
+        precon = KFACPreconditioner(
+            model, factor_update_steps=1, inv_update_steps=1, lr=0.01,
+        )
+        # force the partitioned (non-fully-fused) step shape
+        precon._fused_precondition_update = lambda: False
+        if not fused_apply:
+            os.environ['KFAC_AMD_NO_GROUPED_APPLY'] = '1'
+        try:
+            loss = torch.nn.functional.cross_entropy(
+                model(x0.cuda()), y0.cuda(),
+            )
+            loss.backward()
+            precon.step()
+        finally:
+            os.environ.pop('KFAC_AMD_NO_GROUPED_APPLY', None)
+        results[fused_apply] = {
+            n: p.grad.detach().cpu().clone()
+            for n, p in model.named_parameters()
+        }
+    for name in results[True]:
+        torch.testing.assert_close(
+            results[True][name],
+            results[False][name],
+            rtol=1e-4,
+            atol=1e-6,
+            msg=lambda m: f'{name}: {m}',
+        )
