@@ -69,6 +69,35 @@ def main() -> None:
         loss.backward()
         precon2.step()
         opt.step()
+    # INVERSE compute method + symmetry-aware triu wire format through a
+    # real multi-rank GPU step (round-1 verdict: these were only covered
+    # at the op level on GPU)
+    model3 = LeNet().cuda()
+    for p in model3.parameters():
+        dist.broadcast(p.data, src=0)
+    opt3 = torch.optim.SGD(model3.parameters(), lr=0.01)
+    precon3 = KFACPreconditioner(
+        model3,
+        factor_update_steps=1,
+        inv_update_steps=2,
+        lr=0.01,
+        grad_worker_fraction=DistributedStrategy.HYBRID_OPT,
+        compute_method='inverse',
+        symmetry_aware=True,
+        inv_update_async=False,
+    )
+    losses3 = []
+    for _ in range(8):
+        opt3.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model3(x), y)
+        loss.backward()
+        for p in model3.parameters():
+            dist.all_reduce(p.grad)
+            p.grad /= dist.get_world_size()
+        precon3.step()
+        opt3.step()
+        losses3.append(loss.item())
+    assert losses3[0] > losses3[-1], losses3
     if rank == 0:
         print('dist async smoke ok:', losses[0], '->', losses[-1])
     dist.destroy_process_group()
