@@ -433,7 +433,11 @@ torch::Tensor precond_apply_grouped(
     std::vector<torch::Tensor> qgs,
     std::vector<torch::Tensor> dgdas,
     double kl_clip,
-    double lr) {
+    double lr,
+    // optional PRE-ACCUMULATED kl-clip dot contributions from layers
+    // handled outside this call (e.g. the large layers routed through
+    // the hipBLASLt xf32 chain); the scale covers both sets.
+    c10::optional<torch::Tensor> accum_init) {
   const int L = (int)wgrads.size();
   TORCH_CHECK(L > 0, "empty layer list");
   TORCH_CHECK(
@@ -479,7 +483,15 @@ torch::Tensor precond_apply_grouped(
   auto s1 = torch::empty({total}, dev_opts);
   auto s2 = torch::empty({total}, dev_opts);
   auto outbuf = torch::zeros({total}, dev_opts);
-  auto work = torch::empty({2}, dev_opts);  // [accum, scale]
+  auto work = torch::zeros({2}, dev_opts);  // [accum, scale]
+  if (accum_init.has_value()) {
+    check_gpu_contig(*accum_init, "accum_init");
+    TORCH_CHECK(
+        accum_init->scalar_type() == torch::kFloat32 &&
+            accum_init->numel() == 1,
+        "accum_init must be a 1-element fp32 tensor");
+    work.narrow(0, 0, 1).copy_(accum_init->reshape({1}));
+  }
 
   auto desc_cpu = torch::empty(
       {L * (int64_t)(sizeof(PrecondDescHost) / 8)},
@@ -1036,7 +1048,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def(
       "precond_apply_grouped",
       &precond_apply_grouped,
-      "fused gather->precondition->kl-clip->scaled in-place grad update");
+      "fused gather->precondition->kl-clip->scaled in-place grad update",
+      pybind11::arg("wgrads"),
+      pybind11::arg("bgrads"),
+      pybind11::arg("qas"),
+      pybind11::arg("qgs"),
+      pybind11::arg("dgdas"),
+      pybind11::arg("kl_clip"),
+      pybind11::arg("lr"),
+      pybind11::arg("accum_init") = pybind11::none());
   m.def(
       "precond_eigen_grouped",
       &precond_eigen_grouped,

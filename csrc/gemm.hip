@@ -573,7 +573,8 @@ hipError_t precond_apply_grouped_f32(
   grouped_precond_kernel<2><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
   grouped_precond_kernel<3><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
   grouped_precond_kernel<4><<<total_tiles, 256, 0, stream>>>(desc, n_layers);
-  zero_kernel<<<1, 1, 0, stream>>>(accum);
+  // accum arrives zeroed (or PRE-accumulated with contributions from
+  // layers preconditioned outside this call) from the host binding.
   KFAC_HIP_CHECK(
       kl_clip_accum_t<float>(stream, accum, outbuf, gbuf, total_elems));
   grad_scale_kernel<<<1, 1, 0, stream>>>(scale, accum, kl_clip, lr);

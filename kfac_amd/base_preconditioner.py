@@ -365,6 +365,7 @@ class BaseKFACPreconditioner:
         qas = []
         qgs = []
         dgdas = []
+        big = []  # large layers: hipBLASLt xf32 chain beats the grouped launch
         for name, layer in reversed(list(self._layers.values())):
             if not self._assignment.is_grad_worker(name):
                 return False
@@ -382,28 +383,57 @@ class BaseKFACPreconditioner:
             wg = layer.module.get_weight_grad()
             if wg is None or wg.dtype != torch.float32 or not wg.is_contiguous():
                 return False
-            wgrads.append(wg.view(wg.size(0), -1))
+            wgv = wg.view(wg.size(0), -1)
             if layer.module.has_bias():
                 bg = layer.module.get_bias_grad()
                 if bg.dtype != torch.float32 or not bg.is_contiguous():
                     return False
-                bgrads.append(bg)
             else:
-                bgrads.append(wg.new_empty(0))
-            qas.append(qa)
-            qgs.append(qg)
-            dgdas.append(dgda)
-        if not wgrads:
+                bg = wg.new_empty(0)
+            m = qg.size(0)
+            n = qa.size(0)
+            if ops.chain_flops(m, n) > ops.CHAIN_FLOPS_XF32_THRESHOLD:
+                big.append((layer, wgv, bg, qa, qg, dgda))
+            else:
+                wgrads.append(wgv)
+                bgrads.append(bg)
+                qas.append(qa)
+                qgs.append(qg)
+                dgdas.append(dgda)
+        if not wgrads and not big:
             return True
-        ops.precond_apply_grouped(
-            wgrads,
-            bgrads,
-            qas,
-            qgs,
-            dgdas,
-            0.0 if kl_clip is None else float(kl_clip),
-            float(lr),
-        )
+        accum = None
+        big_outs = []
+        if big:
+            device = big[0][1].device
+            accum = torch.zeros((1,), dtype=torch.float32, device=device)
+            for layer, wgv, bg, qa, qg, dgda in big:
+                g = layer.module.get_grad()
+                out = ops.precond_eigen_xf32(g, qa, qg, dgda)
+                ops.kl_clip_accum(accum[0], out, g)
+                big_outs.append((layer, out))
+        if wgrads:
+            scale = ops.precond_apply_grouped(
+                wgrads,
+                bgrads,
+                qas,
+                qgs,
+                dgdas,
+                0.0 if kl_clip is None else float(kl_clip),
+                float(lr),
+                accum,
+            )
+        else:
+            assert accum is not None
+            if kl_clip is None:
+                scale = torch.ones(1, device=accum.device)
+            else:
+                scale = ops.grad_scale_from_accum(
+                    accum[0], float(kl_clip), lr,
+                ).reshape(1)
+        for layer, out in big_outs:
+            layer.grad = out
+            layer.update_grad(scale=scale[0])
         return True
 
     def _grouped_precondition(self) -> set[str]:
