@@ -405,13 +405,49 @@ class BaseKFACPreconditioner:
         accum = None
         big_outs = []
         if big:
+            from kfac_amd.ops import blocked
+
             device = big[0][1].device
             accum = torch.zeros((1,), dtype=torch.float32, device=device)
-            for layer, wgv, bg, qa, qg, dgda in big:
-                g = layer.module.get_grad()
-                out = ops.precond_eigen_xf32(g, qa, qg, dgda)
-                ops.kl_clip_accum(accum[0], out, g)
-                big_outs.append((layer, out))
+            # batch same-shape big layers through bmm chains; the
+            # stacked QA/QG/dgda only change at inverse phases, so they
+            # are cached keyed by the identity of the per-layer tensors
+            shape_groups: dict[tuple[int, int], list] = {}
+            for item in big:
+                _, _, _, qa, qg, _ = item
+                shape_groups.setdefault(
+                    (qg.size(0), qa.size(0)), [],
+                ).append(item)
+            cache = getattr(self, '_big_stack_cache', None)
+            if cache is None:
+                cache = {}
+                self._big_stack_cache = cache
+            for shape, items in shape_groups.items():
+                ids = tuple(id(it[3]) for it in items) + tuple(
+                    id(it[4]) for it in items
+                )
+                ent = cache.get(shape)
+                if ent is None or ent[0] != ids:
+                    qa_s = torch.stack([it[3] for it in items])
+                    qg_s = torch.stack([it[4] for it in items])
+                    dgda_s = torch.stack([it[5] for it in items])
+                    ent = (ids, qa_s, qg_s, dgda_s)
+                    cache[shape] = ent
+                _, qa_s, qg_s, dgda_s = ent
+                g_s = torch.stack(
+                    [
+                        it[0].module.get_grad().to(torch.float32)
+                        for it in items
+                    ],
+                )
+                with blocked.gemm_engine(True):
+                    v1 = (qg_s.transpose(-1, -2) @ g_s) @ qa_s
+                    v2 = v1 * dgda_s
+                    out_s = (qg_s @ v2) @ qa_s.transpose(-1, -2)
+                out_s = out_s.contiguous()
+                ops.kl_clip_accum(accum[0], out_s, g_s.contiguous())
+                for i, it in enumerate(items):
+                    big_outs.append((it[0], out_s[i]))
         if wgrads:
             scale = ops.precond_apply_grouped(
                 wgrads,

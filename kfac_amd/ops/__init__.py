@@ -162,11 +162,16 @@ def precond_eigen(
     if grad.is_cuda:
         ext = _require_ext('precond_eigen')
         if ext is not None:
+            # the device kernels and the xf32 library chain are fp32;
+            # non-fp32 inv_dtype state is cast here (no-op by default)
+            qa = qa.to(torch.float32)
+            qg = qg.to(torch.float32)
             big = (
                 chain_flops(grad.size(0), grad.size(1))
                 > CHAIN_FLOPS_XF32_THRESHOLD
             )
             if dgda is not None:
+                dgda = dgda.to(torch.float32)
                 if big:
                     return precond_eigen_xf32(grad, qa, qg, dgda)
                 return ext.precond_eigen_fused(
@@ -176,6 +181,8 @@ def precond_eigen(
                     dgda.contiguous(),
                 )
             assert da is not None and dg is not None
+            da = da.to(torch.float32)
+            dg = dg.to(torch.float32)
             if big:
                 from kfac_amd.ops import blocked
 
