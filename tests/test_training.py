@@ -157,3 +157,11 @@ def test_distributed_matches_single_process() -> None:
     with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
         run_distributed(1, _single_reference, f.name)
         run_distributed(2, _half_batch_distributed, f.name)
+
+
+def test_distributed_training_hybrid_world8() -> None:
+    # the driver's 8-GPU scaling run shape: world 8, HYBRID 0.5 ->
+    # 4 gradient workers per layer in a 4x2 grid, inverse AND gradient
+    # broadcasts active. Protocol-only on CPU gloo; RCCL first contact
+    # is the driver's run by design.
+    run_distributed(8, _dist_case, 'HYBRID_OPT')
