@@ -30,7 +30,20 @@ NB = 128
 
 @contextlib.contextmanager
 def gemm_engine(tf32: bool) -> Iterator[None]:
-    """Scope the hipBLASLt xf32 (bf16x3) path on or off."""
+    """Scope the hipBLASLt xf32 (bf16x3) path on or off.
+
+    ``allow_tf32`` is PROCESS-GLOBAL state: flipping it from the async
+    inverse worker thread while the main thread trains could silently
+    run a user's fp32 matmuls at reduced precision.  Off the main
+    thread this is therefore a no-op — the async phase's GEMMs run on
+    the exact-fp32 engine (2x slower, but that work is overlapped with
+    training steps by design).
+    """
+    import threading
+
+    if threading.current_thread() is not threading.main_thread():
+        yield
+        return
     prev = torch.backends.cuda.matmul.allow_tf32
     torch.backends.cuda.matmul.allow_tf32 = tf32
     try:
