@@ -165,3 +165,9 @@ def test_distributed_training_hybrid_world8() -> None:
     # broadcasts active. Protocol-only on CPU gloo; RCCL first contact
     # is the driver's run by design.
     run_distributed(8, _dist_case, 'HYBRID_OPT')
+
+
+def test_distributed_training_mem_opt_world8() -> None:
+    # BASELINE.json's grad_worker_fraction=1/8 config shape: every
+    # layer has ONE grad worker broadcasting to the other 7 ranks.
+    run_distributed(8, _dist_case, 'MEM_OPT')
