@@ -41,6 +41,7 @@ hipError_t syevj_small_f32(hipStream_t, const float*, float*, float*, int, int, 
 hipError_t triu_unpack_f32(hipStream_t, float*, const float*, int);
 hipError_t chol_diag_inv_f32(hipStream_t, float*, float*, int, int, int, int, float);
 hipError_t gather_cols_f32(hipStream_t, const float*, const long*, const long*, float*, int, int, int, int);
+hipError_t bucket_unpack_f32(hipStream_t, const float*, const void*, int, long);
 hipError_t scatter_cols_f32(hipStream_t, float*, const long*, const long*, const float*, int, int, int, int);
 
 }  // namespace kfac
@@ -530,6 +531,41 @@ torch::Tensor precond_apply_grouped(
   return work.narrow(0, 1, 1);
 }
 
+// Fused allreduce-bucket unpack (K13): one kernel scatters the flat
+// buffer into every member tensor (fp32 factors only; mixed dtypes
+// keep the per-tensor copy path).
+struct BucketDescHost {
+  float* dst;
+  long offset;
+  long numel;
+};
+static_assert(sizeof(BucketDescHost) == 24, "bucket desc layout");
+
+void bucket_unpack(torch::Tensor flat, std::vector<torch::Tensor> dsts) {
+  check_gpu_contig(flat, "flat");
+  TORCH_CHECK(flat.scalar_type() == torch::kFloat32, "fp32 only");
+  const int n = (int)dsts.size();
+  TORCH_CHECK(n > 0, "empty tensor list");
+  auto desc_cpu = torch::empty(
+      {n * 3},
+      torch::TensorOptions().dtype(torch::kInt64).pinned_memory(true));
+  auto* d = (BucketDescHost*)desc_cpu.data_ptr<int64_t>();
+  long off = 0;
+  for (int i = 0; i < n; ++i) {
+    check_gpu_contig(dsts[i], "dst");
+    TORCH_CHECK(dsts[i].scalar_type() == torch::kFloat32, "fp32 only");
+    d[i].dst = dsts[i].data_ptr<float>();
+    d[i].offset = off;
+    d[i].numel = dsts[i].numel();
+    off += d[i].numel;
+  }
+  TORCH_CHECK(off == flat.numel(), "bucket size mismatch");
+  auto desc_dev = desc_cpu.to(flat.device(), /*non_blocking=*/true);
+  CHECK_OK(kfac::bucket_unpack_f32(
+      current_stream(flat), flat.data_ptr<float>(),
+      desc_dev.data_ptr<int64_t>(), n, off));
+}
+
 // Batched column gather/scatter for the warm block-Jacobi rounds:
 // out[p] = t[mat[p]][:, idx[p]] without materializing index grids.
 torch::Tensor gather_cols(
@@ -1015,6 +1051,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("precond_eigen_fused", &precond_eigen_fused, "Kronecker precondition (prediv)");
   m.def("precond_eigen", &precond_eigen, "Kronecker precondition (dg/da)");
   m.def("precond_inverse", &precond_inverse, "G^-1 grad A^-1");
+  m.def(
+      "bucket_unpack",
+      &bucket_unpack,
+      "fused flat-bucket scatter into member tensors (K13)");
   m.def(
       "gather_cols",
       &gather_cols,

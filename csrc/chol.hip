@@ -195,4 +195,54 @@ hipError_t scatter_cols_f32(
   return hipGetLastError();
 }
 
+// ---- fused allreduce-bucket unpack (K13) ----
+//
+// The bucket pack is a single torch.cat launch; the unpack previously
+// issued one copy kernel per tensor (~30-50 launches per factor step).
+// One kernel scatters the whole flat buffer back through a descriptor
+// table of (dst pointer, offset, numel).
+
+struct BucketDesc {
+  float* dst;
+  long offset;
+  long numel;
+};
+
+__global__ void bucket_unpack_kernel(
+    const float* __restrict__ flat,
+    const BucketDesc* __restrict__ desc,
+    int n_tensors,
+    long total) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    // binary search the tensor containing flat element i
+    int lo = 0;
+    int hi = n_tensors - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (desc[mid].offset <= i) {
+        lo = mid;
+      } else {
+        hi = mid - 1;
+      }
+    }
+    desc[lo].dst[i - desc[lo].offset] = flat[i];
+  }
+}
+
+hipError_t bucket_unpack_f32(
+    hipStream_t stream,
+    const float* flat,
+    const void* desc,
+    int n_tensors,
+    long total) {
+  const int threads = 256;
+  const int blocks =
+      (int)min((total + threads - 1) / threads, (long)8192);
+  bucket_unpack_kernel<<<blocks, threads, 0, stream>>>(
+      flat, (const BucketDesc*)desc, n_tensors, total);
+  return hipGetLastError();
+}
+
 }  // namespace kfac

@@ -159,6 +159,18 @@ class AllreduceTensorBucket:
             raise RuntimeError('bucket waited before communicate()')
         if self._work is not None:
             self._work.wait()
+        if (
+            self._flat.is_cuda
+            and self._flat.dtype == torch.float32
+            and all(t.is_contiguous() for t in self._tensors)
+        ):
+            # fused scatter (K13): one kernel instead of one copy
+            # launch per member tensor
+            ext = ops._load_ext()
+            if ext is not None:
+                ext.bucket_unpack(self._flat, self._tensors)
+                self._unpacked = True
+                return
         offset = 0
         for t in self._tensors:
             n = t.numel()
