@@ -179,3 +179,49 @@ def test_warm_bail_falls_back_cleanly() -> None:
     dvals, q = BaseKFACPreconditioner._group_eigh(f1, layers, 'a')
     gates(f1, dvals, q)
     assert all(layer._warm_phases_a == 0 for layer in layers)
+
+
+def test_warm_training_trajectory_matches_dense() -> None:
+    """Multi-phase training with the warm solver must track the dense
+    (syevd) trajectory: same model/data/seeds, 12 steps with an inverse
+    phase every 2 steps, final parameters within the solver tolerance."""
+    import os
+
+    from kfac_amd import KFACPreconditioner
+
+    results = {}
+    for warm in (True, False):
+        torch.manual_seed(11)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(640, 640),
+            torch.nn.ReLU(),
+            torch.nn.Linear(640, 10),
+        ).cuda()
+        precon = KFACPreconditioner(
+            model,
+            factor_update_steps=1,
+            inv_update_steps=2,
+            lr=0.05,
+            inv_update_async=False,
+        )
+        opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        x = torch.randn(256, 640, device='cuda')
+        y = torch.randint(0, 10, (256,), device='cuda')
+        if not warm:
+            os.environ['KFAC_AMD_WARM_EIGH'] = '0'
+        try:
+            for _ in range(12):
+                opt.zero_grad(set_to_none=True)
+                loss = torch.nn.functional.cross_entropy(model(x), y)
+                loss.backward()
+                precon.step()
+                opt.step()
+        finally:
+            os.environ.pop('KFAC_AMD_WARM_EIGH', None)
+        results[warm] = {
+            n: p.detach().cpu().clone() for n, p in model.named_parameters()
+        }
+    for name in results[True]:
+        ref = results[False][name]
+        diff = (results[True][name] - ref).norm() / ref.norm().clamp_min(1e-12)
+        assert float(diff) < 2e-3, f'{name}: rel param drift {float(diff)}'
