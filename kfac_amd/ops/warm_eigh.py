@@ -12,18 +12,23 @@ dense block pairs.
 Algorithm (per same-size factor group, batched over the group):
   1. T = Q_prev^T F' Q_prev           — two batched xf32 GEMMs
   2. diagonal-block pass              — batched in-LDS Jacobi (b x b)
-  3. adaptive rounds: pick a maximal matching of block pairs whose
-     off-diagonal Frobenius mass matters, solve the 2b x 2b subproblems
-     with the in-LDS Jacobi kernel (+ one Newton orthogonality polish),
-     apply the rotations as batched GEMMs using T's symmetry
-     (T'[E, :] = V^T T[E, :] on full rows; T[:, E] mirrored; T'[E, E]
-     = diag from the subproblem), and update Q's columns.
-  4. stop when sum of off-block mass <= tol * ||F'||_F per matrix.
+  3. dense band correction: clustered mixing (a contiguous range of
+     high-degree blocks) is solved by ONE batched dense sub-eigh and
+     applied with GEMMs — pairwise rotations would need >= width/b
+     sequential rounds for a clique.
+  4. adaptive rounds: maximal matching of block pairs whose
+     off-diagonal Frobenius mass matters; 2b x 2b subproblems on the
+     in-LDS Jacobi kernel (+ one Newton orthogonality polish); the
+     classic parallel round applies T <- V^T T on pair rows (batched
+     globally) then T <- T V on pair columns via payload-only
+     gather/scatter kernels, then updates Q's columns.
+  5. stop when the off-block mass <= tol * ||F'||_F, per matrix.
 
-Safety: if the warm start is bad (off0 above ``bail_rel`` — e.g. after
-a big learning-rate event) the caller falls back to rocSOLVER syevd,
-which also serves the cold first phase.  Replaces the reference's
-``torch.linalg.eigh`` (kfac/layers/eigen.py:309-344) on the warm path.
+Safety is PER MATRIX: bad warm starts (off0 above ``bail_rel``),
+stalled progress, and broad scattered drift at small n are reported in
+the returned mask and the caller dense-solves just those matrices.
+Replaces the reference's ``torch.linalg.eigh``
+(kfac/layers/eigen.py:309-344) on the warm path.
 """
 
 from __future__ import annotations
@@ -272,7 +277,6 @@ def warm_eigh_batched(
     tol_sq = (tol * tn) ** 2
     arange_b = torch.arange(b, device=dev)
     entry_offsq: torch.Tensor | None = None
-    offsq = tol_sq  # placeholder; overwritten in the loop
     for rnd in range(max_rounds):
         bn = _block_off_norms(t, b)
         offsq = (bn * bn).sum(dim=(-2, -1))
