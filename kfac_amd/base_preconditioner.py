@@ -183,11 +183,20 @@ class BaseKFACPreconditioner:
 
     # -- checkpointing ------------------------------------------------------
 
-    def state_dict(self, include_factors: bool = True) -> dict[str, Any]:
+    def state_dict(
+        self,
+        include_factors: bool = True,
+        include_second_order: bool = False,
+    ) -> dict[str, Any]:
         """K-FAC state: steps, non-callable hyperparams, per-layer factors.
 
         Format matches the reference (base_preconditioner.py:215-247) so
-        checkpoints are interchangeable.
+        checkpoints are interchangeable.  ``include_second_order``
+        additionally embeds the eigendecompositions/inverses so
+        ``load_state_dict`` can skip the inverse recomputation phase
+        (the warm solver also keeps its basis continuity across the
+        restart) — an extension over the reference, which always
+        recomputes.
         """
         state_dict: dict[str, Any] = {'steps': self.steps}
         for key, value in (
@@ -202,7 +211,9 @@ class BaseKFACPreconditioner:
                 state_dict[key] = value
         if include_factors:
             state_dict['layers'] = {
-                name: layer.state_dict()
+                name: layer.state_dict(
+                    include_second_order=include_second_order,
+                )
                 for name, layer in self._layers.values()
             }
         return state_dict
@@ -233,9 +244,16 @@ class BaseKFACPreconditioner:
                     'loaded state dict contains a different number of layers',
                 )
             by_name = {name: layer for name, layer in self._layers.values()}
+            restored_so: set[str] = set()
             for found_name, layer_state in state_dict['layers'].items():
                 if found_name in by_name:
                     by_name[found_name].load_state_dict(layer_state)
+                    if any(
+                        k.startswith('so_')
+                        and isinstance(v, torch.Tensor)
+                        for k, v in layer_state.items()
+                    ):
+                        restored_so.add(found_name)
         elif compute_inverses:
             warnings.warn(
                 'Layer factors are not included in the state_dict so '
@@ -245,6 +263,13 @@ class BaseKFACPreconditioner:
             compute_inverses = False
         if compute_inverses:
             for name, layer in self._layers.values():
+                if name in restored_so:
+                    # the checkpoint embedded this layer's second-order
+                    # state (include_second_order=True at save): every
+                    # rank restored identical eigendecompositions /
+                    # inverses, so neither recomputation nor broadcast
+                    # is needed.
+                    continue
                 layer.compute_a_inv(damping=self.damping)
                 layer.compute_g_inv(damping=self.damping)
                 if self._assignment.broadcast_inverses():

@@ -136,12 +136,32 @@ class KFACBaseLayer:
 
     # -- serialization ----------------------------------------------------
 
-    def state_dict(self) -> dict[str, torch.Tensor | None]:
-        """Factors only; inverses are recomputable (reference base.py:130-142)."""
-        return {'A': self.a_factor, 'G': self.g_factor}
+    # second-order attributes included in a state_dict when the caller
+    # opts in (include_second_order=True); subclasses list theirs
+    SECOND_ORDER_KEYS: tuple[str, ...] = ()
+
+    def state_dict(
+        self, include_second_order: bool = False,
+    ) -> dict[str, torch.Tensor | None]:
+        """Factors only by default — the reference-compatible format
+        (inverses recomputable, reference base.py:130-142).  With
+        ``include_second_order`` the eigendecompositions/inverses are
+        embedded so a resume can skip the recomputation phase entirely
+        (and keeps the warm-solver basis continuity across restarts).
+        """
+        sd: dict[str, torch.Tensor | None] = {
+            'A': self.a_factor, 'G': self.g_factor,
+        }
+        if include_second_order:
+            for key in self.SECOND_ORDER_KEYS:
+                value = getattr(self, key)
+                if isinstance(value, torch.Tensor):
+                    sd[f'so_{key}'] = value
+        return sd
 
     def load_state_dict(self, state_dict: dict[str, torch.Tensor | None]) -> None:
-        """Load A/G factors, moving them to the module's device."""
+        """Load A/G factors (and any embedded second-order state),
+        moving them to the module's device."""
         if 'A' not in state_dict or 'G' not in state_dict:
             raise KeyError(
                 "KFACBaseLayer state_dict must contain keys 'A' and 'G'",
@@ -151,6 +171,18 @@ class KFACBaseLayer:
             self.a_factor = state_dict['A'].to(device)
         if state_dict['G'] is not None:
             self.g_factor = state_dict['G'].to(device)
+        for key in self.SECOND_ORDER_KEYS:
+            value = state_dict.get(f'so_{key}')
+            if isinstance(value, torch.Tensor):
+                setattr(self, key, value.to(device))
+
+    def has_second_order_state(self) -> bool:
+        """True if this layer's preconditioning state is ready (either
+        freshly computed or restored from a checkpoint)."""
+        return any(
+            isinstance(getattr(self, key), torch.Tensor)
+            for key in self.SECOND_ORDER_KEYS
+        )
 
     def memory_usage(self) -> dict[str, int]:
         """Bytes used by per-layer state (reference base.py:167-184)."""

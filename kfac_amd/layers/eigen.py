@@ -24,6 +24,8 @@ from kfac_amd.layers.base import KFACBaseLayer
 class KFACEigenLayer(KFACBaseLayer):
     """K-FAC layer preconditioning via eigendecomposition of A and G."""
 
+    SECOND_ORDER_KEYS = ('qa', 'qg', 'da', 'dg', 'dgda')
+
     def __init__(self, *args: Any, prediv_eigenvalues: bool = False, **kwargs: Any) -> None:
         """Init KFACEigenLayer.
 

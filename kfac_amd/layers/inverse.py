@@ -21,6 +21,8 @@ from kfac_amd.layers.base import KFACBaseLayer
 class KFACInverseLayer(KFACBaseLayer):
     """K-FAC layer preconditioning via explicit damped inverses."""
 
+    SECOND_ORDER_KEYS = ('a_inv', 'g_inv')
+
     def __init__(self, *args: Any, **kwargs: Any) -> None:
         super().__init__(*args, **kwargs)
         self._a_inv: Any = None

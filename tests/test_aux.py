@@ -515,3 +515,58 @@ def test_bucket_key_includes_average_flag() -> None:
     k1 = (tdc._group_key(None)) + (a.dtype, a.device, True)
     k2 = (tdc._group_key(None)) + (a.dtype, a.device, False)
     assert k1 != k2
+
+
+def test_state_dict_second_order_resume() -> None:
+    """include_second_order embeds eigendecompositions so a resume can
+    skip the inverse recomputation (extension over the reference)."""
+    torch.manual_seed(4)
+    model = TinyModel()
+    p = KFACPreconditioner(
+        model, factor_update_steps=1, inv_update_steps=1, lr=0.1,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    x = torch.randn(16, 10)
+    y = torch.randint(0, 3, (16,))
+    for _ in range(2):
+        opt.zero_grad()
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        p.step()
+        opt.step()
+    sd = p.state_dict(include_second_order=True)
+    layers_sd = sd['layers']
+    assert any(
+        k.startswith('so_') for v in layers_sd.values() for k in v
+    )
+    # plain format unchanged when not requested
+    sd_plain = p.state_dict()
+    assert all(
+        set(v) == {'A', 'G'} for v in sd_plain['layers'].values()
+    )
+
+    model2 = TinyModel()
+    p2 = KFACPreconditioner(
+        model2, factor_update_steps=1, inv_update_steps=1, lr=0.1,
+    )
+    calls = {'n': 0}
+    orig = type(list(p2._layers.values())[0][1]).compute_a_inv
+
+    def counting(self, *a, **k):
+        calls['n'] += 1
+        return orig(self, *a, **k)
+
+    import unittest.mock as mock
+
+    with mock.patch.object(
+        type(list(p2._layers.values())[0][1]), 'compute_a_inv', counting,
+    ):
+        p2.load_state_dict(sd, compute_inverses=True)
+    assert calls['n'] == 0, 'recomputation should be skipped'
+    for _, layer in p2._layers.values():
+        assert layer.has_second_order_state()
+    # training continues from the restored state
+    opt2 = torch.optim.SGD(model2.parameters(), lr=0.1)
+    opt2.zero_grad()
+    torch.nn.functional.cross_entropy(model2(x), y).backward()
+    p2.step()
+    opt2.step()
