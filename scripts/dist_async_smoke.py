@@ -98,6 +98,55 @@ def main() -> None:
         opt3.step()
         losses3.append(loss.item())
     assert losses3[0] > losses3[-1], losses3
+    # GPT-NeoX TP layer protocol on CUDA tensors: gather sharded
+    # activations -> precondition on primary -> true dist.scatter back
+    # (the reference emulates scatter with a zero-padded reduce_scatter)
+    from kfac_amd.distributed import TorchDistributedCommunicator
+    from kfac_amd.gpt_neox.layer import GPTNeoXKFACEigenLayer
+    from kfac_amd.gpt_neox.modules import GPTNeoXLinearModuleHelper
+    from testing.gpt_neox import RowParallelLinear
+
+    mp_group = dist.new_group([0, 1])
+    dp_groups = [dist.new_group([r]) for r in range(2)]
+    torch.manual_seed(5)
+    in_dim, out_dim, batch = 128, 48, 64
+    x = torch.randn(batch, in_dim, device='cuda')
+    full_grad = torch.randn(out_dim, in_dim, device='cuda')
+    bias_grad = torch.randn(out_dim, device='cuda')
+    shard = in_dim // 2
+    module = RowParallelLinear(shard, out_dim, bias=True).cuda()
+    module.weight.grad = (
+        full_grad[:, rank * shard : (rank + 1) * shard].clone()
+    )
+    module.bias.grad = bias_grad.clone()
+    tdc4 = TorchDistributedCommunicator()
+    tp_layer = GPTNeoXKFACEigenLayer(
+        GPTNeoXLinearModuleHelper(module, mp_group, parallelism='input'),
+        parallelism='input',
+        model_parallel_group=mp_group,
+        data_parallel_group=dp_groups[rank],
+        pipe_parallel_peer_group=mp_group,
+        primary_rank=0,
+        tdc=tdc4,
+        prediv_eigenvalues=False,
+    )
+    tp_layer.save_layer_input([x[:, rank * shard : (rank + 1) * shard]])
+    tp_layer.save_layer_grad_output(
+        (torch.randn(batch, out_dim, device='cuda'),),
+    )
+    tp_layer.update_a_factor(0.95)
+    tp_layer.update_g_factor(0.95)
+    tp_layer.reduce_a_factor()
+    tp_layer.reduce_g_factor()
+    tdc4.flush_allreduce_buckets()
+    if rank == 0:
+        tp_layer.compute_a_inv(damping=1e-3)
+        tp_layer.compute_g_inv(damping=1e-3)
+    tp_layer.preconditioned_grad(damping=1e-3)
+    g = tp_layer.grad
+    assert g is not None and g.shape == (out_dim, shard + 1), g.shape
+    assert bool(torch.isfinite(g).all())
+
     if rank == 0:
         print('dist async smoke ok:', losses[0], '->', losses[-1])
     dist.destroy_process_group()
