@@ -250,22 +250,36 @@ __device__ __forceinline__ void gemm_tile_body(
   f32x4 acc[4][4] = {};
 
   if constexpr (SPLIT) {
-    // Register-staged pipeline: loads for slice t+1 issue while slice
-    // t's MFMAs run; LDS writes happen between the two barriers.
+    // Register-staged pipeline with DOUBLE-BUFFERED LDS: slice t+1's
+    // staging writes land in the other slab while slice t's MFMAs read
+    // — ONE barrier per slice instead of two (the stage kernels
+    // measured issue/barrier-bound at ~15% MFMA-pipe occupancy,
+    // profiles/pmc_r2.txt).  Slabs are lds_a + p*GBK rows.
     StageRegs ra, rb;
     stage_load(a, M, K, lda, !ta, 0, i0, tid, ra);
     stage_load(b, N, K, ldb, tb, 0, j0, tid, rb);
+    stage_store(ra, !ta, tid, lds_a);
+    stage_store(rb, tb, tid, lds_b);
+    if (GBK < K) {
+      stage_load(a, M, K, lda, !ta, GBK, i0, tid, ra);
+      stage_load(b, N, K, ldb, tb, GBK, j0, tid, rb);
+    }
+    __syncthreads();  // slab 0 ready
+    int p = 0;
     for (int k0 = 0; k0 < K; k0 += GBK) {
-      __syncthreads();  // prior slice's LDS reads complete
-      stage_store(ra, !ta, tid, lds_a);
-      stage_store(rb, tb, tid, lds_b);
+      // stage slice t+1 into the OTHER slab (no readers there)
       if (k0 + GBK < K) {
-        stage_load(a, M, K, lda, !ta, k0 + GBK, i0, tid, ra);
-        stage_load(b, N, K, ldb, tb, k0 + GBK, j0, tid, rb);
+        stage_store(ra, !ta, tid, lds_a + (p ^ 1) * GBK);
+        stage_store(rb, tb, tid, lds_b + (p ^ 1) * GBK);
+        if (k0 + 2 * GBK < K) {
+          stage_load(a, M, K, lda, !ta, k0 + 2 * GBK, i0, tid, ra);
+          stage_load(b, N, K, ldb, tb, k0 + 2 * GBK, j0, tid, rb);
+        }
       }
-      __syncthreads();  // this slice's LDS image ready
       // One 16x16x32 bf16 MFMA triple covers the whole 32-deep slice:
       // lane l supplies elements k = 8*(l>>4)..+7 of column (l&15).
+      float(*cur_a)[GLDS] = lds_a + p * GBK;
+      float(*cur_b)[GLDS] = lds_b + p * GBK;
       const int kbase = (lane >> 4) * 8;
       gemm_bf16x8 ah[4], al[4], bh[4], bl[4];
 #pragma unroll
@@ -275,8 +289,8 @@ __device__ __forceinline__ void gemm_tile_body(
         float va[8], vb[8];
 #pragma unroll
         for (int q = 0; q < 8; ++q) {
-          va[q] = lds_a[kbase + q][lds_swz(kbase + q, ca)];
-          vb[q] = lds_b[kbase + q][lds_swz(kbase + q, cb)];
+          va[q] = cur_a[kbase + q][lds_swz(kbase + q, ca)];
+          vb[q] = cur_b[kbase + q][lds_swz(kbase + q, cb)];
         }
         split_bf16(va, ah[f], al[f]);
         split_bf16(vb, bh[f], bl[f]);
@@ -294,6 +308,8 @@ __device__ __forceinline__ void gemm_tile_body(
               ah[fi], bh[fj], acc[fi][fj], 0, 0, 0);
         }
       }
+      __syncthreads();  // this slice's reads AND next slab's writes done
+      p ^= 1;
     }
   } else {
     for (int k0 = 0; k0 < K; k0 += GBK) {
@@ -358,8 +374,10 @@ __global__ __launch_bounds__(256) void gemm_kernel(
     const float* __restrict__ e1,
     const float* __restrict__ e2,
     float damping) {
-  __shared__ float lds_a[GBK][GLDS];
-  __shared__ float lds_b[GBK][GLDS];
+  // 2x GBK rows: the SPLIT path ping-pongs two slabs (single-barrier
+  // pipeline); the f32 path uses only the first GBK rows.
+  __shared__ float lds_a[2 * GBK][GLDS];
+  __shared__ float lds_b[2 * GBK][GLDS];
   gemm_tile_body<EPI, SPLIT>(
       c, a, b, M, N, K, ta, tb, ta ? (long)M : (long)K,
       tb ? (long)K : (long)N, (long)N, e1, e2, damping,
@@ -409,8 +427,8 @@ __global__ __launch_bounds__(256) void grouped_precond_kernel(
   const int i0 = (local / ntj) * GBT;
   const int j0 = (local % ntj) * GBT;
 
-  __shared__ float lds_a[GBK][GLDS];
-  __shared__ float lds_b[GBK][GLDS];
+  __shared__ float lds_a[2 * GBK][GLDS];
+  __shared__ float lds_b[2 * GBK][GLDS];
 
   const long sp = d.spad;
   if constexpr (STAGE == 1) {
