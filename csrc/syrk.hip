@@ -611,22 +611,34 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16_big(
   const int m_end = min(accl.M, m_begin + m_per_split);
   if (m_begin >= m_end) return;
 
-  __shared__ __bf16 lds_l[BTB][BSTR];
-  __shared__ __bf16 lds_r[BTB][BSTR];
+  // Double-buffered slabs: slice t+1 stages into slab p^1 while slice
+  // t's MFMAs read slab p — one barrier per m-slice.  The kernel is
+  // HBM-streaming-bound (~55% of peak single-buffered); overlapping
+  // the staging with the MFMAs recovers most of the latency.
+  __shared__ __bf16 lds_l[2][BTB][BSTR];
+  __shared__ __bf16 lds_r[2][BTB][BSTR];
   const bool diag = same_tile_ok && (ti == tj);
 
   f32x4 acc[4][4] = {};
 
-  for (int m0 = m_begin; m0 < m_end; m0 += BKB) {
+  auto stage_into = [&](int slab, int m0) {
     if constexpr (AccL::kLaneAlongCols) {
-      stage_tile_bf16_big_colfast(accl, lds_l, m0, i0, tid);
-      if (!diag) stage_tile_bf16_big_colfast(accr, lds_r, m0, j0, tid);
+      stage_tile_bf16_big_colfast(accl, lds_l[slab], m0, i0, tid);
+      if (!diag) stage_tile_bf16_big_colfast(accr, lds_r[slab], m0, j0, tid);
     } else {
-      stage_tile_bf16_big(accl, lds_l, m0, i0, tid);
-      if (!diag) stage_tile_bf16_big(accr, lds_r, m0, j0, tid);
+      stage_tile_bf16_big(accl, lds_l[slab], m0, i0, tid);
+      if (!diag) stage_tile_bf16_big(accr, lds_r[slab], m0, j0, tid);
     }
-    __syncthreads();
-    auto rbuf = diag ? lds_l : lds_r;
+  };
+
+  stage_into(0, m_begin);
+  __syncthreads();
+  int p = 0;
+  for (int m0 = m_begin; m0 < m_end; m0 += BKB) {
+    if (m0 + BKB < m_end) {
+      stage_into(p ^ 1, m0 + BKB);
+    }
+    auto rbuf = diag ? lds_l[p] : lds_r[p];
 #pragma unroll
     for (int kt = 0; kt < BKB; kt += 32) {
       const int kfrag = kt + (lane >> 4) * 8;
@@ -636,7 +648,7 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16_big(
       for (int f = 0; f < 4; ++f) {
         const int ra = wr * 64 + f * 16 + (lane & 15);
         const int rb = wc * 64 + f * 16 + (lane & 15);
-        av[f] = *(const bf16x8*)&lds_l[ra][kswz(ra, kfrag)];
+        av[f] = *(const bf16x8*)&lds_l[p][ra][kswz(ra, kfrag)];
         bv[f] = *(const bf16x8*)&rbuf[rb][kswz(rb, kfrag)];
       }
 #pragma unroll
@@ -649,6 +661,7 @@ __global__ __launch_bounds__(256) void syrk_kernel_bf16_big(
       }
     }
     __syncthreads();
+    p ^= 1;
   }
 
   const bool mirror = (ti != tj);
