@@ -622,15 +622,15 @@ class BaseKFACPreconditioner:
 
     # -- async inverse pipeline --------------------------------------------
     #
-    # rocSOLVER's eigendecomposition is host-launch-bound (~48k small
-    # kernels per phase), so even on a side stream the *host* is busy for
-    # the whole phase. The async path runs the batched eigendecomposition
-    # in a worker THREAD on its own HIP stream, overlapping the next
-    # ``inv_async_delay`` training steps; all ranks swap in the new
-    # second-order state (and issue the inverse broadcasts) at exactly
-    # boundary+delay so collectives stay matched. Preconditioning between
-    # boundary and swap uses the previous inverses — well within K-FAC's
-    # by-design staleness (inverses are already inv_update_steps old).
+    # The inverse phase (warm block-Jacobi, or the host-launch-bound
+    # rocSOLVER fallback with ~48k small kernels) keeps the host busy,
+    # so the async path runs it in a worker THREAD on its own HIP
+    # stream, overlapping the next ``inv_async_delay`` training steps;
+    # all ranks swap in the new second-order state (and issue the
+    # inverse broadcasts) at exactly boundary+delay so collectives stay
+    # matched. Preconditioning between boundary and swap uses the
+    # previous inverses — well within K-FAC's by-design staleness
+    # (inverses are already inv_update_steps old).
 
     def _broadcast_inverses(self) -> None:
         if self._assignment.broadcast_inverses():
