@@ -1,0 +1,85 @@
+"""Multi-phase soak: many warm-solver cycles, memory and loss sanity.
+
+500 steps with an inverse phase every 50 (10 warm/cold cycles through
+the solver, cooldowns, re-anchors and the async pipeline), fresh
+batches. Fails loudly on NaN loss or unbounded memory growth.
+
+Run: python scripts/soak_warm.py [--model gptneox125m]
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument('--model', default='resnet50')
+    ap.add_argument('--steps', type=int, default=500)
+    args = ap.parse_args()
+
+    from kfac_amd import KFACPreconditioner
+    from kfac_amd.models import gptneox_125m, resnet50
+    from kfac_amd.models.gptneox import KFAC_SKIP_LAYERS
+
+    torch.manual_seed(0)
+    is_lm = args.model == 'gptneox125m'
+    model = (gptneox_125m() if is_lm else resnet50()).cuda()
+    precon = KFACPreconditioner(
+        model,
+        factor_update_steps=10,
+        inv_update_steps=50,
+        lr=0.1,
+        inv_update_async=True,
+        skip_layers=KFAC_SKIP_LAYERS if is_lm else [],
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    crit = torch.nn.CrossEntropyLoss()
+    gen = torch.Generator(device='cuda').manual_seed(3)
+    vocab = 50304
+    mem0 = None
+    t0 = time.time()
+    for step in range(args.steps):
+        opt.zero_grad(set_to_none=True)
+        if is_lm:
+            x = torch.randint(0, vocab, (8, 2048), device='cuda', generator=gen)
+            y = torch.randint(0, vocab, (8 * 2048,), device='cuda', generator=gen)
+        else:
+            x = torch.randn(64, 3, 224, 224, device='cuda', generator=gen)
+            y = torch.randint(0, 1000, (64,), device='cuda', generator=gen)
+        with torch.autocast('cuda', dtype=torch.bfloat16):
+            out = model(x)
+            loss = crit(out.view(-1, out.size(-1)) if is_lm else out, y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        if step % 100 == 99:
+            torch.cuda.synchronize()
+            lv = float(loss)
+            mem = torch.cuda.memory_allocated() / 2**30
+            if mem0 is None:
+                mem0 = mem
+            print(
+                f'step {step + 1}: loss={lv:.4f} mem={mem:.2f} GiB '
+                f'({(time.time() - t0):.1f}s)',
+                flush=True,
+            )
+            assert lv == lv, 'NaN loss'
+            assert mem < mem0 * 1.5 + 2.0, f'memory growth: {mem0} -> {mem}'
+    # warm counters actually advanced over the soak
+    warm_counts = [
+        getattr(layer, '_warm_phases_a', 0)
+        for _, (_, layer) in precon._layers.items()
+    ]
+    print(f'warm phase counters: max={max(warm_counts)}, soak OK')
+
+
+if __name__ == '__main__':
+    main()
