@@ -202,7 +202,7 @@ def warm_eigh_batched(
     tol: float = 1e-4,
     bail_rel: float = 0.25,
     max_rounds: int = 40,
-) -> tuple[torch.Tensor, torch.Tensor, bool]:
+) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Batched warm-started eigendecomposition.
 
     Args:
