@@ -996,7 +996,13 @@ class BaseKFACPreconditioner:
                         layer, attr_cnt, getattr(layer, attr_cnt, 0) + 1,
                     )
                 else:
-                    setattr(layer, attr_cd, 3)
+                    # skip ONE phase then retry: a failed attempt costs
+                    # ~30-60 ms (bail or early progress-exit) while a
+                    # skipped-but-would-succeed phase wastes 100+ ms of
+                    # dense solve — at inv_update_steps=100 a longer
+                    # cooldown blanks warm for hundreds of steps while
+                    # the drift regime is changing.
+                    setattr(layer, attr_cd, 1)
                     setattr(layer, attr_cnt, 0)
             return d, q
         d, q = _ops.eigh_batched(stack)
