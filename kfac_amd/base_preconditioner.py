@@ -790,6 +790,16 @@ class BaseKFACPreconditioner:
         import os
         import time as _time
 
+        # Join any in-flight async inverse job FIRST: a synchronous
+        # phase (bench's forced measurement, checkpoint resume, a
+        # direct caller) would otherwise race the worker thread — the
+        # worker's kernels read layer.qa on its own stream while this
+        # thread replaces the attribute, freeing memory still being
+        # read (observed as a GPU memory fault when a phase launched
+        # near the end of a run was still computing).
+        if self._async_job is not None:
+            self._finish_async_inverses()
+
         trace = os.environ.get('KFAC_AMD_PHASE_TRACE', '0') == '1'
 
         def _mark(label: str, t0: float) -> float:
