@@ -677,7 +677,24 @@ class BaseKFACPreconditioner:
                 continue
             a = layer.a_factor.detach().clone() if a_mine else None
             g = layer.g_factor.detach().clone() if g_mine else None
-            work.append((layer, a, g))
+            # snapshot the previous eigenbases too: the worker must
+            # never read LIVE layer attributes — the main thread may
+            # replace (and free) them while worker-stream kernels are
+            # still reading (the same hazard class as the factor
+            # clones above).
+            qa_prev = layer.qa
+            qg_prev = layer.qg
+            qa_c = (
+                qa_prev.detach().clone()
+                if isinstance(qa_prev, torch.Tensor) and a_mine
+                else None
+            )
+            qg_c = (
+                qg_prev.detach().clone()
+                if isinstance(qg_prev, torch.Tensor) and g_mine
+                else None
+            )
+            work.append((layer, a, g, qa_c, qg_c))
 
         device = list(self._layers.values())[0][1].module.device
         ready = torch.cuda.Event()
@@ -698,25 +715,32 @@ class BaseKFACPreconditioner:
                     stream.wait_event(ready)
                     results: dict[Any, dict[str, torch.Tensor | None]] = {}
                     for which in ('a', 'g'):
-                        groups: dict[int, list[tuple[Any, torch.Tensor]]] = dd(list)
-                        for layer, a, g in work:
+                        groups: dict[
+                            int,
+                            list[tuple[Any, torch.Tensor, Any]],
+                        ] = dd(list)
+                        for layer, a, g, qa_c, qg_c in work:
                             fac = a if which == 'a' else g
+                            prev = qa_c if which == 'a' else qg_c
                             if fac is not None:
-                                groups[fac.shape[0]].append((layer, fac))
+                                groups[fac.shape[0]].append(
+                                    (layer, fac, prev),
+                                )
                         for n, items in groups.items():
                             stack = torch.stack(
-                                [f.to(torch.float32) for _, f in items],
+                                [f.to(torch.float32) for _, f, _ in items],
                             )
                             if stack.is_cuda:
                                 d, q = BaseKFACPreconditioner._group_eigh(
                                     stack,
-                                    [lyr for lyr, _ in items],
+                                    [lyr for lyr, _, _ in items],
                                     which,
+                                    prev_override=[p for _, _, p in items],
                                 )
                             else:
                                 d, q = torch.linalg.eigh(stack)
                             d = torch.clamp(d, min=0.0)
-                            for i, (layer, _) in enumerate(items):
+                            for i, (layer, _, _) in enumerate(items):
                                 res = results.setdefault(layer, {})
                                 res[f'q{which}'] = (
                                     q[i].to(layer.inv_dtype).contiguous()
@@ -755,7 +779,13 @@ class BaseKFACPreconditioner:
     def _finish_async_inverses(self) -> None:
         job = self._async_job
         assert job is not None
-        job['thread'].join()
+        job['thread'].join(timeout=600)
+        if job['thread'].is_alive():
+            raise RuntimeError(
+                'async inverse worker did not finish within 600 s — the '
+                'GPU context is likely wedged (check for a preceding '
+                'memory fault)',
+            )
         self._async_job = None
         if job['error'] is not None:
             raise job['error']
@@ -915,6 +945,7 @@ class BaseKFACPreconditioner:
         stack: torch.Tensor,
         group: list[Any],
         which: str,
+        prev_override: list[Any] | None = None,
     ) -> tuple[torch.Tensor, torch.Tensor]:
         """Eigendecompose one same-size factor group.
 
@@ -952,7 +983,11 @@ class BaseKFACPreconditioner:
                     n, dtype=stack.dtype, device=stack.device,
                 ).expand_as(stack).contiguous()
                 return diag.clone(), eye
-        prev = [getattr(layer, attr_q, None) for layer in group]
+        prev = (
+            prev_override
+            if prev_override is not None
+            else [getattr(layer, attr_q, None) for layer in group]
+        )
         cooldown = max(
             (getattr(layer, attr_cd, 0) for layer in group), default=0,
         )
