@@ -266,3 +266,67 @@ def test_grouped_apply_matches_per_layer() -> None:
             atol=1e-6,
             msg=lambda m: f'{name}: {m}',
         )
+
+
+@pytest.mark.timeout(300)
+def test_forced_phase_joins_inflight_async_job() -> None:
+    """Regression: a synchronous inverse phase while an async job is in
+    flight must join the worker FIRST (base_preconditioner.py join-guard).
+
+    Before the guard, the forced phase (bench's all-in measurement,
+    checkpoint resume) replaced and freed layer.qa/.qg while the worker
+    thread's kernels were still reading them — observed as a GPU memory
+    access fault on a 300-step GPT-NeoX run. The worker now also
+    snapshots the previous eigenbases at launch (prev_override) so it
+    never reads live layer attributes. Widths >= 512 make the worker
+    take the warm prev_override path from phase 2 on.
+    """
+    from kfac_amd import KFACPreconditioner
+
+    torch.manual_seed(7)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(512, 512),
+        torch.nn.ReLU(),
+        torch.nn.Linear(512, 512),
+        torch.nn.ReLU(),
+        torch.nn.Linear(512, 10),
+    ).cuda()
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    precon = KFACPreconditioner(
+        model,
+        factor_update_steps=1,
+        inv_update_steps=6,
+        lr=0.01,
+        inv_update_async=True,
+        inv_async_delay=4,
+    )
+    x = torch.randn(32, 512, device='cuda')
+    y = torch.randint(0, 10, (32,), device='cuda')
+    forced = 0
+    for i in range(1, 26):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        # steps that are inverse-phase boundaries just LAUNCHED an async
+        # job (due 4 steps later): force a synchronous phase while it is
+        # in flight, exactly like bench.py's all-in measurement.
+        if precon._async_job is not None and precon.steps % 6 == 0:
+            precon._compute_local_inverses()
+            precon._broadcast_inverses()
+            assert precon._async_job is None  # guard joined it
+            forced += 1
+    assert forced >= 3  # the race window was actually exercised
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+    # the forced phase left coherent eigen state behind
+    for _, layer in precon._layers.values():
+        qa = layer.qa.to(torch.float32)
+        n = qa.shape[0]
+        eye = torch.eye(n, device=qa.device)
+        assert torch.allclose(qa.T @ qa, eye, atol=5e-3), (
+            (qa.T @ qa - eye).abs().max()
+        )
+    if precon._async_job is not None:
+        precon._finish_async_inverses()

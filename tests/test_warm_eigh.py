@@ -120,3 +120,42 @@ def test_warm_multiple_pairs_same_matrix_cross_blocks() -> None:
         atol=1e-4,
         rtol=1e-4,
     )
+
+
+def test_group_eigh_prev_override_plumbing() -> None:
+    """The async worker passes snapshot clones via prev_override instead
+    of letting _group_eigh read live layer attributes (race fix); the
+    cold CPU path must accept both forms and agree with torch eigh."""
+    from kfac_amd.base_preconditioner import BaseKFACPreconditioner
+
+    class _FakeLayer:
+        qa = None
+        qg = None
+
+    torch.manual_seed(0)
+    m = torch.randn(3, 32, 32)
+    stack = (m @ m.transpose(-1, -2) + 32 * torch.eye(32)).contiguous()
+    layers = [_FakeLayer() for _ in range(3)]
+    d_ref, q_ref = torch.linalg.eigh(stack)
+    for prev in (None, [None, None, None], list(q_ref.unbind(0))):
+        d, q = BaseKFACPreconditioner._group_eigh(
+            stack, layers, 'a', prev_override=prev,
+        )
+        rec = q @ torch.diag_embed(d) @ q.transpose(-1, -2)
+        assert torch.allclose(rec, stack, atol=1e-3), (
+            (rec - stack).abs().max()
+        )
+
+
+def test_async_worker_snapshots_prev_eigenbases() -> None:
+    """_launch_async_inverses must clone qa/qg into the work items (the
+    worker may run while the main thread frees the live tensors)."""
+    import inspect
+
+    from kfac_amd.base_preconditioner import BaseKFACPreconditioner
+
+    src = inspect.getsource(BaseKFACPreconditioner._launch_async_inverses)
+    assert 'prev_override' in src
+    assert '.detach().clone()' in src
+    sig = inspect.signature(BaseKFACPreconditioner._group_eigh)
+    assert 'prev_override' in sig.parameters
