@@ -254,6 +254,68 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
             loglevel=loglevel,
         )
 
+    def _compute_grad_scale(self) -> float | torch.Tensor:
+        """Tensor-parallel-consistent kl-clip scale.
+
+        The base implementation sums <precon_grad, grad> over this
+        rank's layer tensors (base_preconditioner.py). Under tensor
+        parallelism those tensors are SHARDS, so model-parallel peers
+        would compute different scales and apply them to shards of the
+        same logical weight — silent shard-inconsistency the reference
+        inherits (reference base_preconditioner.py:411-435 applied to
+        sharded grads). Here: weight-shard products are disjoint across
+        mp ranks and sum to the full-model product; replicated bias
+        columns (input parallelism keeps the full bias on every mp
+        rank) are down-weighted by 1/mp_world; the total is allreduced
+        over the model-parallel group so every peer applies the SAME
+        scale — equal to the single-process scale on the full model
+        (tests/test_gpt_neox.py::test_tp_training_matches_single_process).
+        """
+        import math
+
+        from kfac_amd import ops
+
+        layers = list(self._layers.values())
+        if len(layers) == 0:
+            return 1.0
+        mp_group = self.model_parallel_group
+        mp_world = (
+            torch.distributed.get_world_size(mp_group)
+            if mp_group is not None
+            else 1
+        )
+        if mp_world <= 1:
+            return super()._compute_grad_scale()
+        lr = self.lr
+        kl_clip = self.kl_clip
+        assert kl_clip is not None
+        device = layers[0][1].module.device
+        vg = torch.zeros((), dtype=torch.float32, device=device)
+        for _, layer in reversed(layers):
+            grad = layer.grad
+            if grad is None:
+                raise AssertionError(
+                    'layer gradient has not been preconditioned',
+                )
+            g32 = grad.to(torch.float32)
+            wgrad = layer.module.get_grad().to(torch.float32)
+            dot = (g32 * wgrad).sum()
+            if (
+                cast(GPTNeoXKFACEigenLayer, layer).parallelism == 'input'
+                and layer.module.has_bias()
+            ):
+                dot = dot - (g32[:, -1] * wgrad[:, -1]).sum() * (
+                    (mp_world - 1) / mp_world
+                )
+            vg += dot
+        torch.distributed.all_reduce(vg, group=mp_group)
+        if device.type == 'cuda' and ops.extension_available():
+            return ops.grad_scale_from_accum(vg, kl_clip, lr)
+        vg_sum = float(vg) * lr * lr
+        if vg_sum == 0.0:
+            return 1.0
+        return min(1.0, math.sqrt(kl_clip / abs(vg_sum)))
+
     # -- sharded checkpointing ---------------------------------------------
 
     def state_dict(self, include_factors: bool = True) -> dict[str, Any]:
