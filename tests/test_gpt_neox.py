@@ -360,3 +360,97 @@ def test_tp_training_matches_single_process(kl_clip: float | None) -> None:
             torch.testing.assert_close(
                 tp[k], full[k], rtol=1e-3, atol=1e-5,
             )
+
+
+def _tp_dp_mlp_training(tmpdir: str, kl_clip: float | None) -> None:
+    """world=4 as a 2x2 grid: mp groups [0,1],[2,3]; dp groups [0,2],[1,3]."""
+    from kfac_amd.gpt_neox import GPTNeoXKFACPreconditioner
+    from kfac_amd.gpt_neox.topology import PipeModelDataTopology
+    from testing.gpt_neox import FullMLP
+    from testing.gpt_neox import ShardedParallelMLP
+
+    rank = dist.get_rank()
+    mp_rank, dp_rank = rank % 2, rank // 2
+    topo = PipeModelDataTopology(num_pp=1, num_mp=2, num_dp=2)
+    mp_pg = [dist.new_group([0, 1]), dist.new_group([2, 3])]
+    dp_pg = [dist.new_group([0, 2]), dist.new_group([1, 3])]
+    mp_group = mp_pg[dp_rank]
+    dp_group = dp_pg[mp_rank]
+    torch.manual_seed(21)
+    full = FullMLP()
+    model = ShardedParallelMLP(full, mp_rank, 2, mp_group)
+    precon = GPTNeoXKFACPreconditioner(
+        model,
+        topology=topo,
+        data_parallel_group=dp_group,
+        model_parallel_group=mp_group,
+        kl_clip=kl_clip,
+        **_TP_HP,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=_TP_HP['lr'])
+    xs, ys = _tp_batches()
+    losses = []
+    half = 8
+    for x, y in zip(xs, ys):
+        xh = x[dp_rank * half : (dp_rank + 1) * half]
+        yh = y[dp_rank * half : (dp_rank + 1) * half]
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(xh), yh)
+        loss.backward()
+        # DDP-equivalent: average gradients over the dp replicas (K-FAC
+        # assumes pre-averaged grads, SURVEY.md 2.3)
+        for p in model.parameters():
+            dist.all_reduce(p.grad, group=dp_group)
+            p.grad /= 2
+        precon.step()
+        opt.step()
+        # full-batch loss = mean of the two half-batch losses
+        lt = loss.detach().clone()
+        dist.all_reduce(lt, group=dp_group)
+        losses.append(lt.item() / 2)
+    # assemble full weights from the mp shards (collective within mp group)
+    w1s = [torch.empty_like(model.dense_h_to_4h.weight) for _ in range(2)]
+    b1s = [torch.empty_like(model.dense_h_to_4h.bias) for _ in range(2)]
+    w2s = [torch.empty_like(model.dense_4h_to_h.weight) for _ in range(2)]
+    dist.all_gather(w1s, model.dense_h_to_4h.weight.data, group=mp_group)
+    dist.all_gather(b1s, model.dense_h_to_4h.bias.data, group=mp_group)
+    dist.all_gather(w2s, model.dense_4h_to_h.weight.data, group=mp_group)
+    w1, b1, w2 = torch.cat(w1s, 0), torch.cat(b1s, 0), torch.cat(w2s, 1)
+    # dp replicas must agree exactly on the assembled model
+    for t in (w1, b1, w2):
+        gathered = [torch.empty_like(t) for _ in range(4)]
+        dist.all_gather(gathered, t)
+        if rank == 0:
+            for g in gathered[1:]:
+                torch.testing.assert_close(g, gathered[0], rtol=1e-5, atol=1e-6)
+    if rank == 0:
+        torch.save(
+            {'losses': losses, 'w1': w1, 'b1': b1, 'w2': w2},
+            f'{tmpdir}/tp.pt',
+        )
+
+
+@pytest.mark.parametrize('kl_clip', [None, 0.001])
+def test_tp_dp_training_matches_single_process(kl_clip: float | None) -> None:
+    """2x2 (mp=2, dp=2) K-FAC == single-process K-FAC on the full model.
+
+    Half-batches per dp replica with DDP-style grad averaging; factors
+    dp-averaged by the layer's reduce routing; preconditioned grads
+    broadcast over dp (MEM-OPT); gather->precondition->scatter within
+    each mp group. Losses and assembled weights must match the world-1
+    run, and the two dp replicas must stay bitwise-consistent."""
+    with tempfile.TemporaryDirectory() as td:
+        run_distributed(4, _tp_dp_mlp_training, td, kl_clip)
+        run_distributed(1, _full_mlp_training, td, kl_clip)
+        tp = torch.load(f'{td}/tp.pt')
+        full = torch.load(f'{td}/full.pt')
+        torch.testing.assert_close(
+            torch.tensor(tp['losses']),
+            torch.tensor(full['losses']),
+            rtol=1e-4,
+            atol=1e-5,
+        )
+        for k in ('w1', 'b1', 'w2'):
+            torch.testing.assert_close(
+                tp[k], full[k], rtol=1e-3, atol=1e-5,
+            )
