@@ -72,14 +72,47 @@ class GPTNeoXAssignment(WorkAssignment):
             if topology.get_coord(r).pipe == self.pipe_parallel_rank
         ]
 
-        if set(self.pipe_parallel_peers) == set(self.model_parallel_peers):
+        # Reuse an existing group when the peer set coincides with it.
+        # A None mp/dp group means size-1 (no parallelism on that axis)
+        # and must NOT be reused: group=None is the GLOBAL group to
+        # torch.distributed, and with pp > 1 a factor allreduce over the
+        # global group would cross pipeline stages that own different
+        # layers (mismatched collectives).
+        if (
+            set(self.pipe_parallel_peers) == set(self.model_parallel_peers)
+            and self.model_parallel_group is not None
+        ):
             self.pipe_parallel_peer_group = self.model_parallel_group
-        elif set(self.pipe_parallel_peers) == set(self.data_parallel_peers):
+        elif (
+            set(self.pipe_parallel_peers) == set(self.data_parallel_peers)
+            and self.data_parallel_group is not None
+        ):
             self.pipe_parallel_peer_group = self.data_parallel_group
         else:
-            self.pipe_parallel_peer_group = dist.new_group(
-                self.pipe_parallel_peers,
-            )
+            # dist.new_group is collective over the WORLD: every rank
+            # must create every stage's peer group, in the same order,
+            # keeping its own. (Per-stage peer lists differ by rank, so
+            # a single new_group(self.pipe_parallel_peers) call would
+            # violate the collective contract.)
+            self.pipe_parallel_peer_group = None
+            if dist.is_initialized():
+                num_stages = 1 + max(
+                    topology.get_coord(r).pipe
+                    for r in range(topology.world_size())
+                )
+                for stage in range(num_stages):
+                    peers = [
+                        r
+                        for r in range(topology.world_size())
+                        if topology.get_coord(r).pipe == stage
+                    ]
+                    group = dist.new_group(peers)
+                    if stage == self.pipe_parallel_rank:
+                        self.pipe_parallel_peer_group = group
+                assert self.pipe_parallel_peer_group is not None
+            # else: serial/unit-test construction — no groups to make,
+            # and every collective is a no-op without an initialized
+            # world
 
         # Greedy lowest-load balance over the pipe peers (colocated
         # factors: MEM-OPT needs A and G on one rank).

@@ -454,3 +454,118 @@ def test_tp_dp_training_matches_single_process(kl_clip: float | None) -> None:
             torch.testing.assert_close(
                 tp[k], full[k], rtol=1e-3, atol=1e-5,
             )
+
+
+def _pp_stage_training(tmpdir: str) -> None:
+    """pp=2, mp=1, dp=1: each rank owns ONE layer of a replicated model.
+
+    Pipeline mechanics are simulated: both ranks run the full
+    forward/backward locally (identical data and params), each stage's
+    preconditioner is registered over only its own layer, and after the
+    optimizer step each layer's parameters are broadcast from the stage
+    that owns them — exactly the state a real pipeline would hold."""
+    from kfac_amd.gpt_neox import GPTNeoXKFACPreconditioner
+    from kfac_amd.gpt_neox.topology import PipeModelDataTopology
+    from testing.gpt_neox import ParallelMLP
+
+    rank = dist.get_rank()
+    topo = PipeModelDataTopology(num_pp=2, num_mp=1, num_dp=1)
+    dp_groups = [dist.new_group([r]) for r in range(2)]
+    torch.manual_seed(21)
+    model = ParallelMLP()
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    stage_name = 'dense_h_to_4h' if rank == 0 else 'dense_4h_to_h'
+    stage = torch.nn.ModuleDict({stage_name: getattr(model, stage_name)})
+    precon = GPTNeoXKFACPreconditioner(
+        stage,
+        topology=topo,
+        data_parallel_group=dp_groups[rank],
+        model_parallel_group=None,
+        kl_clip=None,  # stage-local by design; global clip needs pp comms
+        **_TP_HP,
+    )
+    # work balanced over this stage's single peer: itself
+    assert precon._assignment.pipe_parallel_peers == [rank]
+    opt = torch.optim.SGD(
+        getattr(model, stage_name).parameters(), lr=_TP_HP['lr'],
+    )
+    xs, ys = _tp_batches()
+    losses = []
+    for x, y in zip(xs, ys):
+        for p in model.parameters():
+            p.grad = None
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        # pipeline state exchange: each layer's params live on its stage
+        for p in model.dense_h_to_4h.parameters():
+            dist.broadcast(p.data, src=0)
+        for p in model.dense_4h_to_h.parameters():
+            dist.broadcast(p.data, src=1)
+        losses.append(loss.item())
+    if rank == 0:
+        torch.save(
+            {
+                'losses': losses,
+                'w1': model.dense_h_to_4h.weight.data,
+                'b1': model.dense_h_to_4h.bias.data,
+                'w2': model.dense_4h_to_h.weight.data,
+                'b2': model.dense_4h_to_h.bias.data,
+            },
+            f'{tmpdir}/pp.pt',
+        )
+
+
+def _serial_parallel_mlp_training(tmpdir: str) -> None:
+    from kfac_amd import KFACPreconditioner
+    from testing.gpt_neox import ParallelMLP
+
+    torch.manual_seed(21)
+    model = ParallelMLP()
+    precon = KFACPreconditioner(model, kl_clip=None, **_TP_HP)
+    opt = torch.optim.SGD(model.parameters(), lr=_TP_HP['lr'])
+    xs, ys = _tp_batches()
+    losses = []
+    for x, y in zip(xs, ys):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    torch.save(
+        {
+            'losses': losses,
+            'w1': model.dense_h_to_4h.weight.data,
+            'b1': model.dense_h_to_4h.bias.data,
+            'w2': model.dense_4h_to_h.weight.data,
+            'b2': model.dense_4h_to_h.bias.data,
+        },
+        f'{tmpdir}/serial.pt',
+    )
+
+
+def test_pp_training_matches_single_process() -> None:
+    """pp=2 per-stage K-FAC == single-process K-FAC on the same model.
+
+    Regression for the pipe-peer group selection: with mp=1 (group None)
+    and pp=2, the peer group must be the rank's own singleton — reusing
+    group=None (the GLOBAL group) would cross stages owning different
+    layers and hang or corrupt the factor allreduce."""
+    with tempfile.TemporaryDirectory() as td:
+        run_distributed(2, _pp_stage_training, td)
+        run_distributed(1, _serial_parallel_mlp_training, td)
+        pp = torch.load(f'{td}/pp.pt')
+        serial = torch.load(f'{td}/serial.pt')
+        torch.testing.assert_close(
+            torch.tensor(pp['losses']),
+            torch.tensor(serial['losses']),
+            rtol=1e-4,
+            atol=1e-5,
+        )
+        for k in ('w1', 'b1', 'w2', 'b2'):
+            torch.testing.assert_close(
+                pp[k], serial[k], rtol=1e-3, atol=1e-5,
+            )
