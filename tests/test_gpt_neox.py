@@ -569,3 +569,130 @@ def test_pp_training_matches_single_process() -> None:
             torch.testing.assert_close(
                 pp[k], serial[k], rtol=1e-3, atol=1e-5,
             )
+
+
+def _pp_tp_dp_training(tmpdir: str) -> None:
+    """world=8 as pp=2 x mp=2 x dp=2 — the full 3D grid.
+
+    rank = (pipe*2 + data)*2 + model. Each rank holds shard `model` of
+    BOTH layers (forward needs the whole model) but registers K-FAC only
+    over its own stage's layer; pipeline state exchange is simulated by
+    broadcasting each layer's shards from the stage that owns them.
+    This is the only configuration where pipe peers != mp peers != dp
+    peers, so the assignment must build per-stage peer groups itself."""
+    from kfac_amd.gpt_neox import GPTNeoXKFACPreconditioner
+    from kfac_amd.gpt_neox.topology import PipeModelDataTopology
+    from testing.gpt_neox import FullMLP
+    from testing.gpt_neox import ShardedParallelMLP
+
+    rank = dist.get_rank()
+    topo = PipeModelDataTopology(num_pp=2, num_mp=2, num_dp=2)
+    c = topo.get_coord(rank)
+    # every rank creates every group, same order (collective contract)
+    mp_group = dp_group = ex_group = None
+    for p in range(2):
+        for d in range(2):
+            g = dist.new_group([topo.get_rank(p, d, 0), topo.get_rank(p, d, 1)])
+            if (p, d) == (c.pipe, c.data):
+                mp_group = g
+    for p in range(2):
+        for m in range(2):
+            g = dist.new_group([topo.get_rank(p, 0, m), topo.get_rank(p, 1, m)])
+            if (p, m) == (c.pipe, c.model):
+                dp_group = g
+    for d in range(2):
+        for m in range(2):
+            g = dist.new_group([topo.get_rank(0, d, m), topo.get_rank(1, d, m)])
+            if (d, m) == (c.data, c.model):
+                ex_group = g
+    assert None not in (mp_group, dp_group, ex_group)
+
+    torch.manual_seed(21)
+    full = FullMLP()
+    model = ShardedParallelMLP(full, c.model, 2, mp_group)
+    stage_name = 'dense_h_to_4h' if c.pipe == 0 else 'dense_4h_to_h'
+    stage = torch.nn.ModuleDict({stage_name: getattr(model, stage_name)})
+    precon = GPTNeoXKFACPreconditioner(
+        stage,
+        topology=topo,
+        data_parallel_group=dp_group,
+        model_parallel_group=mp_group,
+        kl_clip=None,  # stage-local by design; see pp=2 test
+        **_TP_HP,
+    )
+    assert sorted(precon._assignment.pipe_parallel_peers) == sorted(
+        topo.get_rank(c.pipe, d, m) for d in range(2) for m in range(2)
+    )
+    opt = torch.optim.SGD(
+        getattr(model, stage_name).parameters(), lr=_TP_HP['lr'],
+    )
+    xs, ys = _tp_batches()
+    losses = []
+    half = 8
+    stage_srcs = {
+        'dense_h_to_4h': topo.get_rank(0, c.data, c.model),
+        'dense_4h_to_h': topo.get_rank(1, c.data, c.model),
+    }
+    for x, y in zip(xs, ys):
+        xh = x[c.data * half : (c.data + 1) * half]
+        yh = y[c.data * half : (c.data + 1) * half]
+        for p in model.parameters():
+            p.grad = None
+        loss = torch.nn.functional.cross_entropy(model(xh), yh)
+        loss.backward()
+        for p in model.parameters():
+            dist.all_reduce(p.grad, group=dp_group)
+            p.grad /= 2
+        precon.step()
+        opt.step()
+        # pipeline state exchange within each (data, model) column
+        for lname, src in stage_srcs.items():
+            for p in getattr(model, lname).parameters():
+                dist.broadcast(p.data, src=src, group=ex_group)
+        lt = loss.detach().clone()
+        dist.all_reduce(lt, group=dp_group)
+        losses.append(lt.item() / 2)
+    # assemble full weights from this rank's mp shards, then require all
+    # 8 ranks to agree bitwise
+    w1s = [torch.empty_like(model.dense_h_to_4h.weight) for _ in range(2)]
+    b1s = [torch.empty_like(model.dense_h_to_4h.bias) for _ in range(2)]
+    w2s = [torch.empty_like(model.dense_4h_to_h.weight) for _ in range(2)]
+    dist.all_gather(w1s, model.dense_h_to_4h.weight.data, group=mp_group)
+    dist.all_gather(b1s, model.dense_h_to_4h.bias.data, group=mp_group)
+    dist.all_gather(w2s, model.dense_4h_to_h.weight.data, group=mp_group)
+    w1, b1, w2 = torch.cat(w1s, 0), torch.cat(b1s, 0), torch.cat(w2s, 1)
+    for t in (w1, b1, w2):
+        gathered = [torch.empty_like(t) for _ in range(8)]
+        dist.all_gather(gathered, t)
+        if rank == 0:
+            for g in gathered[1:]:
+                torch.testing.assert_close(g, gathered[0], rtol=1e-5, atol=1e-6)
+    if rank == 0:
+        torch.save(
+            {'losses': losses, 'w1': w1, 'b1': b1, 'w2': w2},
+            f'{tmpdir}/grid.pt',
+        )
+
+
+def test_3d_grid_training_matches_single_process() -> None:
+    """pp=2 x mp=2 x dp=2 (world 8) == single-process K-FAC.
+
+    The full 3D protocol at once: per-stage work assignment over 4 pipe
+    peers (built via the fallback group path), TP gather/precondition/
+    scatter within mp pairs, dp-averaged factors and MEM-OPT grad
+    broadcasts, pipeline state exchange across stages."""
+    with tempfile.TemporaryDirectory() as td:
+        run_distributed(8, _pp_tp_dp_training, td)
+        run_distributed(1, _full_mlp_training, td, None)
+        grid = torch.load(f'{td}/grid.pt')
+        full = torch.load(f'{td}/full.pt')
+        torch.testing.assert_close(
+            torch.tensor(grid['losses']),
+            torch.tensor(full['losses']),
+            rtol=1e-4,
+            atol=1e-5,
+        )
+        for k in ('w1', 'b1', 'w2'):
+            torch.testing.assert_close(
+                grid[k], full[k], rtol=1e-3, atol=1e-5,
+            )
