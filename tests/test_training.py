@@ -74,7 +74,7 @@ def _lenet_case() -> None:
     assert losses[0] > losses[-1], losses
 
 
-def _dist_case(strategy_name: str) -> None:
+def _dist_case(strategy_name: str, **kfac_kwargs) -> None:
     strategy = DistributedStrategy[strategy_name]
     torch.manual_seed(42)
     model = TinyModel()
@@ -82,7 +82,9 @@ def _dist_case(strategy_name: str) -> None:
         torch.distributed.broadcast(p.data, src=0)
     x = torch.randn(32, 10)
     y = torch.randint(0, 3, (32,))
-    losses = _train(model, x, y, steps=10, grad_worker_fraction=strategy)
+    losses = _train(
+        model, x, y, steps=10, grad_worker_fraction=strategy, **kfac_kwargs,
+    )
     assert losses[0] > losses[-1], losses
 
 
@@ -171,3 +173,20 @@ def test_distributed_training_mem_opt_world8() -> None:
     # BASELINE.json's grad_worker_fraction=1/8 config shape: every
     # layer has ONE grad worker broadcasting to the other 7 ranks.
     run_distributed(8, _dist_case, 'MEM_OPT')
+
+
+def test_distributed_training_inverse_method_hybrid() -> None:
+    """INVERSE method across ranks: explicit A_inv/G_inv broadcast from
+    the inverse workers to the grad-worker groups (layers/inverse.py)."""
+    run_distributed(2, _dist_case, 'HYBRID_OPT', compute_method='inverse')
+
+
+def test_distributed_training_inverse_method_mem_opt_world4() -> None:
+    run_distributed(4, _dist_case, 'MEM_OPT', compute_method='inverse')
+
+
+def test_distributed_training_symmetry_aware_hybrid() -> None:
+    """Triu-packed factor allreduce + inverse broadcast wire format."""
+    run_distributed(
+        2, _dist_case, 'HYBRID_OPT', symmetry_aware=True,
+    )
