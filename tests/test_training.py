@@ -10,6 +10,8 @@ from __future__ import annotations
 
 import pickle
 import sys
+
+import pytest
 import tempfile
 
 import torch
@@ -120,18 +122,23 @@ def test_distributed_training_comm_opt() -> None:
     run_distributed(4, _dist_case, 'COMM_OPT')
 
 
-def _single_reference(path: str) -> None:
+def _single_reference(path: str, **kfac_kwargs) -> None:
     torch.manual_seed(7)
     model = TinyModel()
     x = torch.randn(32, 10)
     y = torch.randint(0, 3, (32,))
     sd0 = {k: v.clone() for k, v in model.state_dict().items()}
-    losses = _train(model, x, y, steps=5, dist_avg=False)
+    # strategy kwargs are world-size-dependent; only the compute method
+    # changes the single-process math
+    ref_kwargs = {
+        k: v for k, v in kfac_kwargs.items() if k == 'compute_method'
+    }
+    losses = _train(model, x, y, steps=5, dist_avg=False, **ref_kwargs)
     with open(path, 'wb') as f:
         pickle.dump((sd0, x, y, losses), f)
 
 
-def _half_batch_distributed(path: str) -> None:
+def _half_batch_distributed(path: str, **kfac_kwargs) -> None:
     with open(path, 'rb') as fh:
         sd0, x, y, losses_single = pickle.load(fh)
     rank = torch.distributed.get_rank()
@@ -140,7 +147,7 @@ def _half_batch_distributed(path: str) -> None:
     half = x.size(0) // 2
     xs = x[rank * half : (rank + 1) * half]
     ys = y[rank * half : (rank + 1) * half]
-    losses = _train(model, xs, ys, steps=5)
+    losses = _train(model, xs, ys, steps=5, **kfac_kwargs)
     lt = torch.tensor(losses)
     torch.distributed.all_reduce(lt)
     lt /= 2
@@ -149,16 +156,28 @@ def _half_batch_distributed(path: str) -> None:
     )
 
 
-def test_distributed_matches_single_process() -> None:
+@pytest.mark.parametrize(
+    'kfac_kwargs',
+    [
+        {},
+        {'grad_worker_fraction': DistributedStrategy.HYBRID_OPT},
+        {'grad_worker_fraction': DistributedStrategy.MEM_OPT},
+        {'compute_method': 'inverse'},
+    ],
+    ids=['comm-opt', 'hybrid-opt', 'mem-opt', 'inverse'],
+)
+def test_distributed_matches_single_process(kfac_kwargs) -> None:
     """World-2 training on half batches == single-process on the full batch.
 
-    Validates the whole distributed pipeline numerically: factor
-    allreduce-averaging + DDP-style grad averaging reproduce the
-    single-process K-FAC trajectory.
+    Validates the whole distributed pipeline numerically for every
+    worker-placement strategy AND the explicit-inverse method: factor
+    allreduce-averaging, inverse/eigen broadcasts, gradient broadcasts
+    (HYBRID/MEM-OPT) and DDP-style grad averaging together reproduce
+    the single-process K-FAC trajectory.
     """
     with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
-        run_distributed(1, _single_reference, f.name)
-        run_distributed(2, _half_batch_distributed, f.name)
+        run_distributed(1, _single_reference, f.name, **kfac_kwargs)
+        run_distributed(2, _half_batch_distributed, f.name, **kfac_kwargs)
 
 
 def test_distributed_training_hybrid_world8() -> None:
