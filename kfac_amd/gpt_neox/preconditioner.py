@@ -376,20 +376,49 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
             return
         by_name = {name: layer for name, layer in self._layers.values()}
         for found_name, lsd in layers.items():
-            if (
-                found_name in by_name
-                and cast(
+            if found_name not in by_name:
+                continue
+            layer = by_name[found_name]
+            is_worker = (
+                cast(
                     GPTNeoXAssignment, self._assignment,
                 ).factor_worker(found_name, 'A')
                 == get_rank()
-            ):
-                layer = by_name[found_name]
-                layer.load_state_dict(lsd)
-                if compute_inverses:
-                    layer.compute_a_inv(damping=self.damping)
-                    layer.compute_g_inv(damping=self.damping)
+            )
+            # The unsharded-dim factor is allreduce-AVERAGED over all
+            # pipe peers every reduce, so every peer holds (and feeds
+            # back) a local copy: restore it everywhere, or the first
+            # post-resume reduce averages in a freshly-initialized
+            # factor and the trajectory drifts (the reference restores
+            # only on the worker and inherits that drift —
+            # gpt_neox/preconditioner.py:316-350). The sharded-dim
+            # factor is gathered to the primary and exists only there.
+            restored = self._peer_visible_factors(layer, lsd, is_worker)
+            layer.load_state_dict(restored)
+            if compute_inverses and is_worker:
+                layer.compute_a_inv(damping=self.damping)
+                layer.compute_g_inv(damping=self.damping)
         if torch.distributed.is_initialized():
             torch.distributed.barrier()
+
+    @staticmethod
+    def _peer_visible_factors(
+        layer: KFACBaseLayer,
+        lsd: dict[str, Any],
+        is_worker: bool,
+    ) -> dict[str, Any]:
+        """Restrict a factor state dict to what this rank holds live.
+
+        Workers hold both factors; non-worker pipe peers hold only the
+        unsharded-dim factor (G under input parallelism, A under output
+        parallelism).
+        """
+        if is_worker:
+            return lsd
+        restored = dict(lsd)
+        parallelism = cast(GPTNeoXKFACEigenLayer, layer).parallelism
+        restored['A' if parallelism == 'input' else 'G'] = None
+        return restored
 
     def load_factors_from_dir(self, compute_inverses: bool = True) -> None:
         """Load per-layer factor files from ``factor_checkpoint_dir``."""
@@ -404,16 +433,22 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
             return
         for name, layer in self._layers.values():
             assignment = cast(GPTNeoXAssignment, self._assignment)
-            if assignment.factor_worker(name, 'A') == get_rank():
-                filepath = os.path.join(self.factor_checkpoint_dir, name)
-                if os.path.exists(filepath):
-                    logger.info(
-                        f'loading KFAC factors for {name} on rank {get_rank()}',
-                    )
-                    layer.load_state_dict(torch.load(filepath))
-                    if compute_inverses:
-                        layer.compute_a_inv(damping=self.damping)
-                        layer.compute_g_inv(damping=self.damping)
+            is_worker = assignment.factor_worker(name, 'A') == get_rank()
+            filepath = os.path.join(self.factor_checkpoint_dir, name)
+            if os.path.exists(filepath):
+                logger.info(
+                    f'loading KFAC factors for {name} on rank {get_rank()}',
+                )
+                # non-worker peers restore their unsharded-dim factor
+                # copy too (see load_state_dict)
+                layer.load_state_dict(
+                    self._peer_visible_factors(
+                        layer, torch.load(filepath), is_worker,
+                    ),
+                )
+                if compute_inverses and is_worker:
+                    layer.compute_a_inv(damping=self.damping)
+                    layer.compute_g_inv(damping=self.damping)
 
     def save_factors_to_dir(self) -> None:
         """Each inverse worker writes one factor file per owned layer."""

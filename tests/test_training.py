@@ -209,3 +209,40 @@ def test_distributed_training_symmetry_aware_hybrid() -> None:
     run_distributed(
         2, _dist_case, 'HYBRID_OPT', symmetry_aware=True,
     )
+
+
+def _single_reference_lenet(path: str) -> None:
+    torch.manual_seed(9)
+    model = LeNet()
+    x = torch.randn(16, 1, 28, 28)
+    y = torch.randint(0, 10, (16,))
+    sd0 = {k: v.clone() for k, v in model.state_dict().items()}
+    losses = _train(model, x, y, steps=4, dist_avg=False)
+    with open(path, 'wb') as f:
+        pickle.dump((sd0, x, y, losses), f)
+
+
+def _half_batch_distributed_lenet(path: str) -> None:
+    with open(path, 'rb') as fh:
+        sd0, x, y, losses_single = pickle.load(fh)
+    rank = torch.distributed.get_rank()
+    model = LeNet()
+    model.load_state_dict(sd0)
+    half = x.size(0) // 2
+    xs = x[rank * half : (rank + 1) * half]
+    ys = y[rank * half : (rank + 1) * half]
+    losses = _train(model, xs, ys, steps=4)
+    lt = torch.tensor(losses)
+    torch.distributed.all_reduce(lt)
+    lt /= 2
+    torch.testing.assert_close(
+        lt, torch.tensor(losses_single), rtol=1e-3, atol=1e-4,
+    )
+
+
+def test_distributed_conv_matches_single_process() -> None:
+    """Conv2d factor path (im2col A, spatial-averaged G) under world-2:
+    half-batch training must reproduce the single-process trajectory."""
+    with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
+        run_distributed(1, _single_reference_lenet, f.name)
+        run_distributed(2, _half_batch_distributed_lenet, f.name)
