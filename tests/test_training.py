@@ -246,3 +246,65 @@ def test_distributed_conv_matches_single_process() -> None:
     with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
         run_distributed(1, _single_reference_lenet, f.name)
         run_distributed(2, _half_batch_distributed_lenet, f.name)
+
+
+def _resume_case(strategy_name: str) -> None:
+    """Interrupt at step 6, checkpoint, rebuild, resume 4 more steps —
+    must match an uninterrupted 10-step run exactly."""
+    strategy = DistributedStrategy[strategy_name]
+    rank = torch.distributed.get_rank()
+    world = torch.distributed.get_world_size()
+    g = torch.Generator().manual_seed(13)
+    xs = [torch.randn(16, 10, generator=g) for _ in range(10)]
+    ys = [torch.randint(0, 3, (16,), generator=g) for _ in range(10)]
+
+    def build():
+        torch.manual_seed(3)
+        model = TinyModel()
+        precon = KFACPreconditioner(
+            model,
+            factor_update_steps=1,
+            inv_update_steps=2,
+            lr=0.05,
+            grad_worker_fraction=strategy,
+        )
+        return model, precon
+
+    def run(model, precon, lo, hi):
+        opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        out = []
+        half = 16 // world
+        for x, y in zip(xs[lo:hi], ys[lo:hi]):
+            xb = x[rank * half : (rank + 1) * half]
+            yb = y[rank * half : (rank + 1) * half]
+            opt.zero_grad()
+            loss = torch.nn.functional.cross_entropy(model(xb), yb)
+            loss.backward()
+            for p_ in model.parameters():
+                torch.distributed.all_reduce(p_.grad)
+                p_.grad /= world
+            precon.step()
+            opt.step()
+            out.append(loss.item())
+        return out
+
+    model, precon = build()
+    full = run(model, precon, 0, 10)
+
+    model, precon = build()
+    part = run(model, precon, 0, 6)
+    sd = precon.state_dict()
+    msd = {k: v.clone() for k, v in model.state_dict().items()}
+    model2, precon2 = build()
+    model2.load_state_dict(msd)
+    precon2.load_state_dict(sd, compute_inverses=True)
+    assert precon2.steps == precon.steps
+    part += run(model2, precon2, 6, 10)
+    torch.testing.assert_close(
+        torch.tensor(part), torch.tensor(full), rtol=1e-4, atol=1e-6,
+    )
+
+
+@pytest.mark.parametrize('strategy', ['COMM_OPT', 'HYBRID_OPT'])
+def test_distributed_resume_matches_uninterrupted(strategy: str) -> None:
+    run_distributed(2, _resume_case, strategy)
