@@ -255,7 +255,7 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
         )
 
     def _compute_grad_scale(self) -> float | torch.Tensor:
-        """Tensor-parallel-consistent kl-clip scale.
+        """Model-consistent kl-clip scale under 3D parallelism.
 
         The base implementation sums <precon_grad, grad> over this
         rank's layer tensors (base_preconditioner.py). Under tensor
@@ -268,8 +268,14 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
         columns (input parallelism keeps the full bias on every mp
         rank) are down-weighted by 1/mp_world; the total is allreduced
         over the model-parallel group so every peer applies the SAME
-        scale — equal to the single-process scale on the full model
-        (tests/test_gpt_neox.py::test_tp_training_matches_single_process).
+        scale. Under pipeline parallelism each stage only sees its own
+        layers; when ``pipeline_parallel_group`` was provided the stage
+        sums are additionally allreduced over it, making the scale the
+        TRUE full-model Fisher-norm clip — equal to the single-process
+        scale (tests/test_gpt_neox.py::
+        test_tp_training_matches_single_process and
+        test_3d_grid_training_matches_single_process). Without a pipe
+        group the scale stays stage-local like the reference.
         """
         import math
 
@@ -284,7 +290,13 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
             if mp_group is not None
             else 1
         )
-        if mp_world <= 1:
+        pp_group = self.pipeline_parallel_group
+        pp_world = (
+            torch.distributed.get_world_size(pp_group)
+            if pp_group is not None
+            else 1
+        )
+        if mp_world <= 1 and pp_world <= 1:
             return super()._compute_grad_scale()
         lr = self.lr
         kl_clip = self.kl_clip
@@ -301,14 +313,20 @@ class GPTNeoXKFACPreconditioner(BaseKFACPreconditioner):
             wgrad = layer.module.get_grad().to(torch.float32)
             dot = (g32 * wgrad).sum()
             if (
-                cast(GPTNeoXKFACEigenLayer, layer).parallelism == 'input'
+                mp_world > 1
+                and cast(GPTNeoXKFACEigenLayer, layer).parallelism == 'input'
                 and layer.module.has_bias()
             ):
                 dot = dot - (g32[:, -1] * wgrad[:, -1]).sum() * (
                     (mp_world - 1) / mp_world
                 )
             vg += dot
-        torch.distributed.all_reduce(vg, group=mp_group)
+        if mp_world > 1:
+            torch.distributed.all_reduce(vg, group=mp_group)
+        if pp_world > 1:
+            # pipe peers hold disjoint layer sets: summing stage sums
+            # yields the full-model <precon_grad, grad>
+            torch.distributed.all_reduce(vg, group=pp_group)
         if device.type == 'cuda' and ops.extension_available():
             return ops.grad_scale_from_accum(vg, kl_clip, lr)
         vg_sum = float(vg) * lr * lr

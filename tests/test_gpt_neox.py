@@ -456,7 +456,7 @@ def test_tp_dp_training_matches_single_process(kl_clip: float | None) -> None:
             )
 
 
-def _pp_stage_training(tmpdir: str) -> None:
+def _pp_stage_training(tmpdir: str, kl_clip: float | None) -> None:
     """pp=2, mp=1, dp=1: each rank owns ONE layer of a replicated model.
 
     Pipeline mechanics are simulated: both ranks run the full
@@ -471,6 +471,7 @@ def _pp_stage_training(tmpdir: str) -> None:
     rank = dist.get_rank()
     topo = PipeModelDataTopology(num_pp=2, num_mp=1, num_dp=1)
     dp_groups = [dist.new_group([r]) for r in range(2)]
+    pp_group = dist.new_group([0, 1])
     torch.manual_seed(21)
     model = ParallelMLP()
     for p in model.parameters():
@@ -482,7 +483,9 @@ def _pp_stage_training(tmpdir: str) -> None:
         topology=topo,
         data_parallel_group=dp_groups[rank],
         model_parallel_group=None,
-        kl_clip=None,  # stage-local by design; global clip needs pp comms
+        pipeline_parallel_group=pp_group,
+        kl_clip=kl_clip,  # with the pipe group provided, the clip is
+        # the TRUE full-model scale (stage sums allreduced over it)
         **_TP_HP,
     )
     # work balanced over this stage's single peer: itself
@@ -518,13 +521,13 @@ def _pp_stage_training(tmpdir: str) -> None:
         )
 
 
-def _serial_parallel_mlp_training(tmpdir: str) -> None:
+def _serial_parallel_mlp_training(tmpdir: str, kl_clip: float | None) -> None:
     from kfac_amd import KFACPreconditioner
     from testing.gpt_neox import ParallelMLP
 
     torch.manual_seed(21)
     model = ParallelMLP()
-    precon = KFACPreconditioner(model, kl_clip=None, **_TP_HP)
+    precon = KFACPreconditioner(model, kl_clip=kl_clip, **_TP_HP)
     opt = torch.optim.SGD(model.parameters(), lr=_TP_HP['lr'])
     xs, ys = _tp_batches()
     losses = []
@@ -547,16 +550,19 @@ def _serial_parallel_mlp_training(tmpdir: str) -> None:
     )
 
 
-def test_pp_training_matches_single_process() -> None:
+@pytest.mark.parametrize('kl_clip', [None, 0.001])
+def test_pp_training_matches_single_process(kl_clip: float | None) -> None:
     """pp=2 per-stage K-FAC == single-process K-FAC on the same model.
 
     Regression for the pipe-peer group selection: with mp=1 (group None)
     and pp=2, the peer group must be the rank's own singleton — reusing
     group=None (the GLOBAL group) would cross stages owning different
-    layers and hang or corrupt the factor allreduce."""
+    layers and hang or corrupt the factor allreduce. With kl_clip set,
+    the stage sums must combine over the provided pipeline group into
+    the exact full-model clip scale."""
     with tempfile.TemporaryDirectory() as td:
-        run_distributed(2, _pp_stage_training, td)
-        run_distributed(1, _serial_parallel_mlp_training, td)
+        run_distributed(2, _pp_stage_training, td, kl_clip)
+        run_distributed(1, _serial_parallel_mlp_training, td, kl_clip)
         pp = torch.load(f'{td}/pp.pt')
         serial = torch.load(f'{td}/serial.pt')
         torch.testing.assert_close(
@@ -571,7 +577,7 @@ def test_pp_training_matches_single_process() -> None:
             )
 
 
-def _pp_tp_dp_training(tmpdir: str) -> None:
+def _pp_tp_dp_training(tmpdir: str, kl_clip: float | None) -> None:
     """world=8 as pp=2 x mp=2 x dp=2 — the full 3D grid.
 
     rank = (pipe*2 + data)*2 + model. Each rank holds shard `model` of
@@ -617,7 +623,8 @@ def _pp_tp_dp_training(tmpdir: str) -> None:
         topology=topo,
         data_parallel_group=dp_group,
         model_parallel_group=mp_group,
-        kl_clip=None,  # stage-local by design; see pp=2 test
+        pipeline_parallel_group=ex_group,
+        kl_clip=kl_clip,
         **_TP_HP,
     )
     assert sorted(precon._assignment.pipe_parallel_peers) == sorted(
@@ -674,16 +681,19 @@ def _pp_tp_dp_training(tmpdir: str) -> None:
         )
 
 
-def test_3d_grid_training_matches_single_process() -> None:
+@pytest.mark.parametrize('kl_clip', [None, 0.001])
+def test_3d_grid_training_matches_single_process(kl_clip: float | None) -> None:
     """pp=2 x mp=2 x dp=2 (world 8) == single-process K-FAC.
 
     The full 3D protocol at once: per-stage work assignment over 4 pipe
     peers (built via the fallback group path), TP gather/precondition/
     scatter within mp pairs, dp-averaged factors and MEM-OPT grad
-    broadcasts, pipeline state exchange across stages."""
+    broadcasts, pipeline state exchange across stages — with kl-clip,
+    the mp-allreduced stage sums combine over the pipe axis into the
+    exact single-process clip scale."""
     with tempfile.TemporaryDirectory() as td:
-        run_distributed(8, _pp_tp_dp_training, td)
-        run_distributed(1, _full_mlp_training, td, None)
+        run_distributed(8, _pp_tp_dp_training, td, kl_clip)
+        run_distributed(1, _full_mlp_training, td, kl_clip)
         grid = torch.load(f'{td}/grid.pt')
         full = torch.load(f'{td}/full.pt')
         torch.testing.assert_close(
