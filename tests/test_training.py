@@ -308,3 +308,70 @@ def _resume_case(strategy_name: str) -> None:
 @pytest.mark.parametrize('strategy', ['COMM_OPT', 'HYBRID_OPT'])
 def test_distributed_resume_matches_uninterrupted(strategy: str) -> None:
     run_distributed(2, _resume_case, strategy)
+
+
+def _accum_single_reference(path: str) -> None:
+    torch.manual_seed(17)
+    model = TinyModel()
+    x = torch.randn(32, 10)
+    y = torch.randint(0, 3, (32,))
+    sd0 = {k: v.clone() for k, v in model.state_dict().items()}
+    losses = _train(model, x, y, steps=5, dist_avg=False)
+    with open(path, 'wb') as f:
+        pickle.dump((sd0, x, y, losses), f)
+
+
+def _accum_distributed(path: str) -> None:
+    """world-2, accumulation_steps=2: each rank sees two 8-sample
+    micro-batches per optimizer step. Losses use sum-reduction divided
+    by the GLOBAL batch so every sample's grad-output scale matches the
+    single 32-sample pass exactly — G factors are covariances of the
+    raw grad-outputs, so per-sample scaling is part of the semantics.
+    Factors accumulate per mini-step and must average to the full-batch
+    covariance."""
+    with open(path, 'rb') as fh:
+        sd0, x, y, losses_single = pickle.load(fh)
+    rank = torch.distributed.get_rank()
+    model = TinyModel()
+    model.load_state_dict(sd0)
+    precon = KFACPreconditioner(
+        model,
+        factor_update_steps=1,
+        inv_update_steps=2,
+        lr=0.01,
+        accumulation_steps=2,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    quarter = 8
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        total = 0.0
+        for micro in range(2):
+            lo = (2 * rank + micro) * quarter
+            xb, yb = x[lo : lo + quarter], y[lo : lo + quarter]
+            loss = (
+                torch.nn.functional.cross_entropy(
+                    model(xb), yb, reduction='sum',
+                )
+                / 32
+            )
+            loss.backward()
+            total += loss.item()
+        for p_ in model.parameters():
+            torch.distributed.all_reduce(p_.grad)
+        precon.step()
+        opt.step()
+        losses.append(total)
+    lt = torch.tensor(losses)
+    torch.distributed.all_reduce(lt)
+    torch.testing.assert_close(
+        lt, torch.tensor(losses_single), rtol=1e-3, atol=1e-4,
+    )
+
+
+def test_gradient_accumulation_distributed_matches_single() -> None:
+    """2 ranks x 2 micro-batches == one 32-sample single-process step."""
+    with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
+        run_distributed(1, _accum_single_reference, f.name)
+        run_distributed(2, _accum_distributed, f.name)
