@@ -163,8 +163,18 @@ def _half_batch_distributed(path: str, **kfac_kwargs) -> None:
         {'grad_worker_fraction': DistributedStrategy.HYBRID_OPT},
         {'grad_worker_fraction': DistributedStrategy.MEM_OPT},
         {'compute_method': 'inverse'},
+        {'symmetry_aware': True},
+        # prediv (eigenvalue outer product) requires colocated factors
+        {'colocate_factors': False, 'compute_eigenvalue_outer_product': False},
     ],
-    ids=['comm-opt', 'hybrid-opt', 'mem-opt', 'inverse'],
+    ids=[
+        'comm-opt',
+        'hybrid-opt',
+        'mem-opt',
+        'inverse',
+        'symmetry-aware',
+        'no-colocate',
+    ],
 )
 def test_distributed_matches_single_process(kfac_kwargs) -> None:
     """World-2 training on half batches == single-process on the full batch.
@@ -375,3 +385,49 @@ def test_gradient_accumulation_distributed_matches_single() -> None:
     with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
         run_distributed(1, _accum_single_reference, f.name)
         run_distributed(2, _accum_distributed, f.name)
+
+
+def _amp_scaled_case() -> None:
+    """grad_scaler semantics: a loss scaled by S (grad-outputs S-scaled
+    in the hooks, corrected by the 1/S^2 covariance coefficient) plus an
+    unscale_ of .grad before step() must reproduce the unscaled run."""
+    scale = 1024.0
+    torch.manual_seed(31)
+    model_a = TinyModel()
+    model_b = TinyModel()
+    model_b.load_state_dict(model_a.state_dict())
+    x = torch.randn(32, 10)
+    y = torch.randint(0, 3, (32,))
+
+    pa = KFACPreconditioner(
+        model_a, factor_update_steps=1, inv_update_steps=2, lr=0.01,
+    )
+    pb = KFACPreconditioner(
+        model_b,
+        factor_update_steps=1,
+        inv_update_steps=2,
+        lr=0.01,
+        grad_scaler=lambda: scale,
+    )
+    oa = torch.optim.SGD(model_a.parameters(), lr=0.01)
+    ob = torch.optim.SGD(model_b.parameters(), lr=0.01)
+    for _ in range(6):
+        oa.zero_grad()
+        torch.nn.functional.cross_entropy(model_a(x), y).backward()
+        pa.step()
+        oa.step()
+
+        ob.zero_grad()
+        (torch.nn.functional.cross_entropy(model_b(x), y) * scale).backward()
+        for q in model_b.parameters():  # scaler.unscale_(optimizer)
+            q.grad /= scale
+        pb.step()
+        ob.step()
+    for ka, kb in zip(
+        model_a.state_dict().values(), model_b.state_dict().values(),
+    ):
+        torch.testing.assert_close(ka, kb, rtol=1e-4, atol=1e-6)
+
+
+def test_amp_grad_scaler_matches_unscaled() -> None:
+    run_distributed(1, _amp_scaled_case)
