@@ -69,3 +69,35 @@ def test_cifar_example_no_kfac() -> None:
             ],
         )
         assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_gpt_neox_mlp_example_runs() -> None:
+    """2-way tensor-parallel example CLI under torchrun/gloo."""
+    env = dict(os.environ)
+    env['MASTER_ADDR'] = '127.0.0.1'
+    env['MASTER_PORT'] = '29611'
+    r = subprocess.run(
+        [
+            sys.executable,
+            '-m',
+            'torch.distributed.run',
+            '--nproc-per-node',
+            '2',
+            '--master-addr',
+            '127.0.0.1',
+            '--master-port',
+            '29611',
+            'examples/torch_gpt_neox_mlp.py',
+            '--steps',
+            '12',
+            '--backend',
+            'gloo',
+        ],
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=240,
+        env=env,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert 'done' in r.stdout
