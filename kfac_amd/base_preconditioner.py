@@ -662,6 +662,25 @@ class BaseKFACPreconditioner:
             for _, layer in layers
         )
 
+    @staticmethod
+    def _group_async_work(
+        work: list[tuple[Any, ...]],
+        which: str,
+    ) -> dict[int, list[tuple[Any, torch.Tensor, Any]]]:
+        """Group async work items (layer, a, g, qa_prev, qg_prev) by
+        factor size for the requested side; items whose factor is not
+        owned by this rank (None) drop out, keeping the prev-eigenbasis
+        snapshot aligned with its factor."""
+        groups: dict[int, list[tuple[Any, torch.Tensor, Any]]] = {}
+        for layer, a, g, qa_c, qg_c in work:
+            fac = a if which == 'a' else g
+            prev = qa_c if which == 'a' else qg_c
+            if fac is not None:
+                groups.setdefault(fac.shape[0], []).append(
+                    (layer, fac, prev),
+                )
+        return groups
+
     def _launch_async_inverses(self) -> None:
         import threading
 
@@ -709,23 +728,13 @@ class BaseKFACPreconditioner:
 
         def worker() -> None:
             try:
-                from collections import defaultdict as dd
-
                 with torch.cuda.stream(stream):
                     stream.wait_event(ready)
                     results: dict[Any, dict[str, torch.Tensor | None]] = {}
                     for which in ('a', 'g'):
-                        groups: dict[
-                            int,
-                            list[tuple[Any, torch.Tensor, Any]],
-                        ] = dd(list)
-                        for layer, a, g, qa_c, qg_c in work:
-                            fac = a if which == 'a' else g
-                            prev = qa_c if which == 'a' else qg_c
-                            if fac is not None:
-                                groups[fac.shape[0]].append(
-                                    (layer, fac, prev),
-                                )
+                        groups = BaseKFACPreconditioner._group_async_work(
+                            work, which,
+                        )
                         for n, items in groups.items():
                             stack = torch.stack(
                                 [f.to(torch.float32) for _, f, _ in items],

@@ -159,3 +159,28 @@ def test_async_worker_snapshots_prev_eigenbases() -> None:
     assert '.detach().clone()' in src
     sig = inspect.signature(BaseKFACPreconditioner._group_eigh)
     assert 'prev_override' in sig.parameters
+
+
+def test_async_work_grouping_alignment() -> None:
+    """The async worker's grouping must keep each factor aligned with
+    ITS layer's snapshot eigenbasis across mixed ownership."""
+    from kfac_amd.base_preconditioner import BaseKFACPreconditioner
+
+    la, lb, lc = object(), object(), object()
+    a1, a2 = torch.zeros(8, 8), torch.zeros(8, 8)
+    g1 = torch.zeros(4, 4)
+    qa1, qg1 = torch.ones(8, 8), torch.ones(4, 4)
+    work = [
+        (la, a1, g1, qa1, qg1),   # owns both
+        (lb, a2, None, None, None),  # owns A only, cold (no prev)
+        (lc, None, None, None, None),  # owns neither
+    ]
+    ga = BaseKFACPreconditioner._group_async_work(work, 'a')
+    gg = BaseKFACPreconditioner._group_async_work(work, 'g')
+    assert set(ga.keys()) == {8} and set(gg.keys()) == {4}
+    assert [(lyr, p is qa1) for lyr, _, p in ga[8]] == [
+        (la, True),
+        (lb, False),
+    ]
+    assert ga[8][1][2] is None
+    assert gg[4] == [(la, g1, qg1)]
