@@ -73,9 +73,13 @@ def test_cifar_example_no_kfac() -> None:
 
 def test_gpt_neox_mlp_example_runs() -> None:
     """2-way tensor-parallel example CLI under torchrun/gloo."""
+    sys.path.insert(0, REPO)
+    from testing.distributed import find_free_port
+
+    port = str(find_free_port())
     env = dict(os.environ)
     env['MASTER_ADDR'] = '127.0.0.1'
-    env['MASTER_PORT'] = '29611'
+    env['MASTER_PORT'] = port
     r = subprocess.run(
         [
             sys.executable,
@@ -86,7 +90,7 @@ def test_gpt_neox_mlp_example_runs() -> None:
             '--master-addr',
             '127.0.0.1',
             '--master-port',
-            '29611',
+            port,
             'examples/torch_gpt_neox_mlp.py',
             '--steps',
             '12',
