@@ -431,3 +431,37 @@ def _amp_scaled_case() -> None:
 
 def test_amp_grad_scaler_matches_unscaled() -> None:
     run_distributed(1, _amp_scaled_case)
+
+
+def _reduced_precision_case() -> None:
+    """inv_dtype=bf16 second-order state trains stably and is actually
+    stored reduced (reference supports reduced-precision
+    eigendecomposition storage)."""
+    torch.manual_seed(5)
+    model = TinyModel()
+    x = torch.randn(32, 10)
+    y = torch.randint(0, 3, (32,))
+    precon = KFACPreconditioner(
+        model,
+        factor_update_steps=1,
+        inv_update_steps=2,
+        lr=0.01,
+        inv_dtype=torch.bfloat16,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    losses = []
+    for _ in range(12):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[0] > losses[-1], losses
+    layer = next(iter(precon._layers.values()))[1]
+    assert layer.qa.dtype == torch.bfloat16
+    assert layer.qg.dtype == torch.bfloat16
+
+
+def test_reduced_precision_inverse_state() -> None:
+    run_distributed(1, _reduced_precision_case)
