@@ -166,6 +166,7 @@ def _half_batch_distributed(path: str, **kfac_kwargs) -> None:
         {'symmetry_aware': True},
         # prediv (eigenvalue outer product) requires colocated factors
         {'colocate_factors': False, 'compute_eigenvalue_outer_product': False},
+        {'update_factors_in_hook': False},
     ],
     ids=[
         'comm-opt',
@@ -174,6 +175,7 @@ def _half_batch_distributed(path: str, **kfac_kwargs) -> None:
         'inverse',
         'symmetry-aware',
         'no-colocate',
+        'factors-in-step',
     ],
 )
 def test_distributed_matches_single_process(kfac_kwargs) -> None:
