@@ -330,3 +330,38 @@ def test_forced_phase_joins_inflight_async_job() -> None:
         )
     if precon._async_job is not None:
         precon._finish_async_inverses()
+
+
+@pytest.mark.parametrize(
+    'kwargs',
+    [
+        {'compute_method': 'inverse'},
+        {'symmetry_aware': True},
+    ],
+    ids=['inverse-method', 'symmetry-aware'],
+)
+def test_training_variants_gpu(kwargs) -> None:
+    """GPU e2e for the explicit-inverse method (batched Cholesky group
+    path) and the triu-packed symmetric wire format."""
+    from kfac_amd import KFACPreconditioner
+    from testing.models import LeNet
+
+    torch.manual_seed(42)
+    model = LeNet().cuda()
+    x = torch.randn(64, 1, 28, 28, device='cuda')
+    y = torch.randint(0, 10, (64,), device='cuda')
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    precon = KFACPreconditioner(
+        model, factor_update_steps=1, inv_update_steps=2, lr=0.01, **kwargs,
+    )
+    losses = []
+    for _ in range(10):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[0] > losses[-1], losses
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
