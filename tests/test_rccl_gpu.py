@@ -141,3 +141,40 @@ def test_rccl_full_kfac_step(rccl_world1) -> None:
     # kl-clip bounds the per-step movement, so expect steady (not
     # dramatic) descent on the fixed batch: >= 10% in 20 steps.
     assert losses[-1] < 0.9 * losses[0], losses
+
+
+def test_gpt_neox_training_over_rccl(rccl_world1) -> None:
+    """GPT-NeoX preconditioner end-to-end on cuda over a real RCCL
+    group (world 1): gather/precondition/scatter branches, dp-group
+    factor reduce, and the gloo-subgroup checkpoint gather."""
+    from kfac_amd.gpt_neox import GPTNeoXKFACPreconditioner
+    from kfac_amd.gpt_neox.topology import PipeModelDataTopology
+    from testing.gpt_neox import ParallelMLP
+
+    torch.manual_seed(0)
+    model = ParallelMLP(in_dim=32, hidden=64, out_dim=8).cuda()
+    precon = GPTNeoXKFACPreconditioner(
+        model,
+        topology=PipeModelDataTopology(num_pp=1, num_mp=1, num_dp=1),
+        data_parallel_group=dist.new_group([0]),
+        model_parallel_group=None,
+        factor_update_steps=1,
+        inv_update_steps=2,
+        lr=0.05,
+    )
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    x = torch.randn(64, 32, device='cuda')
+    y = torch.randint(0, 8, (64,), device='cuda')
+    losses = []
+    for _ in range(8):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[0] > losses[-1], losses
+    sd = precon.state_dict()
+    assert set(sd['layers'].keys()) == {'dense_h_to_4h', 'dense_4h_to_h'}
+    precon.load_state_dict(sd, compute_inverses=True)
+    torch.cuda.synchronize()
