@@ -57,12 +57,20 @@ class LambdaParamScheduler:
             'lr': lr_lambda,
         }
         for name, fn in self._lambdas.items():
-            if fn is not None and callable(
-                getattr(preconditioner, f'_{name}'),
-            ):
+            if fn is None:
+                continue
+            current = getattr(preconditioner, f'_{name}')
+            if callable(current):
                 raise ValueError(
                     f'preconditioner.{name} is already a callable and '
                     'cannot be updated by the LambdaParamScheduler.',
+                )
+            if current is None:
+                # e.g. kl_clip=None disables clipping; multiplying None
+                # at step time would be a bare TypeError much later
+                raise ValueError(
+                    f'preconditioner.{name} is None (disabled) and '
+                    'cannot be scheduled.',
                 )
 
     def step(self, step: int | None = None) -> None:

@@ -570,3 +570,12 @@ def test_state_dict_second_order_resume() -> None:
     torch.nn.functional.cross_entropy(model2(x), y).backward()
     p2.step()
     opt2.step()
+
+
+def test_scheduler_rejects_disabled_param() -> None:
+    """Scheduling a None (disabled) hyperparameter fails at construction,
+    not with a TypeError deep in step()."""
+    model = TinyModel()
+    p = KFACPreconditioner(model, kl_clip=None)
+    with pytest.raises(ValueError, match='disabled'):
+        LambdaParamScheduler(p, kl_clip_lambda=lambda s: 0.99)
