@@ -365,3 +365,24 @@ def test_training_variants_gpu(kwargs) -> None:
     assert losses[0] > losses[-1], losses
     for p in model.parameters():
         assert torch.isfinite(p).all()
+
+
+def test_trace_cuda_sync_ranges() -> None:
+    """trace(cuda_sync=True) brackets with device sync + roctx/nvtx
+    ranges on a real GPU (the rocprofv3 timeline hook)."""
+    from kfac_amd import tracing
+
+    tracing.clear_trace()
+
+    @tracing.trace(cuda_sync=True)
+    def _work() -> torch.Tensor:
+        x = torch.randn(512, 512, device='cuda')
+        return x @ x
+
+    for _ in range(3):
+        _work()
+    t = tracing.get_trace()
+    key = next(iter(t))
+    assert '_work' in key
+    assert t[key] > 0.0
+    tracing.clear_trace()
