@@ -105,3 +105,20 @@ def test_gpt_neox_mlp_example_runs() -> None:
     )
     assert r.returncode == 0, r.stderr[-2000:]
     assert 'done' in r.stdout
+
+
+def test_imagenet_example_runs() -> None:
+    """ImageNet CLI: 2 synthetic steps of resnet50 on CPU."""
+    r = _run(
+        [
+            'examples/torch_imagenet_resnet.py',
+            '--synthetic',
+            '--epochs', '1',
+            '--max-steps-per-epoch', '2',
+            '--batch-size', '4',
+            '--val-batch-size', '4',
+            '--kfac-inv-update-steps', '2',
+        ],
+        timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
