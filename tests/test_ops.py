@@ -199,3 +199,19 @@ def test_module_helper_get_factor_api() -> None:
     g = torch.randn(12, 3)
     fg = h.get_g_factor(g)
     torch.testing.assert_close(fg, get_cov(g), rtol=1e-5, atol=1e-6)
+
+
+def test_require_ext_fails_loudly(monkeypatch) -> None:
+    """On a GPU box without the built extension, ops must raise — no
+    silent eager fallback (round requirement)."""
+    from kfac_amd import ops as _ops
+
+    monkeypatch.setattr(_ops, '_EXT', None)
+    monkeypatch.setattr(_ops, '_EXT_TRIED', True)
+    monkeypatch.delenv('KFAC_AMD_ALLOW_EAGER', raising=False)
+    monkeypatch.delenv('KFAC_AMD_FORCE_EAGER', raising=False)
+    with pytest.raises(RuntimeError, match='not built'):
+        _ops._require_ext('cov_linear')
+    # the two debug escapes return None instead
+    monkeypatch.setenv('KFAC_AMD_ALLOW_EAGER', '1')
+    assert _ops._require_ext('cov_linear') is None
