@@ -602,6 +602,14 @@ void scatter_cols(
   check_gpu_contig(mat, "mat");
   check_gpu_contig(idx, "idx");
   check_gpu_contig(src, "src");
+  TORCH_CHECK(
+      t.dim() == 3 && t.scalar_type() == torch::kFloat32 &&
+          src.scalar_type() == torch::kFloat32,
+      "t/src must be 3D fp32");
+  TORCH_CHECK(
+      mat.scalar_type() == torch::kInt64 &&
+          idx.scalar_type() == torch::kInt64,
+      "int64 indices required");
   const int p = (int)mat.size(0);
   const int rows = (int)t.size(1);
   const int n = (int)t.size(2);
