@@ -960,7 +960,12 @@ std::tuple<torch::Tensor, torch::Tensor> eigh_jacobi(
   const int B = (int)stack.size(0);
   const int n = (int)stack.size(1);
 
+  // One mutex across the whole call: the handle's bound stream must not
+  // change between set_stream and the launch if another thread (e.g.
+  // the async-inverse worker) calls in concurrently.
+  static std::mutex mu;
   static rocblas_handle handle = nullptr;
+  std::lock_guard<std::mutex> lock(mu);
   if (handle == nullptr) {
     TORCH_CHECK(
         rocblas_create_handle(&handle) == rocblas_status_success,
