@@ -98,11 +98,10 @@ def create_lr_schedule(
                 * (epoch * (workers - 1) / warmup_epochs + 1)
             )
         else:
-            decay_schedule.sort(reverse=True)
-            for e in decay_schedule:
-                if epoch >= e:
-                    lr_adj = alpha ** (decay_schedule.index(e) + 1)
-                    break
+            # one factor of alpha per decay epoch already passed
+            # (reference examples/utils.py:108-111 multiplies alpha for
+            # every e <= epoch)
+            lr_adj = alpha ** sum(1 for e in decay_schedule if epoch >= e)
         return lr_adj
 
     return lr_schedule

@@ -3,6 +3,7 @@
 from __future__ import annotations
 
 import os
+import pytest
 import subprocess
 import sys
 import tempfile
@@ -122,3 +123,66 @@ def test_imagenet_example_runs() -> None:
         timeout=600,
     )
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_example_utils_units() -> None:
+    """Unit semantics of the shared example utilities (previously only
+    covered through the CLI smokes)."""
+    import torch
+
+    sys.path.insert(0, REPO)
+    from examples.utils import accuracy
+    from examples.utils import create_lr_schedule
+    from examples.utils import LabelSmoothLoss
+    from examples.utils import Metric
+
+    # LabelSmoothLoss at smoothing=0 == plain cross entropy
+    torch.manual_seed(0)
+    x = torch.randn(16, 10)
+    y = torch.randint(0, 10, (16,))
+    torch.testing.assert_close(
+        LabelSmoothLoss(0.0)(x, y),
+        torch.nn.functional.cross_entropy(x, y),
+    )
+    # smoothing mixes toward uniform: loss increases on confident preds
+    conf = torch.full((4, 10), -10.0)
+    t = torch.arange(4) % 10
+    conf[torch.arange(4), t] = 10.0
+    assert LabelSmoothLoss(0.1)(conf, t) > LabelSmoothLoss(0.0)(conf, t)
+
+    # Metric averages over updates (single-process path)
+    m = Metric('loss')
+    m.update(torch.tensor(2.0))
+    m.update(torch.tensor(4.0))
+    assert float(m.avg) == 3.0
+
+    # warmup ramps from 1/workers to 1, then one factor of alpha per
+    # decay epoch passed (epoch 25 is past BOTH 10 and 20 -> alpha^2;
+    # this was inverted before this test existed)
+    sched = create_lr_schedule(8, warmup_epochs=4, decay_schedule=[10, 20])
+    assert sched(0) == 1.0 / 8
+    assert sched(5) == 1.0
+    assert sched(10) == pytest.approx(0.1)
+    assert sched(15) == pytest.approx(0.1)
+    assert sched(25) == pytest.approx(0.01)
+
+    assert float(accuracy(x, x.argmax(dim=1))) == 1.0
+
+
+def test_save_checkpoint_roundtrip(tmp_path) -> None:
+    import torch
+
+    sys.path.insert(0, REPO)
+    from examples.utils import save_checkpoint
+    from kfac_amd import KFACPreconditioner
+    from testing.models import TinyModel
+
+    model = TinyModel()
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    precon = KFACPreconditioner(model)
+    path = str(tmp_path / 'ckpt.pth.tar')
+    save_checkpoint(model, opt, precon, [], path, epoch=3)
+    state = torch.load(path, weights_only=False)
+    assert state['epoch'] == 3
+    assert set(state['model'].keys()) == set(model.state_dict().keys())
+    assert state['preconditioner']['steps'] == 0
