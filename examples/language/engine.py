@@ -18,7 +18,7 @@ def train_epoch(
     batch_fn: Callable[[int], tuple[torch.Tensor, torch.Tensor]],
     steps: int,
     vocab: int,
-    grad_clip: float = 0.25,
+    grad_clip: float = 0.5,  # reference examples/language/engine.py:53
 ) -> float:
     """One epoch over ``steps`` batches; returns train perplexity."""
     model.train()
