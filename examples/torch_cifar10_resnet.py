@@ -49,7 +49,9 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--warmup-epochs', type=int, default=5)
     p.add_argument('--momentum', type=float, default=0.9)
     p.add_argument('--weight-decay', type=float, default=5e-4)
-    p.add_argument('--label-smoothing', type=float, default=0.1)
+    # reference CIFAR example uses plain cross entropy (its
+    # torch_cifar10_resnet.py:340); smoothing is opt-in here
+    p.add_argument('--label-smoothing', type=float, default=0.0)
     p.add_argument('--checkpoint-dir', type=str, default='/tmp/kfac_ckpt')
     p.add_argument('--checkpoint-freq', type=int, default=10)
     p.add_argument('--amp', action='store_true', help='fp16 GradScaler AMP')
