@@ -28,22 +28,30 @@ def parse_args() -> argparse.Namespace:
         description='Transformer LM + K-FAC',
         formatter_class=argparse.ArgumentDefaultsHelpFormatter,
     )
-    p.add_argument('--seq-len', type=int, default=35)
+    # flag names and defaults match the reference CLI
+    # (torch_language_model.py:35-160); --vocab replaces its dataset
+    # download flags (offline synthetic token streams)
+    p.add_argument('--seq-len', type=int, default=64)
     p.add_argument('--batch-size', type=int, default=20)
     p.add_argument('--vocab', type=int, default=8192)
-    p.add_argument('--emsize', type=int, default=200)
-    p.add_argument('--nhead', type=int, default=2)
-    p.add_argument('--nhid', type=int, default=200)
-    p.add_argument('--nlayers', type=int, default=2)
+    p.add_argument('--embedding-dim', '--emsize', dest='emsize',
+                   type=int, default=256)
+    p.add_argument('--attention-heads', '--nhead', dest='nhead',
+                   type=int, default=4)
+    p.add_argument('--hidden-dim', '--nhid', dest='nhid',
+                   type=int, default=256)
+    p.add_argument('--layers', '--nlayers', dest='nlayers',
+                   type=int, default=2)
     p.add_argument('--dropout', type=float, default=0.2)
-    p.add_argument('--epochs', type=int, default=10)
+    p.add_argument('--epochs', type=int, default=20)
     p.add_argument('--steps-per-epoch', type=int, default=200)
     p.add_argument('--lr', type=float, default=1.0)
     p.add_argument('--backend', type=str, default=None, choices=['nccl', 'gloo'])
     p.add_argument('--seed', type=int, default=42)
     p.add_argument('--kfac-inv-update-steps', type=int, default=10)
     p.add_argument('--kfac-factor-update-steps', type=int, default=1)
-    p.add_argument('--kfac-damping', type=float, default=0.01)
+    p.add_argument('--kfac-factor-decay', type=float, default=0.95)
+    p.add_argument('--kfac-damping', type=float, default=0.003)
     p.add_argument('--kfac-kl-clip', type=float, default=0.001)
     p.add_argument(
         '--kfac-skip-layers',
@@ -87,6 +95,7 @@ def main() -> None:
             model,
             factor_update_steps=args.kfac_factor_update_steps,
             inv_update_steps=args.kfac_inv_update_steps,
+            factor_decay=args.kfac_factor_decay,
             damping=args.kfac_damping,
             kl_clip=args.kfac_kl_clip,
             lr=lambda x: optimizer.param_groups[0]['lr'],
