@@ -40,8 +40,8 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--data-dir', type=str, default=None)
     p.add_argument('--synthetic', action='store_true', default=True)
     p.add_argument('--model', type=str, default='resnet50', choices=MODELS)
-    p.add_argument('--batch-size', type=int, default=64)
-    p.add_argument('--val-batch-size', type=int, default=64)
+    p.add_argument('--batch-size', type=int, default=32)
+    p.add_argument('--val-batch-size', type=int, default=32)
     p.add_argument('--batches-per-allreduce', type=int, default=1)
     p.add_argument('--epochs', type=int, default=55)
     p.add_argument('--base-lr', type=float, default=0.0125)
