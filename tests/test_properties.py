@@ -263,3 +263,53 @@ def test_warm_eigh_block_sparse_property(
         (d.squeeze(0).sort().values.to(torch.float64) - w_ref).abs().max(),
     )
     assert err < 1e-3, err
+
+
+@SETTINGS
+@given(
+    world=st.integers(min_value=1, max_value=96),
+    gw_pick=st.integers(min_value=0, max_value=10),
+    n_layers=st.integers(min_value=1, max_value=12),
+    seed=st.integers(min_value=0, max_value=999),
+    colocate=st.booleans(),
+)
+def test_grid_and_assignment_any_world(
+    world: int, gw_pick: int, n_layers: int, seed: int, colocate: bool,
+) -> None:
+    """KAISA invariants hold for ARBITRARY world sizes (6, 12, 96, ...),
+    not just powers of two, and the greedy assignment keeps every
+    layer's factors inside one gradient-worker column."""
+    import random
+
+    divisors = [w for w in range(1, world + 1) if world % w == 0]
+    gw = divisors[gw_pick % len(divisors)]
+    cols = KAISAAssignment.partition_grad_workers(world, gw)
+    rows = KAISAAssignment.partition_grad_receivers(world, gw)
+    assert sorted(r for s in cols for r in s) == list(range(world))
+    assert sorted(r for s in rows for r in s) == list(range(world))
+    assert all(len(c) == gw for c in cols)
+    assert all(len(r) == world // gw for r in rows)
+    for c in cols:
+        for r in rows:
+            assert len(c & r) == 1
+
+    rng = random.Random(seed)
+    work = {
+        f'l{i}': {
+            'A': float(rng.randint(1, 4096)) ** 3,
+            'G': float(rng.randint(1, 4096)) ** 3,
+        }
+        for i in range(n_layers)
+    }
+    assignment = KAISAAssignment.greedy_assignment(
+        work,
+        [sorted(ranks) for ranks in sorted(cols, key=lambda s: sorted(s))],
+        world,
+        colocate,
+    )
+    for layer, factors in assignment.items():
+        workers = set(factors.values())
+        if colocate:
+            assert len(workers) == 1
+        # both factors always land inside ONE column
+        assert any(workers <= c for c in cols), (layer, workers)
