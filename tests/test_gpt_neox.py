@@ -788,3 +788,51 @@ def test_tp_checkpoint_resume_matches_uninterrupted(mode: str) -> None:
     else:
         with tempfile.TemporaryDirectory() as td:
             run_distributed(2, _tp_checkpoint_resume, f'{td}/factors')
+
+
+def test_reshape_data_semantics() -> None:
+    from kfac_amd.layers.utils import reshape_data
+
+    a = torch.arange(12.0).reshape(2, 3, 2)
+    b = torch.arange(6.0).reshape(1, 3, 2)
+    out = reshape_data([a, b])
+    assert out.shape == (3, 3, 2)
+    torch.testing.assert_close(out[:2], a)
+    flat = reshape_data([a, b], collapse_dims=True)
+    assert flat.shape == (9, 2)
+    sf = reshape_data(
+        [a.transpose(0, 1), b.transpose(0, 1)], batch_first=False,
+    )
+    assert sf.shape == (3, 3, 2)
+
+
+def _gather_mp_region() -> None:
+    from kfac_amd.gpt_neox.mpu import gather_from_model_parallel_region
+
+    rank = dist.get_rank()
+    group = dist.new_group([0, 1])
+    shard = torch.full((2, 3), float(rank))
+    out = gather_from_model_parallel_region(shard, dst=0, model_parallel_group=group)
+    if rank == 0:
+        assert out.shape == (2, 6)
+        torch.testing.assert_close(out[:, :3], torch.zeros(2, 3))
+        torch.testing.assert_close(out[:, 3:], torch.ones(2, 3))
+    else:
+        assert out is None
+    # bf16 + fp32_allreduce roundtrip keeps the bf16 dtype
+    shard16 = torch.full((2, 2), float(rank + 1), dtype=torch.bfloat16)
+    out16 = gather_from_model_parallel_region(
+        shard16, dst=0, model_parallel_group=group, fp32_allreduce=True,
+    )
+    if rank == 0:
+        assert out16.dtype == torch.bfloat16
+        assert out16.shape == (2, 4)
+    # size-1 group short-circuits to the input
+    same = gather_from_model_parallel_region(
+        shard, dst=rank, model_parallel_group=None,
+    )
+    assert same is shard
+
+
+def test_gather_from_model_parallel_region() -> None:
+    run_distributed(2, _gather_mp_region)
