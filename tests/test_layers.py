@@ -277,3 +277,47 @@ def test_conv3d_end_to_end() -> None:
         opt.step()
         losses.append(loss.item())
     assert losses[-1] < losses[0]
+
+
+def test_nonsymmetric_eigen_path() -> None:
+    """K6: a module helper reporting non-symmetric factors routes
+    through torch.linalg.eig (.real parts); the preconditioned gradient
+    is basis-invariant, so it must match the symmetric eigh path
+    (reference tests/layers/layers_test.py test_nonsymmetric_eigen)."""
+
+    class NonSymHelper(LinearModuleHelper):
+        def has_symmetric_factors(self) -> bool:
+            return False
+
+    torch.manual_seed(3)
+    in_dim, out_dim, batch = 7, 4, 32
+    results = {}
+    for helper_cls in (LinearModuleHelper, NonSymHelper):
+        torch.manual_seed(3)
+        module = torch.nn.Linear(in_dim, out_dim)
+        module.weight.grad = torch.randn(out_dim, in_dim)
+        module.bias.grad = torch.randn(out_dim)
+        layer = KFACEigenLayer(
+            helper_cls(module),
+            tdc=TorchDistributedCommunicator(),
+            allreduce_method=AllreduceMethod.ALLREDUCE,
+        )
+        assert layer.symmetric_factors == (helper_cls is LinearModuleHelper)
+        x = torch.randn(batch, in_dim)
+        g = torch.randn(batch, out_dim)
+        layer.save_layer_input([x])
+        layer.save_layer_grad_output((g,))
+        layer.update_a_factor(0.95)
+        layer.update_g_factor(0.95)
+        layer.compute_a_inv(damping=1e-3)
+        layer.compute_g_inv(damping=1e-3)
+        assert layer.qa.dtype == torch.float32
+        assert (layer.da >= 0).all() and (layer.dg >= 0).all()
+        layer.preconditioned_grad(damping=1e-3)
+        results[helper_cls.__name__] = layer.grad.clone()
+    torch.testing.assert_close(
+        results['NonSymHelper'],
+        results['LinearModuleHelper'],
+        rtol=1e-4,
+        atol=1e-5,
+    )
