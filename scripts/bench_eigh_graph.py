@@ -1,4 +1,7 @@
-"""Micro: graph-replayed batched syevd vs torch.linalg.eigh (ResNet-50 groups)."""
+"""Micro: ops.eigh_batched (cold path) vs torch.linalg.eigh on the
+ResNet-50 factor groups.  (Round 1 benchmarked the hipGraph-replayed
+syevd here; that machinery was removed as unsound in round 2 — this now
+measures the direct batched call the warm solver falls back to.)"""
 from __future__ import annotations
 
 import sys
