@@ -186,3 +186,46 @@ def test_save_checkpoint_roundtrip(tmp_path) -> None:
     assert state['epoch'] == 3
     assert set(state['model'].keys()) == set(model.state_dict().keys())
     assert state['preconditioner']['steps'] == 0
+
+
+def test_vision_engine_accumulation_boundaries() -> None:
+    """engine.train steps the optimizer/preconditioner only at
+    accumulation boundaries and scales micro-losses by 1/steps."""
+    import torch
+
+    sys.path.insert(0, REPO)
+    from examples.vision.engine import train
+    from kfac_amd import KFACPreconditioner
+    from testing.models import TinyModel
+
+    torch.manual_seed(0)
+    model = TinyModel()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    precon = KFACPreconditioner(
+        model, factor_update_steps=1, inv_update_steps=1, lr=0.05,
+    )
+    steps = {'n': 0}
+    orig_step = opt.step
+
+    def counting_step(*a, **k):
+        steps['n'] += 1
+        return orig_step(*a, **k)
+
+    opt.step = counting_step
+    data = [
+        (torch.randn(8, 10), torch.randint(0, 3, (8,)))
+        for _ in range(6)
+    ]
+    train(
+        epoch=0,
+        model=model,
+        optimizer=opt,
+        preconditioner=precon,
+        loss_func=torch.nn.CrossEntropyLoss(),
+        train_loader=data,
+        device=torch.device('cpu'),
+        accumulation_steps=2,
+        log_interval=1000,
+    )
+    assert steps['n'] == 3  # 6 micro-batches / 2
+    assert precon.steps == 3  # preconditioner stepped only at boundaries
