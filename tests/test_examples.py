@@ -229,3 +229,24 @@ def test_vision_engine_accumulation_boundaries() -> None:
     )
     assert steps['n'] == 3  # 6 micro-batches / 2
     assert precon.steps == 3  # preconditioner stepped only at boundaries
+
+
+def test_language_dataset_units() -> None:
+    import torch
+
+    sys.path.insert(0, REPO)
+    from examples.language.dataset import corpus_batch
+    from examples.language.dataset import synthetic_batch
+
+    dev = torch.device('cpu')
+    x1, y1 = synthetic_batch(100, 4, 8, dev, seed=5)
+    x2, y2 = synthetic_batch(100, 4, 8, dev, seed=5)
+    torch.testing.assert_close(x1, x2)  # reproducible per seed
+    assert x1.shape == (8, 4) and y1.shape == (8 * 4,)
+    # target is the next-token shift of the input stream
+    torch.testing.assert_close(x1[1:].reshape(-1), y1[: 7 * 4])
+
+    corpus = torch.arange(1000) % 50
+    cx, cy = corpus_batch(corpus, batch=2, seq=5, step=0, device=dev)
+    assert cx.shape == (5, 2) and cy.shape == (10,)
+    torch.testing.assert_close(cy.view(5, 2)[:-1], cx[1:])  # shifted by one
