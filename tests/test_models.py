@@ -64,3 +64,12 @@ def test_gptneox_125m_size_and_forward() -> None:
     out = m(x)
     assert out.shape == (2, 16, 50304)
     assert torch.isfinite(out).all()
+
+
+def test_resnet_deep_variant_counts() -> None:
+    from kfac_amd.models import resnet101
+    from kfac_amd.models import resnet152
+
+    # torchvision-exact counts
+    assert _nparams(resnet101()) == 44_549_160
+    assert _nparams(resnet152()) == 60_192_808
