@@ -467,3 +467,41 @@ def _reduced_precision_case() -> None:
 
 def test_reduced_precision_inverse_state() -> None:
     run_distributed(1, _reduced_precision_case)
+
+
+def _ddp_half_batch(path: str) -> None:
+    """Real DistributedDataParallel wrapping (the README/examples usage):
+    registration walks the DDP wrapper, DDP itself averages gradients."""
+    with open(path, 'rb') as fh:
+        sd0, x, y, losses_single = pickle.load(fh)
+    rank = torch.distributed.get_rank()
+    model = TinyModel()
+    model.load_state_dict(sd0)
+    ddp = torch.nn.parallel.DistributedDataParallel(model)
+    precon = KFACPreconditioner(
+        ddp, factor_update_steps=1, inv_update_steps=2, lr=0.01,
+    )
+    opt = torch.optim.SGD(ddp.parameters(), lr=0.01)
+    half = x.size(0) // 2
+    xs = x[rank * half : (rank + 1) * half]
+    ys = y[rank * half : (rank + 1) * half]
+    losses = []
+    for _ in range(5):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(ddp(xs), ys)
+        loss.backward()  # DDP averages gradients here
+        precon.step()
+        opt.step()
+        losses.append(loss.item())
+    lt = torch.tensor(losses)
+    torch.distributed.all_reduce(lt)
+    lt /= 2
+    torch.testing.assert_close(
+        lt, torch.tensor(losses_single), rtol=1e-3, atol=1e-4,
+    )
+
+
+def test_ddp_wrapped_matches_single_process() -> None:
+    with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
+        run_distributed(1, _single_reference, f.name)
+        run_distributed(2, _ddp_half_batch, f.name)
