@@ -269,13 +269,11 @@ def test_syevj_small_clamps_negative() -> None:
     assert (w >= 0).all()
 
 
-def test_syevd_batched_graph_replay() -> None:
-    """rocSOLVER syevd behind hipGraph capture/replay (ops.eigh_batched).
-
-    The second and third calls with the same (B, n) shape but different
-    data exercise the REPLAY path (first call captures); results must
-    track the new inputs, not the captured ones.
-    """
+def test_syevd_batched_repeated_calls_track_data() -> None:
+    """Repeated ops.eigh_batched calls with the same (B, n) shape but
+    DIFFERENT data must track the new inputs (persistent workspace
+    buffers are cached per shape; this is the gate that exposed the
+    round-1 hipGraph replay as unsound before it was deleted)."""
     from kfac_amd import ops
 
     torch.manual_seed(7)
@@ -305,3 +303,24 @@ def test_eigh_batched_small_jacobi() -> None:
     torch.cuda.synchronize()
     recon = q @ torch.diag_embed(d) @ q.transpose(1, 2)
     torch.testing.assert_close(recon, stack, rtol=1e-3, atol=1e-3)
+
+
+def test_eigh_batched_degenerate_identity_screen() -> None:
+    """Fully-decayed scalar-identity factors (with denormal
+    off-diagonals) make rocSOLVER syevd fail silently (info != 0) and
+    even torch.linalg.eigh mangle the basis — the near-diagonal screen
+    must return the EXACT (diag, I) answer instead."""
+    from kfac_amd import ops
+
+    b, n = 4, 256
+    stack = 0.569 * torch.eye(n, device='cuda').expand(b, n, n).contiguous()
+    # denormal off-diagonal dust (the observed failure pattern)
+    stack = stack + 1e-42 * torch.randn(b, n, n, device='cuda').abs()
+    stack = 0.5 * (stack + stack.transpose(1, 2))
+    d, q = ops.eigh_batched(stack)
+    torch.cuda.synchronize()
+    eye = torch.eye(n, device='cuda').expand(b, n, n)
+    torch.testing.assert_close(q, eye)
+    torch.testing.assert_close(
+        d, stack.diagonal(dim1=-2, dim2=-1), rtol=0, atol=0,
+    )
