@@ -116,8 +116,8 @@ def test_imagenet_example_runs() -> None:
             '--synthetic',
             '--epochs', '1',
             '--max-steps-per-epoch', '2',
-            '--batch-size', '4',
-            '--val-batch-size', '4',
+            '--batch-size', '2',
+            '--val-batch-size', '2',
             '--kfac-inv-update-steps', '2',
         ],
         timeout=600,
