@@ -54,6 +54,7 @@ def parse_args() -> argparse.Namespace:
     p.add_argument('--checkpoint-freq', type=int, default=5)
     p.add_argument('--amp', action='store_true', help='fp16 GradScaler AMP')
     p.add_argument('--seed', type=int, default=42)
+    p.add_argument('--no-cuda', action='store_true', default=False)
     p.add_argument('--max-steps-per-epoch', type=int, default=None)
     p.add_argument('--backend', type=str, default=None, choices=['nccl', 'gloo'])
     p.add_argument('--kfac-inv-update-steps', type=int, default=100)
@@ -82,7 +83,7 @@ def main() -> None:
     args = parse_args()
     world = int(os.environ.get('WORLD_SIZE', '1'))
     local_rank = int(os.environ.get('LOCAL_RANK', '0'))
-    use_cuda = torch.cuda.is_available()
+    use_cuda = torch.cuda.is_available() and not args.no_cuda
     if world > 1:
         dist.init_process_group(args.backend or ('nccl' if use_cuda else 'gloo'))
     if use_cuda:
