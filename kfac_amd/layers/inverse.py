@@ -95,19 +95,54 @@ class KFACInverseLayer(KFACBaseLayer):
             symmetric=self.symmetric_factors and self.symmetry_aware,
         )
 
+    def _damped_inverse(
+        self,
+        factor: torch.Tensor,
+        prev: torch.Tensor | None,
+        damping: float,
+    ) -> torch.Tensor:
+        """(factor + damping I)^-1, warm-started from the previous
+        phase's inverse when available.
+
+        The INVERSE-method analog of the eigen path's warm solver:
+        factors are slowly-drifting EMAs, so two Newton-Schulz
+        iterations from the previous inverse replace the fresh
+        factorize+invert (ops.refine_inverse docstring has the cost
+        model). Residual-certified — a failed refinement falls back to
+        the exact inverse, so numerics never degrade. Disable with
+        KFAC_AMD_WARM_INV=0.
+        """
+        import os
+
+        if (
+            isinstance(prev, torch.Tensor)
+            and prev.shape == factor.shape
+            and os.environ.get('KFAC_AMD_WARM_INV', '1') == '1'
+        ):
+            f32 = factor.to(torch.float32)
+            m = f32 + damping * torch.eye(
+                f32.size(0), dtype=f32.dtype, device=f32.device,
+            )
+            x, ok = ops.refine_inverse(m, prev.to(torch.float32))
+            if ok:
+                return x.to(self.inv_dtype)
+        return ops.inv_damped(factor, damping).to(self.inv_dtype)
+
     def compute_a_inv(self, damping: float = 0.001) -> None:
         """A^-1 = (A + damping I)^-1 in fp32 (reference inverse.py:186-199)."""
         a = self.a_factor
         if not isinstance(a, torch.Tensor):
             raise RuntimeError('Cannot invert A before A has been computed')
-        self.a_inv = ops.inv_damped(a, damping).to(self.inv_dtype)
+        prev = self._a_inv if isinstance(self._a_inv, torch.Tensor) else None
+        self.a_inv = self._damped_inverse(a, prev, damping)
 
     def compute_g_inv(self, damping: float = 0.001) -> None:
         """G^-1 = (G + damping I)^-1 in fp32 (reference inverse.py:201-213)."""
         g = self.g_factor
         if not isinstance(g, torch.Tensor):
             raise RuntimeError('Cannot invert G before G has been computed')
-        self.g_inv = ops.inv_damped(g, damping).to(self.inv_dtype)
+        prev = self._g_inv if isinstance(self._g_inv, torch.Tensor) else None
+        self.g_inv = self._damped_inverse(g, prev, damping)
 
     def preconditioned_grad(self, damping: float = 0.001) -> None:
         """grad <- G^-1 @ grad @ A^-1 (reference inverse.py:215-234)."""

@@ -376,6 +376,49 @@ def inv_damped(x: torch.Tensor, damping: float) -> torch.Tensor:
     return ref.inv_damped(x, damping)
 
 
+def refine_inverse(
+    m: torch.Tensor,
+    x0: torch.Tensor,
+    *,
+    max_iters: int = 4,
+    tol: float = 1e-6,
+) -> tuple[torch.Tensor, bool]:
+    """Newton-Schulz refinement of an approximate inverse of SPD ``m``.
+
+    X <- X (2I - M X), quadratic convergence while ||I - M X0|| < 1.
+    The INVERSE method's analog of the warm eigensolver: K-FAC factors
+    are slowly-drifting EMAs, so the previous phase's damped inverse is
+    an excellent X0 and 1-2 adaptive iterations (2 GEMMs each — the
+    residual check reuses the M X product) replace a fresh
+    factorize+invert whose rocSOLVER panel chain measured only 2-4 TF
+    at these shapes vs ~300 TF for GEMMs (profiles/qdwh_bench.md).
+    Returns ``(x, ok)`` where ``ok`` certifies
+    ||M X - I||_F <= tol * sqrt(n); callers MUST fall back to the exact
+    inverse when ``ok`` is False (cold start, damping change, or drift
+    too large — a bad X0 DIVERGES quadratically and can never pass the
+    certificate).
+    """
+    n = m.size(-1)
+    gate = tol * (float(n) ** 0.5)
+    eye = torch.eye(n, dtype=m.dtype, device=m.device)
+    x = x0
+    ok = False
+    for it in range(max_iters + 1):
+        r = m @ x
+        resid = float(torch.linalg.norm(r - eye))
+        if resid <= gate:
+            ok = True
+            break
+        if it == max_iters or not (resid < float(n)):
+            # out of budget, or far outside the convergence basin
+            break
+        x = x @ (2.0 * eye - r)
+    # the inverse of an SPD matrix is symmetric; enforce exactly so the
+    # triu wire format remains valid
+    x = 0.5 * (x + x.transpose(-1, -2))
+    return x, ok
+
+
 def triu_pack(x: torch.Tensor) -> torch.Tensor:
     """Symmetric-matrix wire format: upper triangle as a flat vector."""
     if x.is_cuda:

@@ -215,3 +215,70 @@ def test_require_ext_fails_loudly(monkeypatch) -> None:
     # the two debug escapes return None instead
     monkeypatch.setenv('KFAC_AMD_ALLOW_EAGER', '1')
     assert _ops._require_ext('cov_linear') is None
+
+
+def test_refine_inverse_converges_and_certifies() -> None:
+    """Newton-Schulz warm inverse: quadratic convergence from a drifted
+    start, residual certificate, refusal on garbage starts."""
+    from kfac_amd import ops as _ops
+
+    torch.manual_seed(2)
+    n = 96
+    b = torch.randn(n, n)
+    f = b @ b.T / n + 0.5 * torch.eye(n)
+    m = f + 1e-3 * torch.eye(n)
+    exact = torch.linalg.inv(m)
+
+    # drifted start (the EMA-update scenario)
+    pert = torch.randn(n, n) * 1e-3
+    x0 = exact + 0.5 * (pert + pert.T)
+    x, ok = _ops.refine_inverse(m, x0)
+    assert ok
+    torch.testing.assert_close(x, exact, rtol=1e-4, atol=1e-6)
+    torch.testing.assert_close(x, x.T)  # symmetric by construction
+
+    # exact start stays exact
+    x, ok = _ops.refine_inverse(m, exact.clone())
+    assert ok
+    torch.testing.assert_close(x, exact, rtol=1e-5, atol=1e-7)
+
+    # garbage start must be REFUSED, not returned
+    bad = torch.randn(n, n)
+    _, ok = _ops.refine_inverse(m, bad)
+    assert not ok
+
+
+def test_inverse_layer_warm_matches_cold(monkeypatch) -> None:
+    """KFACInverseLayer phases warm-started from the previous inverse
+    must match exact recomputation through a multi-phase run."""
+    from kfac_amd.distributed import TorchDistributedCommunicator
+    from kfac_amd.enums import AllreduceMethod
+    from kfac_amd.layers.inverse import KFACInverseLayer
+    from kfac_amd.layers.modules import LinearModuleHelper
+
+    results = {}
+    for warm in (True, False):
+        monkeypatch.setenv('KFAC_AMD_WARM_INV', '1' if warm else '0')
+        torch.manual_seed(4)
+        module = torch.nn.Linear(24, 12)
+        layer = KFACInverseLayer(
+            LinearModuleHelper(module),
+            tdc=TorchDistributedCommunicator(),
+            allreduce_method=AllreduceMethod.ALLREDUCE,
+        )
+        invs = []
+        for phase in range(4):
+            x = torch.randn(64, 24)
+            g = torch.randn(64, 12)
+            layer.save_layer_input([x])
+            layer.save_layer_grad_output((g,))
+            layer.update_a_factor(0.95)
+            layer.update_g_factor(0.95)
+            layer.compute_a_inv(damping=1e-3)
+            layer.compute_g_inv(damping=1e-3)
+            invs.append((layer.a_inv.clone(), layer.g_inv.clone()))
+        results[warm] = invs
+    for (aw, gw), (ac, gc) in zip(results[True], results[False]):
+        # certificate: ||M X - I||_F <= 1e-6 sqrt(n) -> entrywise ~1e-6
+        torch.testing.assert_close(aw, ac, rtol=1e-3, atol=2e-6)
+        torch.testing.assert_close(gw, gc, rtol=1e-3, atol=2e-6)
