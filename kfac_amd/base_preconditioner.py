@@ -923,6 +923,13 @@ class BaseKFACPreconditioner:
                     'computed',
                 )
             groups[(factor.shape[0], factor.device, factor.dtype)].append(layer)
+        import os
+
+        from kfac_amd import ops as _ops
+
+        warm_env = os.environ.get('KFAC_AMD_WARM_INV', '1') == '1'
+        # property access: resolves any in-flight broadcast future
+        attr = 'a_inv' if which == 'a' else 'g_inv'
         for (n, _dev, _dt), group in groups.items():
             stack = torch.stack(
                 [
@@ -933,11 +940,46 @@ class BaseKFACPreconditioner:
                 ],
             )
             stack.diagonal(dim1=-2, dim2=-1).add_(damping)
-            try:
-                chol = torch.linalg.cholesky(stack)
-                inv = torch.cholesky_inverse(chol)
-            except Exception:
-                inv = torch.linalg.inv(stack)
+            # Warm start: the previous phase's inverses seed a batched
+            # Newton-Schulz refinement (a few bmm launches); matrices
+            # failing the residual certificate fall through to the
+            # exact solve below. The eigen path's warm solver analog
+            # (ops.refine_inverse docstring has the cost model).
+            prev = [
+                p if isinstance(p := getattr(layer, attr), torch.Tensor)
+                and p.shape == (n, n)
+                else None
+                for layer in group
+            ]
+            inv = None
+            need_exact = list(range(len(group)))
+            if warm_env and all(p is not None for p in prev):
+                x0 = torch.stack([p.to(torch.float32) for p in prev])
+                refined, okm = _ops.refine_inverse_batched(stack, x0)
+                ok_host = okm.tolist()
+                if any(ok_host):
+                    inv = refined
+                    need_exact = [
+                        i for i, okv in enumerate(ok_host) if not okv
+                    ]
+            if need_exact:
+                sub = (
+                    stack
+                    if inv is None
+                    else stack[need_exact].contiguous()
+                )
+                try:
+                    chol = torch.linalg.cholesky(sub)
+                    exact = torch.cholesky_inverse(chol)
+                except Exception:
+                    exact = torch.linalg.inv(sub)
+                if inv is None:
+                    inv = exact
+                else:
+                    inv = inv.clone()
+                    for k, i in enumerate(need_exact):
+                        inv[i] = exact[k]
+            assert inv is not None
             for i, layer in enumerate(group):
                 result = inv[i].to(layer.inv_dtype).contiguous()
                 if which == 'a':

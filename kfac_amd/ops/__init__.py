@@ -419,6 +419,37 @@ def refine_inverse(
     return x, ok
 
 
+def refine_inverse_batched(
+    m: torch.Tensor,
+    x0: torch.Tensor,
+    *,
+    max_iters: int = 4,
+    tol: float = 1e-6,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Batched Newton-Schulz inverse refinement over a (B, n, n) stack.
+
+    Same contract as :func:`refine_inverse` but vectorized over the
+    batch (the grouped inverse phase refines every same-size factor in
+    a handful of bmm launches); returns ``(x, ok_mask)`` with a
+    per-matrix certificate — callers exact-solve the ``~ok`` subset.
+    """
+    bsz, n, _ = m.shape
+    gate = tol * (float(n) ** 0.5)
+    eye = torch.eye(n, dtype=m.dtype, device=m.device).expand_as(m)
+    x = x0
+    ok = torch.zeros(bsz, dtype=torch.bool, device=m.device)
+    for it in range(max_iters + 1):
+        r = m @ x
+        resid = torch.linalg.norm((r - eye).reshape(bsz, -1), dim=-1)
+        # NaN/Inf from a diverging (bad-start) matrix compares False
+        ok = resid <= gate
+        if it == max_iters or bool(ok.all()):
+            break
+        x = x @ (2.0 * eye - r)
+    x = 0.5 * (x + x.transpose(-1, -2))
+    return x, ok
+
+
 def triu_pack(x: torch.Tensor) -> torch.Tensor:
     """Symmetric-matrix wire format: upper triangle as a flat vector."""
     if x.is_cuda:

@@ -505,3 +505,44 @@ def test_ddp_wrapped_matches_single_process() -> None:
     with tempfile.NamedTemporaryFile(suffix='.pkl') as f:
         run_distributed(1, _single_reference, f.name)
         run_distributed(2, _ddp_half_batch, f.name)
+
+
+def _inverse_warm_cold_case() -> None:
+    """The batched warm (Newton-Schulz) inverse phase must reproduce the
+    exact-inverse trajectory across multiple phases."""
+    import os
+
+    params = {}
+    for warm in (True, False):
+        os.environ['KFAC_AMD_WARM_INV'] = '1' if warm else '0'
+        try:
+            torch.manual_seed(11)
+            model = TinyModel()
+            g = torch.Generator().manual_seed(6)
+            precon = KFACPreconditioner(
+                model,
+                factor_update_steps=1,
+                inv_update_steps=2,
+                lr=0.01,
+                compute_method='inverse',
+            )
+            opt = torch.optim.SGD(model.parameters(), lr=0.01)
+            for _ in range(8):
+                x = torch.randn(32, 10, generator=g)
+                y = torch.randint(0, 3, (32,), generator=g)
+                opt.zero_grad()
+                loss = torch.nn.functional.cross_entropy(model(x), y)
+                loss.backward()
+                precon.step()
+                opt.step()
+        finally:
+            os.environ.pop('KFAC_AMD_WARM_INV', None)
+        params[warm] = {k: v.clone() for k, v in model.state_dict().items()}
+    for k in params[True]:
+        torch.testing.assert_close(
+            params[True][k], params[False][k], rtol=1e-4, atol=1e-6,
+        )
+
+
+def test_inverse_warm_phase_matches_exact() -> None:
+    run_distributed(1, _inverse_warm_cold_case)
