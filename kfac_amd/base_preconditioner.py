@@ -980,12 +980,20 @@ class BaseKFACPreconditioner:
                     for k, i in enumerate(need_exact):
                         inv[i] = exact[k]
             assert inv is not None
+            exact_set = set(need_exact)
             for i, layer in enumerate(group):
                 result = inv[i].to(layer.inv_dtype).contiguous()
                 if which == 'a':
                     layer.a_inv = result
                 else:
                     layer.g_inv = result
+                # observability: count certified warm refinements (the
+                # analog of _warm_phases_a/g on the eigen path)
+                cnt_attr = f'_warm_inv_phases_{which}'
+                if i not in exact_set:
+                    setattr(
+                        layer, cnt_attr, getattr(layer, cnt_attr, 0) + 1,
+                    )
 
     # Force a dense (syevd) re-anchor after this many consecutive warm
     # phases, bounding any slow random walk of Q's orthogonality.

@@ -537,6 +537,16 @@ def _inverse_warm_cold_case() -> None:
                 opt.step()
         finally:
             os.environ.pop('KFAC_AMD_WARM_INV', None)
+        counts = [
+            getattr(layer, '_warm_inv_phases_a', 0)
+            for _, layer in precon._layers.values()
+        ]
+        # the warm path must actually ENGAGE (certified refinements) in
+        # the warm run — otherwise this equivalence proves nothing
+        if warm:
+            assert max(counts) >= 2, counts
+        else:
+            assert max(counts) == 0, counts
         params[warm] = {k: v.clone() for k, v in model.state_dict().items()}
     for k in params[True]:
         torch.testing.assert_close(
